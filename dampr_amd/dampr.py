@@ -1,0 +1,598 @@
+"""The user-facing fluent DSL.
+
+API parity with the reference's ``dampr/dampr.py`` (reference:
+dampr.py:19-977): same classes (``Dampr``, ``PMap``, ``PReduce``,
+``ARReduce``, ``PJoin``, ``ValueEmitter``), same methods and signatures,
+same lazy-graph semantics — consecutive maps fuse into one stage and a stage
+boundary appears only at ``checkpoint()`` (or the operators that imply one).
+
+Additions over the reference:
+
+* ``PJoin.outer_reduce`` — a *working* full outer join (the reference's
+  OuterJoin is broken dead code).
+* ``PMap.concat`` — declared but never implemented in the reference
+  (disabled ``_test_concat``).
+"""
+import itertools
+import json as json_mod
+import logging
+import operator
+import random
+import sys
+
+from .base import (BlockMapper, Map, MapAllJoin, MapCrossJoin, Mapper,
+                   ComposedMapper, ComposedStreamable, KeyedInnerJoin,
+                   KeyedLeftJoin, KeyedOuterJoin, KeyedCrossJoin,
+                   KeyedReduce, PartialReduceCombiner, Reduce, Reducer,
+                   StreamMapper, StreamReducer, Streamable)
+from .dataset import CatDataset, Chunker
+from .inputs import MemoryInput, PathInput
+from .runner import Graph, MTRunner, Source
+
+log = logging.getLogger("dampr_amd")
+
+
+class ValueEmitter(object):
+    """Reads values out of a finished computation."""
+
+    def __init__(self, dataset):
+        self.dataset = dataset
+
+    def stream(self):
+        for _k, v in self.dataset.read():
+            yield v
+
+    def read(self, k=None):
+        if k is None:
+            return list(self.stream())
+        return list(itertools.islice(self.stream(), k))
+
+    def __iter__(self):
+        return self.stream()
+
+    def delete(self):
+        self.dataset.delete()
+
+
+class PBase(object):
+    def __init__(self, source, pmer):
+        assert isinstance(source, Source)
+        self.source = source
+        self.pmer = pmer
+
+    def run(self, name=None, **kwargs):
+        """Evaluate the composed graph; returns a ValueEmitter."""
+        if name is None:
+            name = "dampr/{}".format(random.random())
+        ds = self.pmer.runner(name, self.pmer.graph, **kwargs) \
+                      .run([self.source])
+        return ValueEmitter(ds[0])
+
+    def read(self, k=None, **kwargs):
+        """run() followed by read()."""
+        return self.run(**kwargs).read(k)
+
+
+def _identity(k, v):
+    yield k, v
+
+
+class PMap(PBase):
+    """A (possibly fused) chain of map operations."""
+
+    def __init__(self, source, pmer, agg=None):
+        super(PMap, self).__init__(source, pmer)
+        self.agg = agg or []
+
+    def run(self, name=None, **kwargs):
+        if self.agg:
+            return self.checkpoint().run(name, **kwargs)
+        return super(PMap, self).run(name, **kwargs)
+
+    # -- fusion ------------------------------------------------------------
+
+    def _add_mapper(self, mapper):
+        assert isinstance(mapper, Streamable)
+        return PMap(self.source, self.pmer, self.agg + [mapper])
+
+    def _add_map(self, f):
+        return self._add_mapper(Map(f))
+
+    def checkpoint(self, force=False, combiner=None, options=None):
+        """Fuse the pending maps into one stage; shares the materialized
+        result among downstream consumers."""
+        if self.agg or force:
+            aggs = self.agg[:] if self.agg else [Map(_identity)]
+            name = "Stage {}: %s" % " -> ".join(str(a) for a in aggs)
+            source, pmer = self.pmer._add_mapper(
+                [self.source], fuse(aggs), combiner=combiner, name=name,
+                options=options)
+            return PMap(source, pmer)
+        return self
+
+    # -- low/medium-level hooks ---------------------------------------------
+
+    def custom_mapper(self, mapper, name=None, **options):
+        """Attach a user Mapper instance directly (does not fuse)."""
+        if isinstance(mapper, Streamable):
+            return self._add_mapper(mapper)
+        assert isinstance(mapper, Mapper)
+        name = name or str(mapper)
+        me = self.checkpoint()
+        source, pmer = me.pmer._add_mapper([me.source], mapper, name=name,
+                                           options=options)
+        return PMap(source, pmer)
+
+    def custom_reducer(self, reducer, name=None, **options):
+        """Attach a user Reducer instance directly."""
+        assert isinstance(reducer, Reducer)
+        me = self.checkpoint(force=True)
+        name = name or str(reducer)
+        source, pmer = me.pmer._add_reducer([me.source], reducer, name=name,
+                                            options=options)
+        return PMap(source, pmer)
+
+    def partition_map(self, f, **options):
+        """Map over a whole partition iterator; runs on empty partitions
+        too."""
+        return self.custom_mapper(StreamMapper(f), **options)
+
+    def partition_reduce(self, f):
+        """Reduce over a whole partition's (key, values) iterator; runs on
+        empty partitions too."""
+        return self.custom_reducer(StreamReducer(f))
+
+    # -- record-level operators ---------------------------------------------
+
+    def map(self, f):
+        """Map each value with f."""
+        def _map(k, v):
+            yield k, f(v)
+        return self._add_map(_map)
+
+    def map_values(self, f):
+        """Map the second element of two-tuple values."""
+        def _map_values(k, v):
+            yield k, (v[0], f(v[1]))
+        return self._add_map(_map_values)
+
+    def map_keys(self, f):
+        """Map the first element of two-tuple values."""
+        def _map_keys(k, v):
+            yield k, (f(v[0]), v[1])
+        return self._add_map(_map_keys)
+
+    def prefix(self, f):
+        """value -> (f(value), value)."""
+        def _prefix(k, v):
+            yield k, (f(v), v)
+        return self._add_map(_prefix)
+
+    def suffix(self, f):
+        """value -> (value, f(value))."""
+        def _suffix(k, v):
+            yield k, (v, f(v))
+        return self._add_map(_suffix)
+
+    def filter(self, f):
+        """Keep values where f(value) is true."""
+        def _filter(k, v):
+            if f(v):
+                yield k, v
+        return self._add_map(_filter)
+
+    def flat_map(self, f):
+        """Map each value to an iterable and flatten."""
+        def _flat_map(k, v):
+            for vi in f(v):
+                yield k, vi
+        return self._add_map(_flat_map)
+
+    def sample(self, prob):
+        """Uniformly keep each record with probability ``prob``."""
+        assert 0 <= prob <= 1.0
+
+        def _sample(k, v):
+            if _get_rand().random() < prob:
+                yield k, v
+        return self._add_map(_sample)
+
+    def inspect(self, prefix="", exit=False):
+        """Debug passthrough that prints each value."""
+        def _inspect(k, v):
+            print("{}: {}".format(prefix, v))
+            yield k, v
+        ins = self._add_map(_inspect)
+        if exit:
+            ins.run()
+            sys.exit(0)
+        return ins
+
+    # -- grouping -----------------------------------------------------------
+
+    def group_by(self, key, vf=lambda x: x):
+        """Group by key(value); general (non-associative) reductions."""
+        def _group_by(_k, value):
+            yield key(value), vf(value)
+        pm = self._add_map(_group_by).checkpoint()
+        return PReduce(pm.source, pm.pmer)
+
+    def a_group_by(self, key, vf=lambda x: x):
+        """Group by key(value) for *associative* reductions: enables the
+        map-side partial reduce (combiner), which is the fast path — on GPU
+        it lowers to the device hash-combine kernel (K6)."""
+        def _a_group_by(_k, value):
+            yield key(value), vf(value)
+        # No checkpoint: ARReduce attaches the combiner to this stage.
+        return ARReduce(self._add_map(_a_group_by))
+
+    def fold_by(self, key, binop, value=lambda x: x, **options):
+        """a_group_by(key, value).reduce(binop)."""
+        return self.a_group_by(key, value).reduce(binop, **options)
+
+    def sort_by(self, key, **options):
+        """Totally order the collection by key(value)."""
+        def _sort_by(_k, value):
+            yield key(value), value
+        return self._add_map(_sort_by).checkpoint(options=options)
+
+    def count(self, key=lambda x: x, **options):
+        """Count occurrences by key(value)."""
+        return self.a_group_by(key, lambda v: 1) \
+                   .reduce(operator.add, **options)
+
+    def mean(self, key=lambda x: 1, value=lambda x: x, **options):
+        """Mean of value(v) grouped by key(v)."""
+        def _mean_binop(x, y):
+            return x[0] + y[0], x[1] + y[1]
+
+        def _average(x):
+            return (x[0], x[1][0] / float(x[1][1]))
+
+        return self.a_group_by(key, lambda v: (value(v), 1)) \
+                   .reduce(_mean_binop, **options) \
+                   .map(_average)
+
+    def len(self):
+        """Number of records in the collection."""
+        def _map_count(items):
+            count = 0
+            for _ in items:
+                count += 1
+            yield 1, count
+
+        def _reduce_count(groups):
+            count = 0
+            seen = False
+            for _k, counts in groups:
+                seen = True
+                for c in counts:
+                    count += c
+            if seen:
+                yield 1, count
+
+        return self.partition_map(_map_count) \
+                   .partition_reduce(_reduce_count) \
+                   .map(lambda x: x[1])
+
+    def topk(self, k, value=None):
+        """Top-k values ordered by value(x) (K11)."""
+        if value is None:
+            value = lambda x: x
+        import heapq
+
+        def map_topk(it):
+            heap = []
+            for x in it:
+                heapq.heappush(heap, (value(x), x))
+                if len(heap) > k:
+                    heapq.heappop(heap)
+            return ((1, x) for x in heap)
+
+        def reduce_topk(it):
+            candidates = (v for _k, vit in it for v in vit)
+            for _score, x in heapq.nlargest(k, candidates):
+                yield x, 1
+
+        return self.partition_map(map_topk) \
+                   .partition_reduce(reduce_topk) \
+                   .map(lambda x: x[0])
+
+    # -- multi-graph operators ----------------------------------------------
+
+    def join(self, other):
+        """Reduce-side equi-join with another grouped computation."""
+        assert isinstance(other, PBase)
+        me = self.checkpoint(True)
+        if isinstance(other, PMap):
+            other = other.checkpoint(True)
+        pmer = Dampr(me.pmer.graph.union(other.pmer.graph))
+        return PJoin(me.source, pmer, other.source)
+
+    def concat(self, other):
+        """Concatenate two collections (new: the reference declares but
+        never implements this)."""
+        assert isinstance(other, PMap)
+        me = self.checkpoint()
+        other = other.checkpoint()
+        pmer = Dampr(me.pmer.graph.union(other.pmer.graph))
+        source, pmer = pmer._add_mapper(
+            [me.source, other.source], Map(_identity),
+            name="Stage {}: Concat", options={"concat": True})
+        return PMap(source, pmer)
+
+    def cross_left(self, other, cross, memory=False, **options):
+        """Cross product; self is the left side."""
+        def _cross(k1, v1, k2, v2):
+            yield k1, cross(v2, v1)
+
+        me = self.checkpoint()
+        other = other.checkpoint()
+        pmer = Dampr(me.pmer.graph.union(other.pmer.graph))
+        source, pmer = pmer._add_mapper(
+            [other.source, me.source], MapCrossJoin(_cross, cache=memory),
+            combiner=None, name="Stage {}: Cross", options=options)
+        return PMap(source, pmer)
+
+    def cross_right(self, other, cross, memory=False):
+        """Cross product; self is the right side.  With ``memory=True`` the
+        small side is cached (GPU path: broadcast, K9)."""
+        assert isinstance(other, PMap)
+        return other.cross_left(self, lambda xi, yi: cross(yi, xi), memory)
+
+    def cross_set(self, other, cross, agg=None, **options):
+        """Cross each value against the whole of ``other``, aggregated once
+        by ``agg`` (default list)."""
+        def _cross(k1, v1, right):
+            yield k1, cross(v1, right)
+
+        if agg is None:
+            agg = list
+
+        def _aggregate(d):
+            return agg(v for _k, v in d)
+
+        me = self.checkpoint()
+        other = other.checkpoint()
+        pmer = Dampr(me.pmer.graph.union(other.pmer.graph))
+        source, pmer = pmer._add_mapper(
+            [other.source, me.source], MapAllJoin(_cross, _aggregate),
+            combiner=None, name="Stage {}: CrossAll", options=options)
+        return PMap(source, pmer)
+
+    # -- caching and sinks --------------------------------------------------
+
+    def cached(self, **options):
+        """Materialize this subgraph in memory for reuse."""
+        options["memory"] = True
+        return self.checkpoint(options=options)
+
+    def sink(self, path):
+        """Write each value (assumed str) as a line into part files under
+        ``path``; exempt from cleanup."""
+        aggs = self.agg[:] if self.agg else [Map(_identity)]
+        name = "Stage {}: %s" % " -> ".join(str(a) for a in aggs)
+        source, pmer = self.pmer._add_sink(
+            [self.source], fuse(aggs), path=path, name=name, options=None)
+        return PMap(source, pmer)
+
+    def sink_tsv(self, path):
+        """Format tuples as TSV lines, then sink."""
+        return self.map(
+            lambda x: u"\t".join(str(p) for p in x)).sink(path)
+
+    def sink_json(self, path):
+        """Serialize values as line-delimited JSON, then sink."""
+        return self.map(json_mod.dumps).sink(path)
+
+
+class ARReduce(object):
+    """Associative reductions (map-side partial reduce enabled)."""
+
+    def __init__(self, pmap):
+        self.pmap = pmap
+
+    def reduce(self, binop, reduce_buffer=1000, **options):
+        """Reduce each group with an associative binop.  ``reduce_buffer``
+        caps the map-side combine dictionary (keys held in memory before a
+        spill) — unlike the reference, it is honored."""
+        def _reduce(key, vs):
+            acc = next(vs)
+            for v in vs:
+                acc = binop(acc, v)
+            return acc
+
+        options.update({"binop": binop, "reduce_buffer": reduce_buffer})
+        pm = self.pmap.checkpoint(
+            True, combiner=PartialReduceCombiner(Reduce(_reduce)),
+            options=options)
+        return PReduce(pm.source, pm.pmer).reduce(_reduce)
+
+    def first(self, **options):
+        """First value seen per key."""
+        return self.reduce(lambda x, _y: x, **options)
+
+    def sum(self, **options):
+        """Sum of values per key."""
+        return self.reduce(lambda x, y: x + y, **options)
+
+
+class PReduce(PBase):
+    """General grouped reductions."""
+
+    def reduce(self, f):
+        """Reduce each group with ``f(key, iter values) -> value``."""
+        source, pmer = self.pmer._add_reducer([self.source], KeyedReduce(f))
+        return PMap(source, pmer)
+
+    def unique(self, key=lambda x: x):
+        """Distinct values per group, by key(value)."""
+        def _uniq(_k, it):
+            seen = set()
+            out = []
+            for v in it:
+                fv = key(v)
+                if fv not in seen:
+                    seen.add(fv)
+                    out.append(v)
+            return out
+
+        return self.reduce(_uniq)
+
+    def join(self, other):
+        """Join with another grouped computation."""
+        assert isinstance(other, PBase)
+        if isinstance(other, PMap):
+            other = other.checkpoint(True)
+        pmer = Dampr(self.pmer.graph.union(other.pmer.graph))
+        return PJoin(self.source, pmer, other.source)
+
+    def partition_reduce(self, f):
+        source, pmer = self.pmer._add_reducer([self.source],
+                                              StreamReducer(f))
+        return PMap(source, pmer)
+
+
+class PJoin(PBase):
+    """Joins between two co-grouped computations (K8)."""
+
+    def __init__(self, source, pmer, right):
+        super(PJoin, self).__init__(source, pmer)
+        self.right = right
+
+    def run(self, name=None, **kwargs):
+        return self.reduce(
+            lambda l, r: (list(l), list(r))).run(name, **kwargs)
+
+    def reduce(self, aggregate, many=False):
+        """Inner join; aggregate(left_iter, right_iter).  ``many=True``
+        flattens an iterable result into separate records."""
+        def _reduce(_k, left, right):
+            return aggregate(left, right)
+
+        source, pmer = self.pmer._add_reducer(
+            [self.source, self.right], KeyedInnerJoin(_reduce, many))
+        return PMap(source, pmer)
+
+    def left_reduce(self, aggregate):
+        """Left join; missing right groups get an empty iterator."""
+        def _reduce(_k, left, right):
+            return aggregate(left, right)
+
+        source, pmer = self.pmer._add_reducer(
+            [self.source, self.right], KeyedLeftJoin(_reduce))
+        return PMap(source, pmer)
+
+    def outer_reduce(self, aggregate):
+        """Full outer join (new: the reference's outer join is broken dead
+        code — reference: base.py:337-366)."""
+        def _reduce(_k, left, right):
+            return aggregate(left, right)
+
+        source, pmer = self.pmer._add_reducer(
+            [self.source, self.right], KeyedOuterJoin(_reduce))
+        return PMap(source, pmer)
+
+    def _cross(self, crosser):
+        def _crossf(k1, v1, _k2, v2):
+            return k1, crosser(v1, v2)
+
+        source, pmer = self.pmer._add_reducer(
+            [self.source, self.right], KeyedCrossJoin(_crossf))
+        return PMap(source, pmer).map(lambda x: x[1])
+
+
+class Dampr(object):
+    """Entry points for building pipelines."""
+
+    def __init__(self, graph=None, runner=None):
+        self.graph = graph if graph is not None else Graph()
+        self.runner = runner if runner is not None else MTRunner
+
+    @classmethod
+    def memory(cls, items, partitions=50):
+        """Pipeline over an in-memory list."""
+        mi = MemoryInput(list(enumerate(items)), partitions)
+        source, ng = Graph().add_input(mi)
+        return PMap(source, cls(ng))
+
+    @classmethod
+    def read_input(cls, *datasets):
+        """Pipeline over custom Dataset/Chunker taps."""
+        if len(datasets) == 1:
+            ds = datasets[0]
+        else:
+            ds = CatDataset(list(datasets))
+        source, ng = Graph().add_input(ds)
+        return PMap(source, cls(ng))
+
+    @classmethod
+    def text(cls, fname, chunk_size=16 * 1024 ** 2, followlinks=False):
+        """Pipeline over newline-delimited files/dirs/globs."""
+        return cls.read_input(PathInput(fname, chunk_size, followlinks))
+
+    @classmethod
+    def json(cls, *args, **kwargs):
+        """text() then json-decode each line."""
+        return cls.text(*args, **kwargs).map(json_mod.loads)
+
+    @classmethod
+    def from_dataset(cls, dataset):
+        """Pipeline over a raw stage-output dataset."""
+        assert isinstance(dataset, Chunker)
+        source, ng = Graph().add_input(dataset)
+        return PMap(source, cls(ng))
+
+    @classmethod
+    def run(cls, *pmers, **kwargs):
+        """Run several pipelines as ONE merged DAG; shared checkpointed
+        subgraphs execute once.  Returns one ValueEmitter per input."""
+        assert len(pmers) > 0, "Need at least one graph to run!"
+        sources = []
+        graph = None
+        last = None
+        for i, pmer in enumerate(pmers):
+            if isinstance(pmer, PMap):
+                pmer = pmer.checkpoint()
+            elif isinstance(pmer, PJoin):
+                pmer = pmer.reduce(lambda l, r: (list(l), list(r)))
+            graph = pmer.pmer.graph if graph is None \
+                else pmer.pmer.graph.union(graph)
+            sources.append(pmer.source)
+            last = pmer
+        name = kwargs.pop("name", "dampr/{}".format(random.random()))
+        ds = last.pmer.runner(name, graph, **kwargs).run(sources)
+        return [ValueEmitter(d) for d in ds]
+
+    def _add_mapper(self, *args, **kwargs):
+        output, ng = self.graph.add_mapper(*args, **kwargs)
+        return output, Dampr(ng)
+
+    def _add_reducer(self, *args, **kwargs):
+        output, ng = self.graph.add_reducer(*args, **kwargs)
+        return output, Dampr(ng)
+
+    def _add_sink(self, *args, **kwargs):
+        output, ng = self.graph.add_sink(*args, **kwargs)
+        return output, Dampr(ng)
+
+
+def fuse(aggs):
+    """Compose a fused mapper chain into a single Mapper."""
+    if len(aggs) == 1:
+        return aggs[0]
+    s = aggs[1]
+    for i in range(2, len(aggs)):
+        s = ComposedStreamable(s, aggs[i])
+    return ComposedMapper(aggs[0], s)
+
+
+_RANDOM = None
+
+
+def _get_rand():
+    global _RANDOM
+    if _RANDOM is None:
+        import time
+        _RANDOM = random.Random(time.time())
+    return _RANDOM

@@ -1,0 +1,55 @@
+"""Module-global tunables.
+
+API parity with the reference's ``dampr/settings.py`` (reference: settings.py:1-37):
+users override by assignment, e.g. ``import dampr_amd.settings as settings;
+settings.partitions = 128``.  GPU knobs are new (no analog in the reference,
+which is CPU-only).
+"""
+import multiprocessing
+import os
+
+# ---------------------------------------------------------------- CPU engine
+
+# Number of worker processes per stage pool (reference: settings.py:5).
+max_processes = multiprocessing.cpu_count()
+
+# zlib compression level for spilled runs.  0 disables compression entirely
+# (fastest on NVMe); the reference always gzips at level 1 (settings.py:8).
+compress_level = 1
+
+# Number of reduce partitions (reference: settings.py:11).
+partitions = 91
+
+# Cap on file fan-in per stage; larger sets are compacted by merge passes
+# (reference: settings.py:16).
+max_files_per_stage = 50
+
+# Records per serialized frame in a spill file (reference: settings.py:20).
+batch_size = 4000
+
+# High-water RSS per worker process, in MB.  Crossing it flushes spill
+# buffers to disk (reference: settings.py:27).
+max_memory_per_worker = 512
+
+# Adaptive memory-check pacing (reference: settings.py:31-37).  The governor
+# estimates bytes/record and schedules the next RSS check; these bound it.
+memory_min_count = 10000
+memory_max_count_before_check = 100000
+
+# ---------------------------------------------------------------- GPU engine
+# New knobs for the MI355X path; no reference analog.
+
+# Device-side record-batch size (records per columnar batch).
+gpu_batch_records = 1 << 24
+
+# Fraction of HBM the buffer pool may occupy before spilling to pinned host.
+hbm_watermark = 0.90
+
+# Hash-table load factor for the device combine table (K6).
+gpu_table_load = 0.50
+
+# Partitions per GPU for the device-side shuffle.
+gpu_partitions_per_rank = 8
+
+# Directory for host-side spill of device batches.
+spill_dir = os.environ.get("DAMPR_SPILL_DIR", "/tmp")
