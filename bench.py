@@ -1,0 +1,143 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: TF-IDF over synthetic Zipf text (BASELINE.json
+config 2), whole-node rows/sec.
+
+One step = the complete TF-IDF job over this rank's resident corpus:
+tokenize + hash + per-doc dedupe + df-count on device, the RCCL exchange of
+(key, df, token-bytes) partials when world_size > 1, the idf epilogue, token
+string materialization and the TSV sink.  Nothing is cached across steps
+(tables are reset each step); data is synthetic (no network for datasets).
+
+Run: python bench.py --gpus N --steps K --warmup W [--mb-per-gpu M]
+For N > 1 the driver launches via torch.distributed.run; ranks read
+RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the environment.
+"""
+import argparse
+import json
+import os
+import shutil
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--mb-per-gpu", type=int,
+                    default=int(os.environ.get("DAMPR_BENCH_MB", "1024")))
+    ap.add_argument("--sink-dir", default="/tmp/dampr_amd_bench_idfs")
+    args = ap.parse_args()
+
+    if not torch.cuda.is_available():
+        raise SystemExit("bench.py requires an MI355X (no GPU visible)")
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    dist_mode = world > 1
+    if dist_mode:
+        import torch.distributed as dist
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+    device = torch.device("cuda", local_rank)
+    torch.cuda.set_device(device)
+
+    from dampr_amd.gpu.corpus import synth_corpus
+    from dampr_amd.gpu.tfidf import TfidfEngine
+    if dist_mode:
+        from dampr_amd.parallel.shuffle import (exchange_keyed_payload,
+                                                all_reduce_scalar)
+        import torch.distributed as dist
+
+    # ---- setup (untimed): per-rank synthetic corpus, resident in HBM
+    n_bytes = args.mb_per_gpu * (1 << 20)
+    text_np = synth_corpus(n_bytes, vocab=100_000, seed=1234 + rank)
+    text = torch.from_numpy(text_np).to(device)
+    n = text.numel()
+    line_bytes = 96                      # synth_corpus: 12 words x 8 bytes
+    docs_local = n // line_bytes
+    eng = TfidfEngine(device)
+    shutil.rmtree(args.sink_dir, ignore_errors=True)
+
+    def step():
+        eng.reset()
+        eng.count_chunk(text)
+        keys, df = eng.extract()
+        if dist_mode:
+            blob, lens = eng.token_strings_dev(keys, text)
+            rk, rdf, rblob, rlens = exchange_keyed_payload(
+                keys, df, blob, lens)
+            eng.merge_exchanged(rk, rdf, rblob, rlens)
+            keys, df = eng.extract()
+            total_docs = all_reduce_scalar(eng.n_docs, device=device)
+            src_text = rblob
+        else:
+            total_docs = eng.n_docs
+            src_text = text
+        idf = eng.idf(df, total_docs)
+        blob, lens = eng.token_strings(keys, src_text)
+        eng.sink_tsv(args.sink_dir, rank, blob, lens, df, idf)
+
+    def barrier_sync():
+        if dist_mode:
+            dist.barrier()
+        torch.cuda.synchronize(device)
+
+    for _ in range(args.warmup):
+        step()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if dist_mode:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        total_docs_job = all_reduce_scalar(docs_local, device=device)
+    else:
+        total_docs_job = docs_local
+
+    ms_per_step = elapsed * 1000.0 / args.steps
+    rows_per_sec = total_docs_job * args.steps / elapsed
+    gb_per_sec = (n * world / (1 << 30)) * args.steps / elapsed
+
+    if rank == 0:
+        n_gpus = world if dist_mode else args.gpus
+        print(json.dumps({
+            "metric": "tfidf_rows_per_sec",
+            "value": rows_per_sec,
+            "unit": "rows/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int64-exact",
+            "data": "synthetic",
+            "config": {
+                "model": "tfidf-docfreq",
+                "corpus_mb_per_gpu": args.mb_per_gpu,
+                "vocab": 100_000,
+                "global_batch": total_docs_job,
+                "seq_len": line_bytes,
+                "parallelism": "dp{}".format(n_gpus),
+                "gb_per_sec_ingest": gb_per_sec,
+            },
+        }))
+
+    if dist_mode:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,438 @@
+// dampr_amd gfx950 kernels: text ingest, tokenize+hash (K1), device
+// hash-table combine (K6), extraction, broadcast-apply (K9) and the idf
+// epilogue.  These are the MI355X-native replacements for the reference's
+// per-record Python hot loops; the roles, not the code,
+// come from dampr/dataset.py + dampr/base.py.
+//
+// Design notes (cdna_hip_programming.md):
+//  - memory-bound kernels: 256-thread blocks, vectorized 16 B/lane loads,
+//    grid capped with grid-stride loops (Guideline 11/13).
+//  - wave64 ballots (u64 masks) for in-block stable compaction.
+//  - all inter-block communication via device-scope atomics on HBM.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+#define BLOCK 256
+#define VBYTES 16                     // bytes per lane per iteration
+#define TILE (BLOCK * VBYTES)         // 4 KiB per block-iteration
+
+// Mark modes for the position scan.
+#define MODE_NEWLINE 0
+#define MODE_TOKEN_START 1
+
+__device__ __forceinline__ bool mark_at(int mode, const u8* text,
+                                        long i, long n) {
+    u8 c = text[i];
+    if (mode == MODE_NEWLINE) return c == '\n';
+    // token start: word char whose predecessor is not a word char
+    if (!is_word(c)) return false;
+    return i == 0 || !is_word(text[i - 1]);
+}
+
+// ---------------------------------------------------------------- scan pass 1
+// Each block owns a contiguous [base, base + iters*TILE) byte range and
+// counts its marks.
+__global__ void count_marks_kernel(const u8* __restrict__ text, long n,
+                                   int mode, int iters,
+                                   u32* __restrict__ counts) {
+    long base = (long)blockIdx.x * iters * TILE;
+    int tid = threadIdx.x;
+    u32 local = 0;
+    for (int it = 0; it < iters; ++it) {
+        long off = base + (long)it * TILE + (long)tid * VBYTES;
+        if (off >= n) break;
+        if (off + VBYTES <= n) {
+            uint4 v = *reinterpret_cast<const uint4*>(text + off);
+            const u8* b = reinterpret_cast<const u8*>(&v);
+            #pragma unroll
+            for (int j = 0; j < VBYTES; ++j) {
+                u8 c = b[j];
+                bool m;
+                if (mode == MODE_NEWLINE) {
+                    m = (c == '\n');
+                } else {
+                    bool prev_word = (j > 0)
+                        ? is_word(b[j - 1])
+                        : (off > 0 ? is_word(text[off - 1]) : false);
+                    m = is_word(c) && !prev_word;
+                }
+                local += m;
+            }
+        } else {
+            for (long i = off; i < n; ++i)
+                local += mark_at(mode, text, i, n);
+        }
+    }
+    // block reduction: wave reduce then LDS
+    __shared__ u32 wsum[BLOCK / WAVE];
+    for (int d = WAVE / 2; d > 0; d >>= 1)
+        local += __shfl_down(local, d, WAVE);
+    int lane = tid & (WAVE - 1), wid = tid / WAVE;
+    if (lane == 0) wsum[wid] = local;
+    __syncthreads();
+    if (tid == 0) {
+        u32 total = 0;
+        for (int w = 0; w < BLOCK / WAVE; ++w) total += wsum[w];
+        counts[blockIdx.x] = total;
+    }
+}
+
+// ---------------------------------------------------------------- scan pass 2
+// Re-scan and write mark positions in ascending order, starting at the
+// block's exclusive offset.  In-block order: per-iteration block-wide
+// exclusive scan of per-lane counts (wave shfl scan + LDS across waves).
+__global__ void write_marks_kernel(const u8* __restrict__ text, long n,
+                                   int mode, int iters,
+                                   const u32* __restrict__ block_offsets,
+                                   u32* __restrict__ out) {
+    long base = (long)blockIdx.x * iters * TILE;
+    int tid = threadIdx.x;
+    int lane = tid & (WAVE - 1), wid = tid / WAVE;
+    __shared__ u32 cursor;
+    __shared__ u32 wtot[BLOCK / WAVE];
+    if (tid == 0) cursor = block_offsets[blockIdx.x];
+    __syncthreads();
+
+    for (int it = 0; it < iters; ++it) {
+        long off = base + (long)it * TILE + (long)tid * VBYTES;
+        // gather this lane's marks (positions within its 16 bytes)
+        u32 cnt = 0;
+        u8 rel[VBYTES];
+        if (off < n) {
+            long lim = min((long)VBYTES, n - off);
+            for (long j = 0; j < lim; ++j)
+                if (mark_at(mode, text, off + j, n)) rel[cnt++] = (u8)j;
+        }
+        // block exclusive scan of cnt
+        u32 scan = cnt;
+        for (int d = 1; d < WAVE; d <<= 1) {
+            u32 x = __shfl_up(scan, d, WAVE);
+            if (lane >= d) scan += x;
+        }
+        if (lane == WAVE - 1) wtot[wid] = scan;
+        __syncthreads();
+        u32 wbase = 0;
+        for (int w = 0; w < wid; ++w) wbase += wtot[w];
+        u32 excl = cursor + wbase + scan - cnt;
+        for (u32 j = 0; j < cnt; ++j)
+            out[excl + j] = (u32)(off + rel[j]);
+        __syncthreads();
+        if (tid == 0) {
+            u32 tile_total = 0;
+            for (int w = 0; w < BLOCK / WAVE; ++w) tile_total += wtot[w];
+            cursor += tile_total;
+        }
+        __syncthreads();
+    }
+}
+
+// ------------------------------------------------------------ tokenize+count
+// One thread per token: hash the token (K1), locate its document (binary
+// search over newline positions), dedupe (doc, token) in the seen table
+// and bump the document-frequency table (K6).  First global occurrence of
+// a token also records (byte offset, length) in the string dictionary so
+// results can be materialized as text.
+__global__ void tfidf_count_kernel(
+        const u8* __restrict__ text, long n,
+        const u32* __restrict__ nl_pos, long n_nl,
+        const u32* __restrict__ tok_start, long n_tok,
+        u64* __restrict__ seen_keys, u64 seen_mask,
+        u64* __restrict__ cnt_keys, u64* __restrict__ cnt_vals,
+        u64 cnt_mask,
+        u64* __restrict__ dict_keys, u64* __restrict__ dict_vals,
+        u64 dict_mask, u64 pos_base, u64 doc_base) {
+    long stride = (long)gridDim.x * blockDim.x;
+    for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < n_tok;
+         t += stride) {
+        u32 start = tok_start[t];
+        long p = start;
+        u64 h = FNV_OFFSET;
+        while (p < n) {
+            u8 c = text[p];
+            if (!is_word(c)) break;
+            h = fnv1a64_step(h, lower_ascii(c));
+            ++p;
+        }
+        u32 len = (u32)(p - start);
+        // doc id = number of newlines strictly before `start`
+        long lo = 0, hi = n_nl;
+        while (lo < hi) {
+            long mid = (lo + hi) >> 1;
+            if (nl_pos[mid] < start) lo = mid + 1; else hi = mid;
+        }
+        u64 doc = (u64)lo + doc_base;
+        u64 tok_key = h ? h : 1ULL;
+        u64 sk = splitmix64(h ^ (doc * 0x9E3779B97F4A7C15ULL));
+        if (!sk) sk = 1;
+        u64 slot;
+        if (table_insert_u64(seen_keys, seen_mask, sk, &slot)) {
+            table_add_u64(cnt_keys, cnt_vals, cnt_mask, tok_key, 1ULL);
+            if (table_insert_u64(dict_keys, dict_mask, tok_key, &slot))
+                dict_vals[slot] = ((pos_base + (u64)start) << 8)
+                                  | (u64)min(len, 255u);
+        }
+    }
+}
+
+// ---------------------------------------------------------------- table ops
+__global__ void table_merge_kernel(const u64* __restrict__ in_keys,
+                                   const long* __restrict__ in_vals,
+                                   long n, u64* __restrict__ keys,
+                                   u64* __restrict__ vals, u64 mask) {
+    long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        u64 k = in_keys[i];
+        if (k) table_add_u64(keys, vals, mask, k, (u64)in_vals[i]);
+    }
+}
+
+__global__ void table_put_kernel(const u64* __restrict__ in_keys,
+                                 const u64* __restrict__ in_vals, long n,
+                                 u64* __restrict__ keys,
+                                 u64* __restrict__ vals, u64 mask) {
+    long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        u64 k = in_keys[i];
+        if (!k) continue;
+        u64 slot;
+        if (table_insert_u64(keys, mask, k, &slot))
+            vals[slot] = in_vals[i];
+    }
+}
+
+__global__ void table_extract_kernel(const u64* __restrict__ keys,
+                                     const u64* __restrict__ vals,
+                                     long cap, u64* __restrict__ out_k,
+                                     long* __restrict__ out_v,
+                                     u64* __restrict__ cursor) {
+    long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < cap;
+         i += stride) {
+        u64 k = keys[i];
+        if (k) {
+            u64 j = atomicAdd(cursor, 1ULL);
+            out_k[j] = k;
+            out_v[j] = (long)vals[i];
+        }
+    }
+}
+
+__global__ void table_lookup_kernel(const u64* __restrict__ keys,
+                                    const u64* __restrict__ vals, u64 mask,
+                                    const u64* __restrict__ query, long n,
+                                    u64* __restrict__ out) {
+    long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        u64 k = query[i];
+        u64 slot = k & mask;
+        u64 r = 0;
+        while (true) {
+            u64 cur = keys[slot];
+            if (cur == k) { r = vals[slot]; break; }
+            if (cur == 0) break;
+            slot = (slot + 1) & mask;
+        }
+        out[i] = r;
+    }
+}
+
+// ------------------------------------------------------------------ epilogue
+// idf = log(1 + total/df): the reference's cross_right scalar apply (K9 +
+// benchmarks/tf-idf-dampr.py:18-20), fused elementwise.
+__global__ void idf_kernel(const long* __restrict__ df, long n,
+                           double total, double* __restrict__ out) {
+    long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        out[i] = log(1.0 + total / (double)df[i]);
+}
+
+// Gather token strings: out[offsets[i] .. offsets[i]+len_i) = bytes of
+// token i (packed = pos<<8 | len).
+__global__ void gather_tokens_kernel(const u8* __restrict__ text,
+                                     const u64* __restrict__ packed, long n,
+                                     const long* __restrict__ offsets,
+                                     u8* __restrict__ out) {
+    long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        u64 pk = packed[i];
+        u32 pos = (u32)(pk >> 8);
+        u32 len = (u32)(pk & 0xFF);
+        long o = offsets[i];
+        for (u32 j = 0; j < len; ++j)
+            out[o + j] = lower_ascii(text[pos + j]);
+    }
+}
+
+// ==========================================================================
+// Host wrappers
+// ==========================================================================
+
+namespace {
+
+inline hipStream_t cur_stream() {
+    return at::hip::getCurrentHIPStream().stream();
+}
+
+inline int grid_for(long work, int block = BLOCK, int cap = 4096) {
+    long g = (work + block - 1) / block;
+    return (int)std::min<long>(std::max<long>(g, 1), cap);
+}
+
+void check_u8(const torch::Tensor& t) {
+    TORCH_CHECK(t.is_cuda() && t.dtype() == torch::kUInt8 &&
+                t.is_contiguous(), "expected contiguous u8 device tensor");
+}
+
+constexpr int SCAN_ITERS = 8;   // 32 KiB per block
+
+}  // namespace
+
+// Returns per-block mark counts; python computes exclusive offsets.
+torch::Tensor mark_counts(torch::Tensor text, long mode) {
+    check_u8(text);
+    long n = text.numel();
+    long span = (long)SCAN_ITERS * TILE;
+    long nblocks = (n + span - 1) / span;
+    auto counts = torch::empty({std::max(nblocks, 1L)},
+        torch::TensorOptions().dtype(torch::kUInt32).device(text.device()));
+    if (n == 0) { counts.zero_(); return counts; }
+    hipLaunchKernelGGL(count_marks_kernel, dim3((u32)nblocks), dim3(BLOCK),
+                       0, cur_stream(), text.data_ptr<u8>(), n, (int)mode,
+                       SCAN_ITERS, (u32*)counts.data_ptr());
+    return counts;
+}
+
+void mark_positions(torch::Tensor text, long mode,
+                    torch::Tensor block_offsets, torch::Tensor out) {
+    check_u8(text);
+    long n = text.numel();
+    if (n == 0 || out.numel() == 0) return;
+    long span = (long)SCAN_ITERS * TILE;
+    long nblocks = (n + span - 1) / span;
+    hipLaunchKernelGGL(write_marks_kernel, dim3((u32)nblocks), dim3(BLOCK),
+                       0, cur_stream(), text.data_ptr<u8>(), n, (int)mode,
+                       SCAN_ITERS, (u32*)block_offsets.data_ptr(),
+                       (u32*)out.data_ptr());
+}
+
+void tfidf_count(torch::Tensor text, torch::Tensor nl_pos,
+                 torch::Tensor tok_start, torch::Tensor seen_keys,
+                 torch::Tensor cnt_keys, torch::Tensor cnt_vals,
+                 torch::Tensor dict_keys, torch::Tensor dict_vals,
+                 long pos_base, long doc_base) {
+    check_u8(text);
+    long n_tok = tok_start.numel();
+    if (n_tok == 0) return;
+    hipLaunchKernelGGL(tfidf_count_kernel,
+        dim3(grid_for(n_tok)), dim3(BLOCK), 0, cur_stream(),
+        text.data_ptr<u8>(), text.numel(),
+        (const u32*)nl_pos.data_ptr(), nl_pos.numel(),
+        (const u32*)tok_start.data_ptr(), n_tok,
+        (u64*)seen_keys.data_ptr(), (u64)(seen_keys.numel() - 1),
+        (u64*)cnt_keys.data_ptr(), (u64*)cnt_vals.data_ptr(),
+        (u64)(cnt_keys.numel() - 1),
+        (u64*)dict_keys.data_ptr(), (u64*)dict_vals.data_ptr(),
+        (u64)(dict_keys.numel() - 1), (u64)pos_base, (u64)doc_base);
+}
+
+void table_merge(torch::Tensor in_keys, torch::Tensor in_vals,
+                 torch::Tensor keys, torch::Tensor vals) {
+    long n = in_keys.numel();
+    if (n == 0) return;
+    hipLaunchKernelGGL(table_merge_kernel, dim3(grid_for(n)), dim3(BLOCK),
+        0, cur_stream(), (const u64*)in_keys.data_ptr(),
+        in_vals.data_ptr<long>(), n, (u64*)keys.data_ptr(),
+        (u64*)vals.data_ptr(), (u64)(keys.numel() - 1));
+}
+
+void table_put(torch::Tensor in_keys, torch::Tensor in_vals,
+               torch::Tensor keys, torch::Tensor vals) {
+    long n = in_keys.numel();
+    if (n == 0) return;
+    hipLaunchKernelGGL(table_put_kernel, dim3(grid_for(n)), dim3(BLOCK),
+        0, cur_stream(), (const u64*)in_keys.data_ptr(),
+        (const u64*)in_vals.data_ptr(), n, (u64*)keys.data_ptr(),
+        (u64*)vals.data_ptr(), (u64)(keys.numel() - 1));
+}
+
+std::vector<torch::Tensor> table_extract(torch::Tensor keys,
+                                         torch::Tensor vals, long n_out) {
+    auto dev = keys.device();
+    auto out_k = torch::empty({n_out},
+        torch::TensorOptions().dtype(torch::kInt64).device(dev));
+    auto out_v = torch::empty({n_out},
+        torch::TensorOptions().dtype(torch::kInt64).device(dev));
+    auto cursor = torch::zeros({1},
+        torch::TensorOptions().dtype(torch::kInt64).device(dev));
+    long cap = keys.numel();
+    if (cap > 0 && n_out > 0)
+        hipLaunchKernelGGL(table_extract_kernel, dim3(grid_for(cap)),
+            dim3(BLOCK), 0, cur_stream(), (const u64*)keys.data_ptr(),
+            (const u64*)vals.data_ptr(), cap, (u64*)out_k.data_ptr(),
+            out_v.data_ptr<long>(), (u64*)cursor.data_ptr());
+    return {out_k, out_v, cursor};
+}
+
+torch::Tensor table_lookup(torch::Tensor keys, torch::Tensor vals,
+                           torch::Tensor query) {
+    long n = query.numel();
+    auto out = torch::zeros({n},
+        torch::TensorOptions().dtype(torch::kInt64).device(query.device()));
+    if (n > 0)
+        hipLaunchKernelGGL(table_lookup_kernel, dim3(grid_for(n)),
+            dim3(BLOCK), 0, cur_stream(), (const u64*)keys.data_ptr(),
+            (const u64*)vals.data_ptr(), (u64)(keys.numel() - 1),
+            (const u64*)query.data_ptr(), n, (u64*)out.data_ptr());
+    return out;
+}
+
+torch::Tensor idf(torch::Tensor df, double total) {
+    long n = df.numel();
+    auto out = torch::empty({n},
+        torch::TensorOptions().dtype(torch::kFloat64).device(df.device()));
+    if (n > 0)
+        hipLaunchKernelGGL(idf_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
+            cur_stream(), df.data_ptr<long>(), n, total,
+            out.data_ptr<double>());
+    return out;
+}
+
+torch::Tensor gather_tokens(torch::Tensor text, torch::Tensor packed,
+                            torch::Tensor offsets, long total_bytes) {
+    check_u8(text);
+    long n = packed.numel();
+    auto out = torch::empty({std::max(total_bytes, 1L)},
+        torch::TensorOptions().dtype(torch::kUInt8).device(text.device()));
+    if (n > 0)
+        hipLaunchKernelGGL(gather_tokens_kernel, dim3(grid_for(n)),
+            dim3(BLOCK), 0, cur_stream(), text.data_ptr<u8>(),
+            (const u64*)packed.data_ptr(), n, offsets.data_ptr<long>(),
+            out.data_ptr<u8>());
+    return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("mark_counts", &mark_counts,
+          "per-block counts of marks (mode 0=newline, 1=token start)");
+    m.def("mark_positions", &mark_positions,
+          "write ascending mark positions given exclusive block offsets");
+    m.def("tfidf_count", &tfidf_count,
+          "tokenize+hash+per-doc-dedupe+df-count in one pass");
+    m.def("table_merge", &table_merge, "add (k,v) pairs into a hash table");
+    m.def("table_put", &table_put,
+          "insert (k,v) pairs if absent (first writer wins)");
+    m.def("table_extract", &table_extract,
+          "compact non-empty table slots to (keys, vals, count)");
+    m.def("table_lookup", &table_lookup, "probe table for query keys");
+    m.def("idf", &idf, "idf = log(1 + total/df)");
+    m.def("gather_tokens", &gather_tokens,
+          "materialize token byte strings from dict entries");
+}

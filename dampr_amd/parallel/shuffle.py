@@ -1,0 +1,153 @@
+"""Multi-GPU partition exchange: RCCL all-to-all over xGMI.
+
+Replaces the reference's filesystem "shuffle" (the parent-side transpose of
+{partition -> [files from all workers]}, reference: runner.py:322-335) with
+a direct all-to-all of (key, value) pair columns — on MI355X xGMI is 7
+point-to-point links per GPU, so a pairwise all-to-all keeps every link
+busy carrying only its pair's partitions.
+
+Backend notes: with the "nccl" backend (RCCL on ROCm) this uses
+``all_to_all_single``.  The "gloo" backend (CPU tests, world_size>1 on this
+no-GPU CI) has no all-to-all, so the same exchange is emulated with
+all_gather — identical results, used only in tests.
+"""
+import torch
+import torch.distributed as dist
+
+
+def owner_of(keys, world):
+    """Partition owner per key; torch.remainder keeps it in [0, world)."""
+    return torch.remainder(keys, world)
+
+
+def exchange_pairs(keys, vals, group=None):
+    """All-to-all: route each (key, val) to ``key % world``; returns the
+    (keys, vals) this rank owns.  Tensors are 1-D int64 on the
+    communication device."""
+    world = dist.get_world_size(group)
+    if world == 1:
+        return keys, vals
+    backend = dist.get_backend(group)
+    owner = owner_of(keys, world)
+    order = torch.argsort(owner)
+    keys, vals, owner = keys[order], vals[order], owner[order]
+    send_counts = torch.bincount(owner, minlength=world)
+
+    if backend == "nccl":
+        recv_counts = torch.empty_like(send_counts)
+        dist.all_to_all_single(recv_counts, send_counts, group=group)
+        in_splits = send_counts.tolist()
+        out_splits = recv_counts.tolist()
+        rk = keys.new_empty(sum(out_splits))
+        rv = vals.new_empty(sum(out_splits))
+        dist.all_to_all_single(rk, keys, out_splits, in_splits, group=group)
+        dist.all_to_all_single(rv, vals, out_splits, in_splits, group=group)
+        return rk, rv
+
+    # gloo fallback (CPU tests): gather everything, keep our partition.
+    rank = dist.get_rank(group)
+    sizes = [torch.zeros(1, dtype=torch.int64) for _ in range(world)]
+    dist.all_gather(sizes, torch.tensor([keys.numel()]), group=group)
+    maxn = max(int(s.item()) for s in sizes)
+    pad_k = torch.zeros(maxn, dtype=keys.dtype)
+    pad_v = torch.zeros(maxn, dtype=vals.dtype)
+    pad_k[:keys.numel()] = keys
+    pad_v[:vals.numel()] = vals
+    all_k = [torch.zeros(maxn, dtype=keys.dtype) for _ in range(world)]
+    all_v = [torch.zeros(maxn, dtype=vals.dtype) for _ in range(world)]
+    dist.all_gather(all_k, pad_k, group=group)
+    dist.all_gather(all_v, pad_v, group=group)
+    outs_k, outs_v = [], []
+    for r in range(world):
+        n = int(sizes[r].item())
+        k = all_k[r][:n]
+        v = all_v[r][:n]
+        mine = owner_of(k, world) == rank
+        outs_k.append(k[mine])
+        outs_v.append(v[mine])
+    return torch.cat(outs_k), torch.cat(outs_v)
+
+
+def reorder_blob(blob, lens, order):
+    """Reorder variable-length byte slices (offsets from cumsum(lens)) into
+    ``order`` — all torch ops, stays on device."""
+    if blob.numel() == 0:
+        return blob
+    offsets = torch.cumsum(lens, 0) - lens
+    sel_starts = offsets[order]
+    sel_lens = lens[order]
+    new_offsets = torch.cumsum(sel_lens, 0) - sel_lens
+    total = int(sel_lens.sum().item())
+    pos = (torch.repeat_interleave(sel_starts - new_offsets, sel_lens)
+           + torch.arange(total, device=blob.device))
+    return blob[pos]
+
+
+def exchange_keyed_payload(keys, vals, blob, lens, group=None):
+    """All-to-all of (key, val, var-len payload) triples routed to
+    ``key % world``.  Returns (keys, vals, blob, lens) owned by this rank.
+
+    NCCL/RCCL path: three all_to_all_single calls (counts, columns, bytes).
+    Gloo path (CPU tests only): all_gather emulation.
+    """
+    world = dist.get_world_size(group)
+    if world == 1:
+        return keys, vals, blob, lens
+    owner = owner_of(keys, world)
+    order = torch.argsort(owner)
+    s_owner = owner[order]
+    s_keys, s_vals, s_lens = keys[order], vals[order], lens[order]
+    s_blob = reorder_blob(blob, lens, order)
+    send_counts = torch.bincount(s_owner, minlength=world)
+    byte_counts = torch.zeros(world, dtype=torch.int64,
+                              device=keys.device)
+    byte_counts.index_add_(0, s_owner, s_lens)
+
+    if dist.get_backend(group) == "nccl":
+        recv_counts = torch.empty_like(send_counts)
+        dist.all_to_all_single(recv_counts, send_counts, group=group)
+        recv_bytes = torch.empty_like(byte_counts)
+        dist.all_to_all_single(recv_bytes, byte_counts, group=group)
+        in_sp = send_counts.tolist()
+        out_sp = recv_counts.tolist()
+        in_bp = byte_counts.tolist()
+        out_bp = recv_bytes.tolist()
+        rk = s_keys.new_empty(sum(out_sp))
+        rv = s_vals.new_empty(sum(out_sp))
+        rl = s_lens.new_empty(sum(out_sp))
+        rb = s_blob.new_empty(sum(out_bp))
+        dist.all_to_all_single(rk, s_keys, out_sp, in_sp, group=group)
+        dist.all_to_all_single(rv, s_vals, out_sp, in_sp, group=group)
+        dist.all_to_all_single(rl, s_lens, out_sp, in_sp, group=group)
+        dist.all_to_all_single(rb, s_blob, out_bp, in_bp, group=group)
+        return rk, rv, rb, rl
+
+    # gloo emulation
+    rank = dist.get_rank(group)
+    obj = [None]
+    gathered = [None] * world
+    dist.all_gather_object(
+        gathered,
+        (keys.cpu(), vals.cpu(), blob.cpu(), lens.cpu()), group=group)
+    outs = ([], [], [], [])
+    for r in range(world):
+        k, v, b, l = gathered[r]
+        own = owner_of(k, world) == rank
+        outs[0].append(k[own])
+        outs[1].append(v[own])
+        outs[3].append(l[own])
+        offs = torch.cumsum(l, 0) - l
+        sel = torch.cat([torch.arange(int(o), int(o) + int(n))
+                         for o, n in zip(offs[own], l[own])]) \
+            if int(own.sum()) else torch.empty(0, dtype=torch.int64)
+        outs[2].append(b[sel])
+    return (torch.cat(outs[0]), torch.cat(outs[1]),
+            torch.cat(outs[2]), torch.cat(outs[3]))
+
+
+def all_reduce_scalar(x, group=None, device=None):
+    """Sum an int across ranks (doc totals, C2 role)."""
+    t = torch.tensor([x], dtype=torch.int64,
+                     device=device if device is not None else "cpu")
+    dist.all_reduce(t, group=group)
+    return int(t.item())

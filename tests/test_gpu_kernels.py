@@ -1,0 +1,122 @@
+"""GPU numerics tests: every HIP kernel vs an exact CPU oracle."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from dampr_amd.ops import native
+    EXT = native.require()
+    DEV = torch.device("cuda:0")
+else:
+    EXT = None
+    DEV = None
+
+
+def _positions(text_t, mode):
+    from dampr_amd.gpu.tfidf import TfidfEngine
+    eng = TfidfEngine(DEV, vocab_capacity=1 << 16)
+    return eng.positions(text_t, mode)
+
+
+@pytest.fixture(scope="module")
+def rng():
+    return np.random.default_rng(42)
+
+
+def test_newline_positions(rng):
+    raw = rng.integers(32, 127, size=1 << 20, dtype=np.uint8)
+    raw[rng.integers(0, raw.size, 5000)] = ord("\n")
+    t = torch.from_numpy(raw).to(DEV)
+    pos, total = _positions(t, 0)
+    want = np.flatnonzero(raw == ord("\n"))
+    assert total == len(want)
+    np.testing.assert_array_equal(pos.cpu().numpy(), want)
+
+
+def test_token_start_positions(rng):
+    raw = rng.integers(32, 127, size=1 << 18, dtype=np.uint8)
+    t = torch.from_numpy(raw).to(DEV)
+    pos, total = _positions(t, 1)
+
+    def is_word(c):
+        c = chr(c).lower()
+        return c.isalnum() and ord(c) < 128 or c == "_"
+    want = [i for i in range(len(raw))
+            if is_word(raw[i]) and (i == 0 or not is_word(raw[i - 1]))]
+    assert total == len(want)
+    np.testing.assert_array_equal(pos.cpu().numpy(), np.array(want))
+
+
+def test_hash_table_count_vs_counter(rng):
+    from collections import Counter
+    keys_np = rng.integers(1, 5000, size=200_000).astype(np.int64)
+    vals_np = np.ones(keys_np.size, dtype=np.int64)
+    cap = 1 << 14
+    tk = torch.zeros(cap, dtype=torch.int64, device=DEV)
+    tv = torch.zeros(cap, dtype=torch.int64, device=DEV)
+    EXT.table_merge(torch.from_numpy(keys_np).to(DEV),
+                    torch.from_numpy(vals_np).to(DEV), tk, tv)
+    n_out = int((tk != 0).sum().item())
+    out_k, out_v, _ = EXT.table_extract(tk, tv, n_out)
+    got = dict(zip(out_k.cpu().numpy().tolist(),
+                   out_v.cpu().numpy().tolist()))
+    want = Counter(keys_np.tolist())
+    assert got == dict(want)
+
+
+def test_table_lookup_and_put(rng):
+    cap = 1 << 12
+    tk = torch.zeros(cap, dtype=torch.int64, device=DEV)
+    tv = torch.zeros(cap, dtype=torch.int64, device=DEV)
+    keys = torch.arange(1, 1001, dtype=torch.int64, device=DEV)
+    vals = keys * 7
+    EXT.table_put(keys, vals, tk, tv)
+    # second put with different vals must NOT overwrite
+    EXT.table_put(keys, vals * 0 + 1, tk, tv)
+    got = EXT.table_lookup(tk, tv, keys)
+    assert torch.equal(got, vals)
+    missing = EXT.table_lookup(tk, tv, keys + 5000)
+    assert int(missing.sum().item()) == 0
+
+
+def test_idf_formula():
+    import math
+    df = torch.tensor([1, 2, 10, 100], dtype=torch.int64, device=DEV)
+    out = EXT.idf(df, 100.0).cpu().numpy()
+    want = [math.log(1 + 100.0 / d) for d in [1, 2, 10, 100]]
+    np.testing.assert_allclose(out, want, rtol=1e-12)
+
+
+def test_tfidf_vs_oracle():
+    from dampr_amd.gpu.corpus import synth_corpus, oracle_df
+    from dampr_amd.gpu.tfidf import run_tfidf
+    text = synth_corpus(1 << 20, vocab=20_000, seed=3)
+    got = run_tfidf(text, device=DEV)
+    want = oracle_df(text)
+    assert len(got) == len(want)
+    for tok, df in want.items():
+        assert got[tok][0] == df
+
+
+def test_tfidf_multi_chunk_equivalence():
+    from dampr_amd.gpu.corpus import synth_corpus
+    from dampr_amd.gpu.tfidf import run_tfidf
+    text = synth_corpus(1 << 20, vocab=10_000, seed=5)
+    one = run_tfidf(text, device=DEV)
+    many = run_tfidf(text, device=DEV, chunk_bytes=100_001)
+    assert one == many
+
+
+def test_tfidf_handles_irregular_text():
+    from dampr_amd.gpu.corpus import oracle_df
+    from dampr_amd.gpu.tfidf import run_tfidf
+    text = ("Hello, WORLD!  hello_world 123 foo-bar\n"
+            "\n"
+            "  tabs\tand  spaces   \n"
+            "last line no newline").encode()
+    arr = np.frombuffer(text, dtype=np.uint8).copy()
+    got = run_tfidf(arr, device=DEV)
+    want = oracle_df(arr)
+    assert {t: v[0] for t, v in got.items()} == want
