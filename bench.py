@@ -78,8 +78,7 @@ def main():
             total_docs = eng.n_docs
             src_text = text
         idf = eng.idf(df, total_docs)
-        blob, lens = eng.token_strings(keys, src_text)
-        eng.sink_tsv(args.sink_dir, rank, blob, lens, df, idf)
+        eng.sink_tsv_device(args.sink_dir, rank, keys, df, idf, src_text)
 
     def barrier_sync():
         if dist_mode:

@@ -141,6 +141,27 @@ class TfidfEngine(object):
         blob = self.ext.gather_tokens(text, packed, offsets, total)
         return blob.cpu().numpy(), lens.cpu().numpy()
 
+    def sink_tsv_device(self, path, part_id, keys, df, idf, src_text):
+        """Device-formatted TSV sink: token\\tdf\\tidf rows are laid out by
+        the tsv_format kernel; the host does one write()."""
+        os.makedirs(path, exist_ok=True)
+        packed = self.ext.table_lookup(self.dict_keys, self.dict_vals, keys)
+        lens = (packed & 0xFF).to(torch.int64)
+        tok_off = torch.cumsum(lens, 0) - lens
+        sizes = self.ext.tsv_sizes(lens, df, idf)
+        row_off = torch.cumsum(sizes, 0) - sizes
+        if keys.numel() == 0:
+            total_tok = 0
+            total = 0
+        else:
+            total_tok = int((tok_off[-1] + lens[-1]).item())
+            total = int((row_off[-1] + sizes[-1]).item())
+        blob = self.ext.gather_tokens(src_text, packed, tok_off, total_tok)
+        out = self.ext.tsv_format(blob, tok_off, lens, df, idf, row_off,
+                                  total)
+        with open(os.path.join(path, "part-{}".format(part_id)), "wb") as fh:
+            fh.write(out.cpu().numpy().tobytes())
+
     def sink_tsv(self, path, part_id, tokens_blob, lens, df, idf):
         """Write the classic `token\\tdf\\tidf` part file."""
         os.makedirs(path, exist_ok=True)
@@ -186,7 +207,7 @@ def run_tfidf(text_np, device="cuda:0", sink_path=None, chunk_bytes=None,
     idf = eng.idf(df, eng.n_docs)
     blob, lens = eng.token_strings(keys, text)
     if sink_path:
-        eng.sink_tsv(sink_path, 0, blob, lens, df, idf)
+        eng.sink_tsv_device(sink_path, 0, keys, df, idf, text)
     out = {}
     pos = 0
     b = blob.tobytes()

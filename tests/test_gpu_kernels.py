@@ -109,6 +109,28 @@ def test_tfidf_multi_chunk_equivalence():
     assert one == many
 
 
+def test_tsv_sink_device(tmp_path):
+    import math
+    from dampr_amd.gpu.corpus import synth_corpus, oracle_df
+    from dampr_amd.gpu.tfidf import run_tfidf
+    text = synth_corpus(1 << 18, vocab=2000, seed=11)
+    path = str(tmp_path / "idfs")
+    got = run_tfidf(text, device=DEV, sink_path=path)
+    want = oracle_df(text)
+    n_docs = int((text == ord("\n")).sum())
+    rows = {}
+    with open(path + "/part-0") as fh:
+        for line in fh:
+            tok, df_s, idf_s = line.rstrip("\n").split("\t")
+            rows[tok] = (int(df_s), float(idf_s))
+    assert set(rows) == set(want)
+    for tok, df in want.items():
+        assert rows[tok][0] == df
+        assert abs(rows[tok][1]
+                   - math.log(1 + n_docs / float(df))) < 1e-8
+    assert got.keys() == want.keys()
+
+
 def test_tfidf_handles_irregular_text():
     from dampr_amd.gpu.corpus import oracle_df
     from dampr_amd.gpu.tfidf import run_tfidf
