@@ -42,6 +42,8 @@ class TfidfEngine(object):
         self.dict_keys = torch.zeros(self.cap, **opts)
         self.dict_vals = torch.zeros(self.cap, **opts)
         self.seen = None
+        self._fb_seen = None
+        self._err = None
         self.n_docs = 0
 
     def reset(self):
@@ -66,18 +68,49 @@ class TfidfEngine(object):
     def count_chunk(self, text, pos_base=0):
         """text: u8 device tensor of newline-delimited ASCII.  ``pos_base``
         is the chunk's absolute byte offset in the job's corpus (dict
-        entries store absolute positions so multi-chunk gathers work)."""
+        entries store absolute positions so multi-chunk gathers work).
+
+        Fast path: the wave-per-doc kernel (LDS staging + LDS-set dedupe).
+        Documents whose distinct-token count overflows both the LDS set and
+        the small global fallback seen-table trip an error flag and the
+        chunk reruns on the fully general token-centric kernel."""
         assert text.numel() < (1 << 31), "chunk must be < 2 GiB"
         nl, n_nl = self.positions(text, MODE_NEWLINE)
-        ts, n_tok = self.positions(text, MODE_TOKEN_START)
         # docs = newlines (+1 unterminated tail line)
         n = text.numel()
         tail = 0
         if n and int(text[-1].item()) != ord("\n"):
             tail = 1
+        n_docs = n_nl + tail
         doc_base = self.n_docs
-        self.n_docs += n_nl + tail
+        self.n_docs += n_docs
 
+        if self._fb_seen is None:
+            self._fb_seen = torch.zeros(1 << 22, dtype=torch.int64,
+                                        device=self.device)
+            self._err = torch.zeros(1, dtype=torch.int32,
+                                    device=self.device)
+        else:
+            self._fb_seen.zero_()
+            self._err.zero_()
+        # Snapshot the tables (~20 us for 64 MB at HBM rate) so an overflow
+        # retry can roll back cleanly.
+        snap = (self.cnt_keys.clone(), self.cnt_vals.clone(),
+                self.dict_keys.clone(), self.dict_vals.clone())
+        self.ext.tfidf_count_docs(text, nl, n_docs, self.cnt_keys,
+                                  self.cnt_vals, self.dict_keys,
+                                  self.dict_vals, pos_base,
+                                  self._fb_seen, self._err)
+        if int(self._err.item()):
+            # Rerun this chunk on the general path with a full-size
+            # (doc,token) seen table.
+            (self.cnt_keys, self.cnt_vals,
+             self.dict_keys, self.dict_vals) = snap
+            self._count_chunk_general(text, nl, pos_base, doc_base)
+        return n_docs
+
+    def _count_chunk_general(self, text, nl, pos_base, doc_base):
+        ts, n_tok = self.positions(text, MODE_TOKEN_START)
         seen_cap = _pow2_at_least(2 * max(n_tok, 1))
         if self.seen is None or self.seen.numel() < seen_cap:
             self.seen = torch.zeros(seen_cap, dtype=torch.int64,
@@ -87,7 +120,6 @@ class TfidfEngine(object):
         self.ext.tfidf_count(text, nl, ts, self.seen, self.cnt_keys,
                              self.cnt_vals, self.dict_keys, self.dict_vals,
                              pos_base, doc_base)
-        return n_tok
 
     # -- table io ------------------------------------------------------------
 

@@ -177,6 +177,129 @@ __global__ void tfidf_count_kernel(
     }
 }
 
+// ------------------------------------------------------- doc-centric count
+// Fast path: one wave per document.  The line is staged into LDS with
+// coalesced loads, token starts are found in-register, and the per-doc
+// `set()` dedupe is a tiny per-wave LDS hash set — no global seen table,
+// no per-token binary search, one pass over the text.  Documents whose
+// distinct-token count overflows the LDS set fall back to a global
+// (doc,hash)-keyed seen table; if THAT overflows its probe bound the
+// kernel sets an error flag and the host reruns the chunk on the fully
+// general token-centric kernel above.
+#define DOC_WAVES 4                  // waves per block
+#define STAGE_B 2048                 // staged line segment bytes
+#define SEG_OVERLAP 272              // > max dict token length (255)
+#define DOC_SET 256                  // per-wave dedupe set slots (pow2)
+#define FB_PROBE_CAP 512
+
+__device__ __forceinline__ int lds_set_insert(u64* set, u64 h) {
+    // 1 = fresh, 0 = dup, -1 = set full (h definitely absent: full scan)
+    u32 slot = (u32)(h & (DOC_SET - 1));
+    for (int probe = 0; probe < DOC_SET; ++probe) {
+        u64 prev = atomicCAS(&set[slot], 0ULL, h);
+        if (prev == 0ULL) return 1;
+        if (prev == h) return 0;
+        slot = (slot + 1) & (DOC_SET - 1);
+    }
+    return -1;
+}
+
+__global__ void __launch_bounds__(DOC_WAVES * WAVE)
+tfidf_docs_kernel(const u8* __restrict__ text, long n,
+                  const u32* __restrict__ nl_pos, long n_nl, long n_docs,
+                  u64* __restrict__ cnt_keys, u64* __restrict__ cnt_vals,
+                  u64 cnt_mask,
+                  u64* __restrict__ dict_keys, u64* __restrict__ dict_vals,
+                  u64 dict_mask, u64 pos_base,
+                  u64* __restrict__ fb_seen, u64 fb_mask,
+                  u32* __restrict__ err_flag) {
+    __shared__ u8 stage[DOC_WAVES][STAGE_B];
+    __shared__ u64 dset[DOC_WAVES][DOC_SET];
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const long gwave = (long)blockIdx.x * DOC_WAVES + wid;
+    const long nwaves = (long)gridDim.x * DOC_WAVES;
+    u8* st = stage[wid];
+    u64* set = dset[wid];
+
+    for (long d = gwave; d < n_docs; d += nwaves) {
+        const long ls = d ? (long)nl_pos[d - 1] + 1 : 0;
+        const long le = (d < n_nl) ? (long)nl_pos[d] : n;
+        for (int s = lane; s < DOC_SET; s += WAVE) set[s] = 0;
+
+        for (long seg = ls; seg < le; seg += STAGE_B - SEG_OVERLAP) {
+            const long seg_end = min(seg + (long)STAGE_B, le);
+            const int seg_len = (int)(seg_end - seg);
+            for (int i = lane; i < seg_len; i += WAVE)
+                st[i] = text[seg + i];
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_wave_barrier();
+
+            const int start_lim = (seg_end == le)
+                ? seg_len : (STAGE_B - SEG_OVERLAP);
+            for (int base = 0; base < start_lim; base += WAVE) {
+                const int p = base + lane;
+                bool is_start = false;
+                if (p < start_lim) {
+                    const u8 c = st[p];
+                    bool pw;
+                    if (p > 0) pw = is_word(st[p - 1]);
+                    else pw = (seg > 0) ? is_word(text[seg - 1]) : false;
+                    is_start = is_word(c) && !pw;
+                }
+                if (!is_start) continue;
+                int q = p;
+                u64 h = FNV_OFFSET;
+                while (q < seg_len) {
+                    const u8 c = st[q];
+                    if (!is_word(c)) break;
+                    h = fnv1a64_step(h, lower_ascii(c));
+                    ++q;
+                }
+                u32 tl = (u32)(q - p);
+                if (q == seg_len && seg_end < le) {
+                    // pathological >2 KiB token: finish from global
+                    long g = seg + q;
+                    while (g < le && is_word(text[g])) {
+                        h = fnv1a64_step(h, lower_ascii(text[g]));
+                        ++g; ++tl;
+                    }
+                }
+                int fresh = lds_set_insert(set, h ? h : 1ULL);
+                if (fresh < 0) {
+                    // overflow: global (doc,hash) seen fallback
+                    u64 sk = splitmix64(h ^ ((u64)d
+                                             * 0x9E3779B97F4A7C15ULL));
+                    if (!sk) sk = 1;
+                    u64 slot = sk & fb_mask;
+                    fresh = 0;
+                    int probe = 0;
+                    while (true) {
+                        u64 prev = atomicCAS(&fb_seen[slot], 0ULL, sk);
+                        if (prev == 0ULL) { fresh = 1; break; }
+                        if (prev == sk) break;
+                        slot = (slot + 1) & fb_mask;
+                        if (++probe > FB_PROBE_CAP) {
+                            atomicOr(err_flag, 1u);
+                            break;
+                        }
+                    }
+                }
+                if (fresh == 1) {
+                    const u64 key = h ? h : 1ULL;
+                    table_add_u64(cnt_keys, cnt_vals, cnt_mask, key, 1ULL);
+                    u64 slot;
+                    if (table_insert_u64(dict_keys, dict_mask, key, &slot))
+                        dict_vals[slot] =
+                            ((pos_base + (u64)(seg + p)) << 8)
+                            | (u64)min(tl, 255u);
+                }
+            }
+            __builtin_amdgcn_wave_barrier();
+        }
+    }
+}
+
 // ---------------------------------------------------------------- table ops
 __global__ void table_merge_kernel(const u64* __restrict__ in_keys,
                                    const long* __restrict__ in_vals,
@@ -408,6 +531,31 @@ void tfidf_count(torch::Tensor text, torch::Tensor nl_pos,
         (u64)(dict_keys.numel() - 1), (u64)pos_base, (u64)doc_base);
 }
 
+// Returns 0 on success, 1 when the fallback seen table overflowed (host
+// must rerun the chunk via the token-centric kernel).
+long tfidf_count_docs(torch::Tensor text, torch::Tensor nl_pos,
+                      long n_docs, torch::Tensor cnt_keys,
+                      torch::Tensor cnt_vals, torch::Tensor dict_keys,
+                      torch::Tensor dict_vals, long pos_base,
+                      torch::Tensor fb_seen, torch::Tensor err_flag) {
+    check_u8(text);
+    if (n_docs == 0) return 0;
+    long waves_needed = n_docs;
+    long blocks = std::min<long>((waves_needed + DOC_WAVES - 1) / DOC_WAVES,
+                                 8192);
+    hipLaunchKernelGGL(tfidf_docs_kernel, dim3((u32)blocks),
+        dim3(DOC_WAVES * WAVE), 0, cur_stream(),
+        text.data_ptr<u8>(), text.numel(),
+        (const u32*)nl_pos.data_ptr(), nl_pos.numel(), n_docs,
+        (u64*)cnt_keys.data_ptr(), (u64*)cnt_vals.data_ptr(),
+        (u64)(cnt_keys.numel() - 1),
+        (u64*)dict_keys.data_ptr(), (u64*)dict_vals.data_ptr(),
+        (u64)(dict_keys.numel() - 1), (u64)pos_base,
+        (u64*)fb_seen.data_ptr(), (u64)(fb_seen.numel() - 1),
+        (u32*)err_flag.data_ptr());
+    return 0;
+}
+
 void table_merge(torch::Tensor in_keys, torch::Tensor in_vals,
                  torch::Tensor keys, torch::Tensor vals) {
     long n = in_keys.numel();
@@ -521,6 +669,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "write ascending mark positions given exclusive block offsets");
     m.def("tfidf_count", &tfidf_count,
           "tokenize+hash+per-doc-dedupe+df-count in one pass");
+    m.def("tfidf_count_docs", &tfidf_count_docs,
+          "wave-per-doc tokenize+dedupe+count (fast path)");
     m.def("table_merge", &table_merge, "add (k,v) pairs into a hash table");
     m.def("table_put", &table_put,
           "insert (k,v) pairs if absent (first writer wins)");

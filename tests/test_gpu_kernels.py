@@ -109,6 +109,22 @@ def test_tfidf_multi_chunk_equivalence():
     assert one == many
 
 
+def test_tfidf_long_docs_overflow_paths():
+    # One doc with thousands of distinct tokens (> LDS dedupe set, > one
+    # 2 KiB staging segment) plus repeats: exercises segment streaming and
+    # the global fallback seen table.
+    from dampr_amd.gpu.corpus import oracle_df
+    from dampr_amd.gpu.tfidf import run_tfidf
+    words = ["w{}x".format(i) for i in range(3000)]
+    line1 = " ".join(words + words)           # dups within the doc
+    line2 = " ".join(words[:50])
+    text = (line1 + "\n" + line2 + "\n").encode()
+    arr = np.frombuffer(text, dtype=np.uint8).copy()
+    got = run_tfidf(arr, device=DEV)
+    want = oracle_df(arr)
+    assert {t: v[0] for t, v in got.items()} == want
+
+
 def test_tsv_sink_device(tmp_path):
     import math
     from dampr_amd.gpu.corpus import synth_corpus, oracle_df
