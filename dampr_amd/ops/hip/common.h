@@ -38,15 +38,22 @@ __device__ __forceinline__ bool is_word(u8 c) {
 
 // Open-addressing u64 table helpers: key 0 is EMPTY (real keys are mixed
 // 64-bit hashes; a key hashing to 0 is remapped to 1 by callers).
+// Probing reads the key with a plain (cached) load first and only CASes on
+// observed-empty: hot keys then cost one read instead of a serializing RMW
+// on a contended line.
 __device__ __forceinline__ void table_add_u64(u64* __restrict__ keys,
                                               u64* __restrict__ vals,
                                               u64 mask, u64 key, u64 inc) {
     u64 slot = key & mask;
     while (true) {
-        u64 prev = atomicCAS(&keys[slot], 0ULL, key);
-        if (prev == 0ULL || prev == key) {
-            atomicAdd(&vals[slot], inc);
-            return;
+        u64 cur = keys[slot];
+        if (cur == key) { atomicAdd(&vals[slot], inc); return; }
+        if (cur == 0ULL) {
+            u64 prev = atomicCAS(&keys[slot], 0ULL, key);
+            if (prev == 0ULL || prev == key) {
+                atomicAdd(&vals[slot], inc);
+                return;
+            }
         }
         slot = (slot + 1) & mask;
     }
@@ -58,9 +65,13 @@ __device__ __forceinline__ bool table_insert_u64(u64* __restrict__ keys,
                                                  u64* slot_out) {
     u64 slot = key & mask;
     while (true) {
-        u64 prev = atomicCAS(&keys[slot], 0ULL, key);
-        if (prev == 0ULL) { *slot_out = slot; return true; }
-        if (prev == key)  { *slot_out = slot; return false; }
+        u64 cur = keys[slot];
+        if (cur == key) { *slot_out = slot; return false; }
+        if (cur == 0ULL) {
+            u64 prev = atomicCAS(&keys[slot], 0ULL, key);
+            if (prev == 0ULL) { *slot_out = slot; return true; }
+            if (prev == key)  { *slot_out = slot; return false; }
+        }
         slot = (slot + 1) & mask;
     }
 }

@@ -191,6 +191,30 @@ __global__ void tfidf_count_kernel(
 #define SEG_OVERLAP 272              // > max dict token length (255)
 #define DOC_SET 256                  // per-wave dedupe set slots (pow2)
 #define FB_PROBE_CAP 512
+#define CCACHE 1024                  // block-level LDS count cache slots
+
+// Two-level counting: Zipf-hot keys would serialize ~50M same-address L2
+// atomics; the block-level LDS cache turns that into one global add per
+// (block, hot key).  Cache misses (cold keys) go straight to the global
+// table — cold keys have no contention.
+__device__ __forceinline__ void block_count_add(
+        u64* __restrict__ cck, u32* __restrict__ ccv, u64 key,
+        u64* __restrict__ gk, u64* __restrict__ gv, u64 gmask) {
+    u32 slot = (u32)(key & (CCACHE - 1));
+    for (int probe = 0; probe < 16; ++probe) {
+        u64 cur = cck[slot];
+        if (cur == key) { atomicAdd(&ccv[slot], 1u); return; }
+        if (cur == 0ULL) {
+            u64 prev = atomicCAS(&cck[slot], 0ULL, key);
+            if (prev == 0ULL || prev == key) {
+                atomicAdd(&ccv[slot], 1u);
+                return;
+            }
+        }
+        slot = (slot + 1) & (CCACHE - 1);
+    }
+    table_add_u64(gk, gv, gmask, key, 1ULL);
+}
 
 __device__ __forceinline__ int lds_set_insert(u64* set, u64 h) {
     // 1 = fresh, 0 = dup, -1 = set full (h definitely absent: full scan)
@@ -215,12 +239,20 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                   u32* __restrict__ err_flag) {
     __shared__ u8 stage[DOC_WAVES][STAGE_B];
     __shared__ u64 dset[DOC_WAVES][DOC_SET];
+    __shared__ u64 cck[CCACHE];
+    __shared__ u32 ccv[CCACHE];
     const int wid = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
     const long gwave = (long)blockIdx.x * DOC_WAVES + wid;
     const long nwaves = (long)gridDim.x * DOC_WAVES;
     u8* st = stage[wid];
     u64* set = dset[wid];
+
+    for (int i = threadIdx.x; i < CCACHE; i += blockDim.x) {
+        cck[i] = 0;
+        ccv[i] = 0;
+    }
+    __syncthreads();
 
     for (long d = gwave; d < n_docs; d += nwaves) {
         const long ls = d ? (long)nl_pos[d - 1] + 1 : 0;
@@ -287,7 +319,8 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                 }
                 if (fresh == 1) {
                     const u64 key = h ? h : 1ULL;
-                    table_add_u64(cnt_keys, cnt_vals, cnt_mask, key, 1ULL);
+                    block_count_add(cck, ccv, key, cnt_keys, cnt_vals,
+                                    cnt_mask);
                     u64 slot;
                     if (table_insert_u64(dict_keys, dict_mask, key, &slot))
                         dict_vals[slot] =
@@ -298,6 +331,13 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
             __builtin_amdgcn_wave_barrier();
         }
     }
+
+    // flush the block's count cache
+    __syncthreads();
+    for (int i = threadIdx.x; i < CCACHE; i += blockDim.x)
+        if (cck[i])
+            table_add_u64(cnt_keys, cnt_vals, cnt_mask, cck[i],
+                          (u64)ccv[i]);
 }
 
 // ---------------------------------------------------------------- table ops
