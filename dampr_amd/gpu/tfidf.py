@@ -44,6 +44,7 @@ class TfidfEngine(object):
         self.seen = None
         self._fb_seen = None
         self._err = None
+        self._ablate = 0        # perf-ablation bits; 0 = full work
         self.n_docs = 0
 
     def reset(self):
@@ -88,7 +89,7 @@ class TfidfEngine(object):
         if self._fb_seen is None:
             self._fb_seen = torch.zeros(1 << 22, dtype=torch.int64,
                                         device=self.device)
-            self._err = torch.zeros(1, dtype=torch.int32,
+            self._err = torch.zeros(2, dtype=torch.int32,
                                     device=self.device)
         else:
             self._fb_seen.zero_()
@@ -100,8 +101,8 @@ class TfidfEngine(object):
         self.ext.tfidf_count_docs(text, nl, n_docs, self.cnt_keys,
                                   self.cnt_vals, self.dict_keys,
                                   self.dict_vals, pos_base,
-                                  self._fb_seen, self._err)
-        if int(self._err.item()):
+                                  self._fb_seen, self._err, self._ablate)
+        if int(self._err[0].item()):
             # Rerun this chunk on the general path with a full-size
             # (doc,token) seen table.
             (self.cnt_keys, self.cnt_vals,

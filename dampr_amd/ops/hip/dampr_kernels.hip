@@ -236,7 +236,7 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                   u64* __restrict__ dict_keys, u64* __restrict__ dict_vals,
                   u64 dict_mask, u64 pos_base,
                   u64* __restrict__ fb_seen, u64 fb_mask,
-                  u32* __restrict__ err_flag) {
+                  u32* __restrict__ err_flag, u32 ablate) {
     __shared__ u8 stage[DOC_WAVES][STAGE_B];
     __shared__ u64 dset[DOC_WAVES][DOC_SET];
     __shared__ u64 cck[CCACHE];
@@ -297,6 +297,10 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                         ++g; ++tl;
                     }
                 }
+                if (ablate & 1) {           // ablation: hash only
+                    if (h == 0xdeadbeefdeadbeefULL) err_flag[1] = 1;
+                    continue;
+                }
                 int fresh = lds_set_insert(set, h ? h : 1ULL);
                 if (fresh < 0) {
                     // overflow: global (doc,hash) seen fallback
@@ -317,12 +321,14 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                         }
                     }
                 }
-                if (fresh == 1) {
+                if (fresh == 1 && !(ablate & 2)) {
                     const u64 key = h ? h : 1ULL;
                     block_count_add(cck, ccv, key, cnt_keys, cnt_vals,
                                     cnt_mask);
                     u64 slot;
-                    if (table_insert_u64(dict_keys, dict_mask, key, &slot))
+                    if (!(ablate & 4)
+                        && table_insert_u64(dict_keys, dict_mask, key,
+                                            &slot))
                         dict_vals[slot] =
                             ((pos_base + (u64)(seg + p)) << 8)
                             | (u64)min(tl, 255u);
@@ -577,7 +583,8 @@ long tfidf_count_docs(torch::Tensor text, torch::Tensor nl_pos,
                       long n_docs, torch::Tensor cnt_keys,
                       torch::Tensor cnt_vals, torch::Tensor dict_keys,
                       torch::Tensor dict_vals, long pos_base,
-                      torch::Tensor fb_seen, torch::Tensor err_flag) {
+                      torch::Tensor fb_seen, torch::Tensor err_flag,
+                      long ablate) {
     check_u8(text);
     if (n_docs == 0) return 0;
     long waves_needed = n_docs;
@@ -592,7 +599,7 @@ long tfidf_count_docs(torch::Tensor text, torch::Tensor nl_pos,
         (u64*)dict_keys.data_ptr(), (u64*)dict_vals.data_ptr(),
         (u64)(dict_keys.numel() - 1), (u64)pos_base,
         (u64*)fb_seen.data_ptr(), (u64)(fb_seen.numel() - 1),
-        (u32*)err_flag.data_ptr());
+        (u32*)err_flag.data_ptr(), (u32)ablate);
     return 0;
 }
 
