@@ -1,5 +1,6 @@
 """Perf ablation of the doc-centric count kernel (run on a GPU box)."""
-import time
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from dampr_amd.gpu.corpus import synth_corpus
 from dampr_amd.gpu.tfidf import TfidfEngine
