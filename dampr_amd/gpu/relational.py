@@ -1,0 +1,140 @@
+"""Device relational ops over (u64-key, payload) columns: sort, group-by
+reduce, join, top-k.
+
+These are the GPU engine's building blocks for the DSL's ``group_by``,
+``sort_by``, ``join`` and ``topk`` when keys/values lower to device columns
+(SURVEY.md §2.4 K2/K3/K5/K7/K8/K11).  Keys are 64-bit (keyhash.key_hash64
+for strings); payloads are row indices into host or device value arenas.
+
+torch is the tensor layer (allocations, cumsum/nonzero metadata glue); all
+per-record hot work is in the hand-written HIP kernels.
+"""
+import torch
+
+from ..ops import native
+
+OP_SUM, OP_MIN, OP_MAX = 0, 1, 2
+
+_I64_MIN = -(1 << 63)
+_I64_MAX = (1 << 63) - 1
+
+
+def _pow2_at_least(n):
+    return 1 << max(4, int(n - 1).bit_length())
+
+
+def radix_sort_pairs(keys, payload=None):
+    """Stable LSD radix sort of int64-as-u64 keys; returns (keys, payload)
+    sorted in unsigned key order.  Skips passes whose digit is constant."""
+    ext = native.require()
+    n = keys.numel()
+    if payload is None:
+        payload = torch.arange(n, dtype=torch.int32, device=keys.device)
+    if n <= 1:
+        return keys.clone(), payload.clone()
+    RS_SPAN = 4096
+    nblocks = (n + RS_SPAN - 1) // RS_SPAN
+    a_k, a_p = keys.contiguous(), payload.contiguous()
+    b_k = torch.empty_like(a_k)
+    b_p = torch.empty_like(a_p)
+    for byte in range(8):
+        shift = byte * 8
+        hist = ext.rs_hist(a_k, shift, nblocks).to(torch.int64)
+        per_bin = hist.view(256, nblocks).sum(1)
+        if int((per_bin != 0).sum().item()) <= 1:
+            continue                      # constant digit: skip pass
+        scanned = torch.cumsum(hist, 0) - hist
+        ext.rs_scatter(a_k, a_p, scanned, shift, nblocks, b_k, b_p)
+        a_k, b_k = b_k, a_k
+        a_p, b_p = b_p, a_p
+    return a_k, a_p
+
+
+def segment_ids(sorted_keys):
+    """(seg_ids i64, unique_keys) for a key-sorted column."""
+    n = sorted_keys.numel()
+    if n == 0:
+        return (torch.zeros(0, dtype=torch.int64,
+                            device=sorted_keys.device), sorted_keys)
+    flags = torch.ones(n, dtype=torch.int64, device=sorted_keys.device)
+    flags[1:] = (sorted_keys[1:] != sorted_keys[:-1]).to(torch.int64)
+    seg = torch.cumsum(flags, 0) - 1
+    uniq = sorted_keys[flags.bool()]
+    return seg, uniq
+
+
+def group_reduce_sorted(sorted_keys, vals, op=OP_SUM):
+    """Segmented reduce over a key-sorted column; returns (unique_keys,
+    aggregates)."""
+    ext = native.require()
+    seg, uniq = segment_ids(sorted_keys)
+    n_seg = uniq.numel()
+    if vals.dtype == torch.float64:
+        init = {OP_SUM: 0.0, OP_MIN: float("inf"),
+                OP_MAX: float("-inf")}[op]
+        out = torch.full((max(n_seg, 1),), init, dtype=torch.float64,
+                         device=vals.device)
+    else:
+        init = {OP_SUM: 0, OP_MIN: _I64_MAX, OP_MAX: _I64_MIN}[op]
+        out = torch.full((max(n_seg, 1),), init, dtype=torch.int64,
+                         device=vals.device)
+    ext.seg_reduce(seg, vals, out, op)
+    return uniq, out[:n_seg]
+
+
+def group_sum(keys, vals):
+    """Unsorted group-by-sum: radix sort then segmented reduce (the
+    high-cardinality path; for low-cardinality the hash-combine table in
+    tfidf.py is the faster route)."""
+    sk, sp = radix_sort_pairs(keys)
+    sv = vals[sp.to(torch.int64)]
+    return group_reduce_sorted(sk, sv)
+
+
+def hash_join(keys_l, keys_r, how="inner"):
+    """Equi-join on u64 keys.  Returns (l_idx, r_idx) int64 row-index pairs;
+    for "left"/"outer", missing matches carry index -1.
+
+    how: "inner" | "left" | "outer"
+    """
+    assert how in ("inner", "left", "outer")
+    ext = native.require()
+    dev = keys_l.device
+    nr = keys_r.numel()
+    cap = _pow2_at_least(max(2 * nr, 16))
+    t_keys = torch.zeros(cap, dtype=torch.int64, device=dev)
+    t_head = torch.full((cap,), -1, dtype=torch.int64, device=dev)
+    nxt = torch.empty(max(nr, 1), dtype=torch.int64, device=dev)
+    ext.hj_build(keys_r, t_keys, t_head, nxt)
+    left_outer = 1 if how in ("left", "outer") else 0
+    counts = ext.hj_count(keys_l, t_keys, t_head, nxt, left_outer)
+    offsets = torch.cumsum(counts, 0) - counts
+    total = int(counts.sum().item())
+    out_l, out_r, matched = ext.hj_emit(
+        keys_l, t_keys, t_head, nxt, offsets, total, left_outer,
+        1 if how == "outer" else 0, nr)
+    out_l, out_r = out_l[:total], out_r[:total]
+    if how == "outer" and nr:
+        un = torch.nonzero(matched[:nr] == 0).flatten()
+        if un.numel():
+            out_l = torch.cat([out_l, torch.full_like(un, -1)])
+            out_r = torch.cat([out_r, un])
+    return out_l, out_r
+
+
+def topk_by(values_u64, k, largest=True):
+    """Top-k row indices ordered by a u64 sort column (K11 via the radix
+    sort; the column is a key-encoded ordering, e.g. flipped-sign floats)."""
+    keys = values_u64 if not largest else ~values_u64
+    sk, sp = radix_sort_pairs(keys)
+    return sp[:k].to(torch.int64)
+
+
+def encode_f64_sortable(x):
+    """Map float64 to u64 preserving order (IEEE trick): sortable keys for
+    sort_by/topk on floats."""
+    b = x.view(torch.int64)
+    neg = b < 0
+    sign_bit = -(1 << 63)                 # 0x8000000000000000 as int64
+    out = torch.where(neg, ~b, b ^ sign_bit)
+    return out

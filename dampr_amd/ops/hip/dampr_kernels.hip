@@ -707,7 +707,34 @@ torch::Tensor tsv_format(torch::Tensor blob, torch::Tensor tok_off,
     return out;
 }
 
+// Implemented in dampr_sort.hip
+torch::Tensor rs_hist(torch::Tensor keys, long shift, long nblocks);
+void rs_scatter(torch::Tensor keys, torch::Tensor payload,
+                torch::Tensor scanned, long shift, long nblocks,
+                torch::Tensor out_k, torch::Tensor out_p);
+void seg_reduce(torch::Tensor seg, torch::Tensor vals, torch::Tensor out,
+                long op);
+void hj_build(torch::Tensor keys_r, torch::Tensor t_keys,
+              torch::Tensor t_head, torch::Tensor next);
+torch::Tensor hj_count(torch::Tensor keys_l, torch::Tensor t_keys,
+                       torch::Tensor t_head, torch::Tensor next,
+                       long left_outer);
+std::vector<torch::Tensor> hj_emit(torch::Tensor keys_l,
+                                   torch::Tensor t_keys,
+                                   torch::Tensor t_head,
+                                   torch::Tensor next,
+                                   torch::Tensor offsets, long total,
+                                   long left_outer, long track_matched,
+                                   long nr);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("rs_hist", &rs_hist, "radix pass histogram (bin-major)");
+    m.def("rs_scatter", &rs_scatter, "stable radix scatter pass");
+    m.def("seg_reduce", &seg_reduce,
+          "segmented reduce over sorted runs (op 0=sum,1=min,2=max)");
+    m.def("hj_build", &hj_build, "hash-join build (chained)");
+    m.def("hj_count", &hj_count, "hash-join probe match counts");
+    m.def("hj_emit", &hj_emit, "hash-join emit (l,r) row-index pairs");
     m.def("tsv_sizes", &tsv_sizes, "per-row TSV byte sizes");
     m.def("tsv_format", &tsv_format, "format token/df/idf rows as TSV");
     m.def("mark_counts", &mark_counts,

@@ -15,7 +15,8 @@ _EXT = None
 _ERR = None
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
-SOURCES = [os.path.join(_HERE, "hip", "dampr_kernels.hip")]
+SOURCES = [os.path.join(_HERE, "hip", "dampr_kernels.hip"),
+           os.path.join(_HERE, "hip", "dampr_sort.hip")]
 BUILD_DIR = os.path.join(_HERE, "_build")
 
 
