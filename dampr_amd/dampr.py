@@ -20,6 +20,7 @@ import operator
 import random
 import sys
 
+from . import funcs
 from .base import (BlockMapper, Map, MapAllJoin, MapCrossJoin, Mapper,
                    ComposedMapper, ComposedStreamable, KeyedInnerJoin,
                    KeyedLeftJoin, KeyedOuterJoin, KeyedCrossJoin,
@@ -61,11 +62,18 @@ class PBase(object):
         self.pmer = pmer
 
     def run(self, name=None, **kwargs):
-        """Evaluate the composed graph; returns a ValueEmitter."""
+        """Evaluate the composed graph; returns a ValueEmitter.
+
+        The engine is picked per graph: columnar inputs
+        (``Dampr.columns``) run on the device engine (gpu/engine.py);
+        everything else on the multi-process host engine.  Override with
+        ``runner=`` (a runner class)."""
         if name is None:
             name = "dampr/{}".format(random.random())
-        ds = self.pmer.runner(name, self.pmer.graph, **kwargs) \
-                      .run([self.source])
+        runner_cls = kwargs.pop("runner", None) or \
+            _pick_runner(self.pmer.graph, self.pmer.runner)
+        ds = runner_cls(name, self.pmer.graph, **kwargs) \
+            .run([self.source])
         return ValueEmitter(ds[0])
 
     def read(self, k=None, **kwargs):
@@ -75,6 +83,17 @@ class PBase(object):
 
 def _identity(k, v):
     yield k, v
+
+
+def _pick_runner(graph, explicit):
+    """Engine selection: explicit wins; columnar inputs get the device
+    engine; everything else the multi-process host engine."""
+    if explicit is not None:
+        return explicit
+    from .gpu.engine import ColumnSource, GpuRunner
+    if any(isinstance(d, ColumnSource) for d in graph.inputs.values()):
+        return GpuRunner
+    return MTRunner
 
 
 class PMap(PBase):
@@ -103,6 +122,10 @@ class PMap(PBase):
         result among downstream consumers."""
         if self.agg or force:
             aggs = self.agg[:] if self.agg else [Map(_identity)]
+            if not self.agg:
+                # pure materialization point: free on the device engine
+                options = dict(options or {})
+                options.setdefault("device_map", ("identity",))
             name = "Stage {}: %s" % " -> ".join(str(a) for a in aggs)
             source, pmer = self.pmer._add_mapper(
                 [self.source], fuse(aggs), combiner=combiner, name=name,
@@ -217,16 +240,29 @@ class PMap(PBase):
         pm = self._add_map(_group_by).checkpoint()
         return PReduce(pm.source, pm.pmer)
 
-    def a_group_by(self, key, vf=lambda x: x):
+    def a_group_by(self, key=None, vf=None):
         """Group by key(value) for *associative* reductions: enables the
         map-side partial reduce (combiner), which is the fast path — on GPU
-        it lowers to the device hash-combine kernel (K6)."""
+        it lowers to the device hash-combine kernel (K6).
+
+        When ``key``/``vf`` are recognized named funcs (dampr_amd.funcs,
+        e.g. the defaults) AND this PMap has no pending opaque maps, the
+        stage carries a device spec so the columnar engine (gpu/engine.py)
+        runs it on the CDNA4 kernels."""
+        key = funcs.identity if key is None else key
+        vf = funcs.identity if vf is None else vf
+
         def _a_group_by(_k, value):
             yield key(value), vf(value)
+        kname = funcs.column_func_name(key)
+        vname = funcs.column_func_name(vf)
+        dev_map = None
+        if kname and vname and not self.agg:
+            dev_map = ("kv", kname, vname)
         # No checkpoint: ARReduce attaches the combiner to this stage.
-        return ARReduce(self._add_map(_a_group_by))
+        return ARReduce(self._add_map(_a_group_by), device_map=dev_map)
 
-    def fold_by(self, key, binop, value=lambda x: x, **options):
+    def fold_by(self, key, binop, value=None, **options):
         """a_group_by(key, value).reduce(binop)."""
         return self.a_group_by(key, value).reduce(binop, **options)
 
@@ -236,9 +272,9 @@ class PMap(PBase):
             yield key(value), value
         return self._add_map(_sort_by).checkpoint(options=options)
 
-    def count(self, key=lambda x: x, **options):
+    def count(self, key=None, **options):
         """Count occurrences by key(value)."""
-        return self.a_group_by(key, lambda v: 1) \
+        return self.a_group_by(key, funcs.one) \
                    .reduce(operator.add, **options)
 
     def mean(self, key=lambda x: 1, value=lambda x: x, **options):
@@ -370,10 +406,14 @@ class PMap(PBase):
     def sink(self, path):
         """Write each value (assumed str) as a line into part files under
         ``path``; exempt from cleanup."""
+        # No pending opaque maps -> the columnar engine can format the
+        # value column on device (tfidf.py's tsv kernels are the model).
+        options = {"device_sink": "values"} if not self.agg else None
         aggs = self.agg[:] if self.agg else [Map(_identity)]
         name = "Stage {}: %s" % " -> ".join(str(a) for a in aggs)
         source, pmer = self.pmer._add_sink(
-            [self.source], fuse(aggs), path=path, name=name, options=None)
+            [self.source], fuse(aggs), path=path, name=name,
+            options=options)
         return PMap(source, pmer)
 
     def sink_tsv(self, path):
@@ -389,13 +429,11 @@ class PMap(PBase):
 class ARReduce(object):
     """Associative reductions (map-side partial reduce enabled)."""
 
-    def __init__(self, pmap):
+    def __init__(self, pmap, device_map=None):
         self.pmap = pmap
+        self._device_map = device_map
 
-    def reduce(self, binop, reduce_buffer=1000, **options):
-        """Reduce each group with an associative binop.  ``reduce_buffer``
-        caps the map-side combine dictionary (keys held in memory before a
-        spill) — unlike the reference, it is honored."""
+    def _run(self, binop, dev_reduce, reduce_buffer, options):
         def _reduce(key, vs):
             acc = next(vs)
             for v in vs:
@@ -403,26 +441,43 @@ class ARReduce(object):
             return acc
 
         options.update({"binop": binop, "reduce_buffer": reduce_buffer})
+        if self._device_map is not None and dev_reduce is not None:
+            options["device_map"] = self._device_map
         pm = self.pmap.checkpoint(
             True, combiner=PartialReduceCombiner(Reduce(_reduce)),
             options=options)
-        return PReduce(pm.source, pm.pmer).reduce(_reduce)
+        red_opts = None
+        if self._device_map is not None and dev_reduce is not None:
+            red_opts = {"device_reduce": dev_reduce}
+        return PReduce(pm.source, pm.pmer).reduce(_reduce,
+                                                  options=red_opts)
+
+    def reduce(self, binop, reduce_buffer=1000, **options):
+        """Reduce each group with an associative binop.  ``reduce_buffer``
+        caps the map-side combine dictionary (keys held in memory before a
+        spill) — unlike the reference, it is honored.
+        Recognized binops (operator.add, min, max — dampr_amd.funcs) lower
+        to the device segmented-reduce kernel on the columnar engine."""
+        name = funcs.binop_name(binop)
+        return self._run(binop, (name,) if name else None, reduce_buffer,
+                         options)
 
     def first(self, **options):
         """First value seen per key."""
-        return self.reduce(lambda x, _y: x, **options)
+        return self._run(lambda x, _y: x, ("first",), 1000, options)
 
     def sum(self, **options):
         """Sum of values per key."""
-        return self.reduce(lambda x, y: x + y, **options)
+        return self._run(operator.add, ("sum",), 1000, options)
 
 
 class PReduce(PBase):
     """General grouped reductions."""
 
-    def reduce(self, f):
+    def reduce(self, f, options=None):
         """Reduce each group with ``f(key, iter values) -> value``."""
-        source, pmer = self.pmer._add_reducer([self.source], KeyedReduce(f))
+        source, pmer = self.pmer._add_reducer([self.source], KeyedReduce(f),
+                                              options=options)
         return PMap(source, pmer)
 
     def unique(self, key=lambda x: x):
@@ -466,12 +521,22 @@ class PJoin(PBase):
 
     def reduce(self, aggregate, many=False):
         """Inner join; aggregate(left_iter, right_iter).  ``many=True``
-        flattens an iterable result into separate records."""
+        flattens an iterable result into separate records.
+
+        ``funcs.pair_sum/pair_product/pair_left/pair_right`` with
+        ``many=True`` lower to the device hash-join kernel (K8) on the
+        columnar engine."""
         def _reduce(_k, left, right):
             return aggregate(left, right)
 
+        options = None
+        pname = funcs.join_pair_name(aggregate)
+        if pname and many:
+            options = {"device_reduce": ("join", "inner"),
+                       "device_join_pair": pname}
         source, pmer = self.pmer._add_reducer(
-            [self.source, self.right], KeyedInnerJoin(_reduce, many))
+            [self.source, self.right], KeyedInnerJoin(_reduce, many),
+            options=options)
         return PMap(source, pmer)
 
     def left_reduce(self, aggregate):
@@ -507,13 +572,25 @@ class Dampr(object):
 
     def __init__(self, graph=None, runner=None):
         self.graph = graph if graph is not None else Graph()
-        self.runner = runner if runner is not None else MTRunner
+        self.runner = runner          # None = auto (see _pick_runner)
 
     @classmethod
     def memory(cls, items, partitions=50):
         """Pipeline over an in-memory list."""
         mi = MemoryInput(list(enumerate(items)), partitions)
         source, ng = Graph().add_input(mi)
+        return PMap(source, cls(ng))
+
+    @classmethod
+    def columns(cls, vals, keys=None):
+        """Pipeline over typed columns (numpy arrays / torch tensors of
+        int64 or float64): records are (key_i, val_i), keys default to the
+        row index.  Runs on the device engine — built-in ops (count, sum,
+        fold_by with recognized binops, joins, first, ...) execute as
+        gfx950 kernels over HBM-resident columns."""
+        from .gpu.engine import ColumnSource
+        src_obj = ColumnSource.from_data(vals, keys)
+        source, ng = Graph().add_input(src_obj)
         return PMap(source, cls(ng))
 
     @classmethod
@@ -561,7 +638,9 @@ class Dampr(object):
             sources.append(pmer.source)
             last = pmer
         name = kwargs.pop("name", "dampr/{}".format(random.random()))
-        ds = last.pmer.runner(name, graph, **kwargs).run(sources)
+        runner_cls = kwargs.pop("runner", None) or \
+            _pick_runner(graph, last.pmer.runner)
+        ds = runner_cls(name, graph, **kwargs).run(sources)
         return [ValueEmitter(d) for d in ds]
 
     def _add_mapper(self, *args, **kwargs):

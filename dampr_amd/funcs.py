@@ -1,0 +1,101 @@
+"""Named, *recognizable* function objects used as DSL defaults.
+
+The reference's API is lambda-everywhere (e.g. ``count(key=lambda x: x)``,
+reference: dampr.py:439-448); opaque lambdas can only run on the host.  The
+MI355X engine lowers a stage to device kernels when the functions it was
+built from are these named objects (or a handful of stdlib ones:
+``operator.add``, builtin ``min``/``max``) — identity-preserving `is`
+checks, never bytecode inspection.  Users who pass their own callables get
+the host path for that stage automatically; passing these (or just using
+the DSL defaults) gets the device path.
+"""
+import operator
+
+
+def identity(x):
+    return x
+
+
+def one(_x):
+    return 1
+
+
+def fst(x):
+    return x[0]
+
+
+def snd(x):
+    return x[1]
+
+
+add = operator.add
+mul = operator.mul
+
+
+# --- recognition tables ----------------------------------------------------
+
+# binop -> segmented-reduce op name understood by the device backend
+ASSOC_BINOPS = {
+    add: "sum",
+    operator.add: "sum",
+    min: "min",
+    max: "max",
+}
+
+# key/value extractors the columnar engine can apply without host trips
+# (fst/snd await pair-column support in the engine)
+COLUMN_FUNCS = {
+    identity: "identity",
+    one: "one",
+}
+
+
+# --- join pair aggregates ---------------------------------------------------
+# PJoin.reduce(aggregate, many=True) aggregates the cartesian product of
+# each key's (left values, right values).  These named per-pair forms let
+# the device engine emit the product straight from the hash-join kernel.
+
+def pair_sum(left, right):
+    return [lv + rv for lv in left for rv in right]
+
+
+def pair_product(left, right):
+    return [lv * rv for lv in left for rv in right]
+
+
+def pair_left(left, right):
+    return [lv for lv in left for _rv in right]
+
+
+def pair_right(left, right):
+    return [rv for _lv in left for rv in right]
+
+
+JOIN_PAIR_FUNCS = {
+    pair_sum: "sum",
+    pair_product: "mul",
+    pair_left: "left",
+    pair_right: "right",
+}
+
+
+def join_pair_name(f):
+    try:
+        return JOIN_PAIR_FUNCS.get(f)
+    except TypeError:
+        return None
+
+
+def binop_name(f):
+    """Device op name for an associative binop, or None."""
+    try:
+        return ASSOC_BINOPS.get(f)
+    except TypeError:
+        return None
+
+
+def column_func_name(f):
+    try:
+        return COLUMN_FUNCS.get(f)
+    except TypeError:
+        return None

@@ -1,0 +1,686 @@
+"""The device dataflow engine: executes DSL plans (runner.Graph) over
+columnar record batches resident in HBM.
+
+Architecture (MI355X-first, no translation of the reference's process
+pools):
+
+* Records are **typed columns** — an i64 key column + an i64/f64 value
+  column — not streams of pickled Python objects.  This is the layout the
+  CDNA4 kernels (ops/hip/*.hip) operate on: radix sort (K2/K3), segmented
+  reduce (K5/K7), hash join (K8), top-k (K11).  Pipelines over arbitrary
+  Python objects belong to the host engine (runner.MTRunner), which is the
+  complete, conformance-tested implementation of the same plans; pipelines
+  over text use the tokenizer/string-dict specialization (gpu/tfidf.py).
+* A stage lowers to device kernels when the DSL built it from *recognized*
+  functions (dampr_amd.funcs); anything else falls back per-stage to the
+  host operators with decode/encode at the boundary (SURVEY.md §7 "hard
+  parts": opaque UDFs run on the host, the shuffle/sort/combine core stays
+  on device).
+* Out-of-core: partitions live in an ``HbmPool`` with a watermark; least
+  recently used runs spill to (pinned) host memory and are paged back per
+  partition at reduce time — the reference's RSS-watermark spill files
+  (dampr/memory.py, dataset.py:190-262) re-expressed at HBM granularity.
+  Only the active partition must be resident, so jobs scale past 288 GB.
+* Multi-GPU: after every partitioning map stage the engine exchanges
+  partitions to their owning rank over RCCL all-to-all (xGMI); partition p
+  is owned by rank ``p % world`` (parallel/shuffle.py).
+"""
+import logging
+
+import torch
+
+from .. import settings
+from ..runner import GMap, GReduce, GSink, RunnerBase
+from .backend import ops_for
+
+log = logging.getLogger("dampr_amd")
+
+
+# --------------------------------------------------------------------------
+# Columnar data
+# --------------------------------------------------------------------------
+
+_VAL_DTYPES = (torch.int64, torch.float64)
+
+
+def _as_column(x, device, dtype=None):
+    if isinstance(x, torch.Tensor):
+        t = x
+    else:
+        import numpy as np
+        t = torch.from_numpy(np.ascontiguousarray(x))
+    if t.dtype not in _VAL_DTYPES:
+        t = t.to(torch.float64 if t.is_floating_point() else torch.int64)
+    if dtype is not None:
+        t = t.to(dtype)
+    return t.to(device)
+
+
+class ColumnSource(object):
+    """A device-columnar input: (keys i64, vals i64|f64)."""
+
+    def __init__(self, keys, vals):
+        assert keys.dtype == torch.int64
+        assert vals.dtype in _VAL_DTYPES
+        assert keys.numel() == vals.numel()
+        self.keys = keys
+        self.vals = vals
+
+    @classmethod
+    def from_data(cls, vals, keys=None, device="cpu"):
+        v = _as_column(vals, device)
+        if keys is None:
+            k = torch.arange(v.numel(), dtype=torch.int64, device=v.device)
+        else:
+            k = _as_column(keys, device, torch.int64)
+        return cls(k, v)
+
+
+class PartStore(dict):
+    """{partition -> [DeviceRun]} plus column metadata:
+
+    * ``keyed``: output of a keyed reducer — records follow the host
+      convention (k, (k, v)) at decode time (base.py KeyedReduce).
+    * ``fkeys``: the key column is an order-preserving i64 encoding of
+      float64 keys (relational.encode_f64_sortable); decoded on read.
+    """
+
+    def __init__(self, keyed=False, fkeys=False):
+        super(PartStore, self).__init__()
+        self.keyed = keyed
+        self.fkeys = fkeys
+
+
+def _decode_f64_sortable(enc):
+    """Inverse of relational.encode_f64_sortable."""
+    sign_bit = -(1 << 63)
+    b = torch.where(enc < 0, enc ^ sign_bit, ~enc)
+    return b.view(torch.float64)
+
+
+def _encode_f64_sortable(x):
+    b = x.view(torch.int64)
+    sign_bit = -(1 << 63)
+    return torch.where(b < 0, ~b, b ^ sign_bit)
+
+
+class DeviceRun(object):
+    """One (keys, vals) run of a partition; spillable to host."""
+
+    __slots__ = ("keys", "vals", "sorted", "_host")
+
+    def __init__(self, keys, vals, sorted=False):
+        self.keys = keys
+        self.vals = vals
+        self.sorted = sorted
+        self._host = None
+
+    @property
+    def nbytes(self):
+        if self.keys is None:
+            return 0
+        return self.keys.numel() * 8 + self.vals.element_size() * \
+            self.vals.numel()
+
+    @property
+    def resident(self):
+        return self.keys is not None
+
+    def spill(self):
+        """Move to (pinned) host memory; frees the HBM."""
+        if self._host is not None or self.keys is None:
+            return
+        pin = self.keys.device.type == "cuda"
+        hk = torch.empty_like(self.keys, device="cpu", pin_memory=pin)
+        hv = torch.empty_like(self.vals, device="cpu", pin_memory=pin)
+        hk.copy_(self.keys)
+        hv.copy_(self.vals)
+        self._host = (hk, hv)
+        self.keys = None
+        self.vals = None
+
+    def load(self, device):
+        if self.keys is None:
+            hk, hv = self._host
+            self.keys = hk.to(device, non_blocking=True)
+            self.vals = hv.to(device, non_blocking=True)
+            self._host = None
+        return self
+
+
+class HbmPool(object):
+    """Watermark governor over device-resident run bytes (the reference's
+    MemoryChecker/MaxMemoryWriter analog, memory.py:72-113, at HBM
+    granularity)."""
+
+    def __init__(self, capacity_bytes):
+        self.capacity = capacity_bytes
+        self.used = 0
+        self._lru = []                 # DeviceRun insertion order
+
+    def admit(self, run):
+        self.used += run.nbytes
+        self._lru.append(run)
+        self.balance()
+
+    def touch(self, run, device):
+        if not run.resident:
+            run.load(device)
+            self.used += run.nbytes
+            self._lru.append(run)
+            self.balance(exclude=run)
+        return run
+
+    def release(self, run):
+        if run.resident:
+            self.used -= run.nbytes
+        if run in self._lru:
+            self._lru.remove(run)
+
+    def balance(self, exclude=None):
+        while self.used > self.capacity and self._lru:
+            victim = None
+            for r in self._lru:
+                if r is not exclude and r.resident:
+                    victim = r
+                    break
+            if victim is None:
+                return
+            self._lru.remove(victim)
+            self.used -= victim.nbytes
+            victim.spill()
+
+
+# --------------------------------------------------------------------------
+# Output dataset (feeds ValueEmitter / downstream host code)
+# --------------------------------------------------------------------------
+
+class ColumnDataset(object):
+    """Dataset-duck-typed view over result columns: read() yields (k, v)
+    Python scalars; columns() hands back the tensors for zero-copy
+    composition.  ``keyed`` reproduces the host reducers' value
+    convention (k, (k, v)); ``fkeys`` decodes float64 keys."""
+
+    def __init__(self, keys, vals, keyed=False, fkeys=False):
+        self.keys_t = keys
+        self.vals_t = vals
+        self.keyed = keyed
+        self.fkeys = fkeys
+
+    def columns(self):
+        return self.keys_t, self.vals_t
+
+    def read(self):
+        kt = self.keys_t
+        if self.fkeys:
+            kt = _decode_f64_sortable(kt)
+        k = kt.cpu().tolist()
+        v = self.vals_t.cpu().tolist()
+        if self.keyed:
+            return iter((kk, (kk, vv)) for kk, vv in zip(k, v))
+        return iter(zip(k, v))
+
+    def grouped_read(self):
+        import itertools
+        for key, group in itertools.groupby(self.read(), key=lambda p: p[0]):
+            yield key, (v for _k, v in group)
+
+    def delete(self):
+        self.keys_t = self.vals_t = None
+
+    def __iter__(self):
+        return self.read()
+
+
+# --------------------------------------------------------------------------
+# The engine
+# --------------------------------------------------------------------------
+
+class GpuRunner(RunnerBase):
+    """Interprets a runner.Graph over device columns.
+
+    Stage dispatch: ``stage.options["device_map"] / ["device_reduce"]``
+    descriptors (attached by the DSL when built from recognized funcs) run
+    on the kernels; untagged stages run the host operators with a
+    decode/encode boundary (records must stay numeric scalars).
+    """
+
+    def __init__(self, name, graph, device=None, n_partitions=None,
+                 hbm_bytes=None):
+        # note: RunnerBase.__init__ builds a /tmp FileSystem we don't use;
+        # keep it for interface parity (sinks reuse its naming).
+        super(GpuRunner, self).__init__(name, graph)
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        self.ops = ops_for(self.device)
+        self.n_partitions = n_partitions or settings.gpu_partitions
+        cap = hbm_bytes or settings.hbm_pool_mb * (1 << 20)
+        self.pool = HbmPool(cap)
+        if torch.distributed.is_available() and \
+                torch.distributed.is_initialized():
+            self.world = torch.distributed.get_world_size()
+            self.rank = torch.distributed.get_rank()
+        else:
+            self.world, self.rank = 1, 0
+
+    # -- plan walk ---------------------------------------------------------
+
+    def run(self, outputs, cleanup=True):
+        data = {}
+        for src, inp in self.graph.inputs.items():
+            data[src] = self._ingest(inp)
+        for stage_id, stage in enumerate(self.graph.stages):
+            log.info("[device] Stage %s/%s: %r", stage_id + 1,
+                     len(self.graph.stages), stage)
+            ins = [data[i] for i in stage.inputs]
+            if isinstance(stage, GMap):
+                out = self.run_map(stage, ins)
+            elif isinstance(stage, GReduce):
+                out = self.run_reduce(stage, ins)
+            elif isinstance(stage, GSink):
+                out = self.run_sink(stage, ins)
+            else:
+                raise TypeError(stage)
+            data[stage.output] = out
+        rets = []
+        for source in outputs:
+            store = data[source]
+            rets.append(self._collect(store))
+        return rets
+
+    # -- ingest / collect --------------------------------------------------
+
+    def _ingest(self, inp):
+        """Input -> partition store.  ColumnSource goes straight to device;
+        host Datasets/Chunkers decode through the numeric encoder."""
+        if isinstance(inp, ColumnSource):
+            keys = inp.keys.to(self.device)
+            vals = inp.vals.to(self.device)
+            if self.world > 1:
+                # each rank keeps an equal slice of the input
+                n = keys.numel()
+                lo = n * self.rank // self.world
+                hi = n * (self.rank + 1) // self.world
+                keys, vals = keys[lo:hi], vals[lo:hi]
+            return self._partition(keys, vals)
+        # host dataset / chunker: stream records, encode
+        records = self._host_records_of_input(inp)
+        return self._encode_records(records)
+
+    @staticmethod
+    def _host_records_of_input(inp):
+        from ..dataset import Chunker, Dataset
+        if isinstance(inp, Dataset):
+            return list(inp.read())
+        if isinstance(inp, Chunker):
+            out = []
+            for c in inp.chunks():
+                out.extend(c.read())
+            return out
+        raise TypeError("GPU engine cannot ingest {!r}".format(inp))
+
+    def _encode_records(self, records):
+        import numpy as np
+        if not records:
+            return self._partition(
+                torch.zeros(0, dtype=torch.int64),
+                torch.zeros(0, dtype=torch.int64))
+        ks = [k for k, _ in records]
+        vs = [v for _, v in records]
+        # host keyed-reducer convention: value = (key, scalar)
+        keyed = all(
+            isinstance(v, tuple) and len(v) == 2 and v[0] == k
+            and isinstance(v[1], (int, float, bool))
+            for (k, _), v in zip(records, vs))
+        if keyed:
+            vs = [v[1] for v in vs]
+        if all(isinstance(v, (int, bool)) for v in vs):
+            vt = torch.from_numpy(np.asarray(vs, dtype=np.int64))
+        elif all(isinstance(v, (int, float, bool)) for v in vs):
+            vt = torch.from_numpy(np.asarray(vs, dtype=np.float64))
+        else:
+            raise TypeError(
+                "device engine requires numeric values; use the host "
+                "engine for object records")
+        fkeys = False
+        if all(isinstance(k, (int, bool)) for k in ks):
+            kt = torch.from_numpy(np.asarray(ks, dtype=np.int64))
+        elif all(isinstance(k, (int, float, bool)) for k in ks):
+            kt = _encode_f64_sortable(
+                torch.from_numpy(np.asarray(ks, dtype=np.float64)))
+            fkeys = True
+        else:
+            raise TypeError(
+                "device engine requires numeric keys; use the host engine "
+                "for object records")
+        return self._partition(kt.to(self.device), vt.to(self.device),
+                               keyed=keyed, fkeys=fkeys)
+
+    def _collect(self, store):
+        """Partition store -> one key-sorted ColumnDataset (the engine's
+        MergeDataset analog: partitions are key-sorted, output is their
+        merge)."""
+        if isinstance(store, ColumnDataset):
+            return store
+        keyed = getattr(store, "keyed", False)
+        fkeys = getattr(store, "fkeys", False)
+        ks, vs = [], []
+        for p in sorted(store):
+            for run in store[p]:
+                self.pool.touch(run, self.device)
+                ks.append(run.keys)
+                vs.append(run.vals)
+                self.pool.release(run)
+        if not ks:
+            z = torch.zeros(0, dtype=torch.int64)
+            return ColumnDataset(z, z.clone(), keyed, fkeys)
+        keys = torch.cat(ks)
+        vals = torch.cat(vs)
+        sk, sp = self.ops.sort_pairs(keys)
+        return ColumnDataset(sk, vals[sp.to(torch.int64)], keyed, fkeys)
+
+    # -- partitioning ------------------------------------------------------
+
+    def _partition(self, keys, vals, already_sorted=False, keyed=False,
+                   fkeys=False):
+        """Split columns into the partition store {p: [DeviceRun]} (K1+K2):
+        partition ids, stable sort by id, slice contiguous segments."""
+        P = self.n_partitions
+        store = PartStore(keyed=keyed, fkeys=fkeys)
+        if keys.numel() == 0:
+            return store
+        pid = self.ops.partition_of(keys, P)
+        order = torch.argsort(pid, stable=True)
+        keys, vals, pid = keys[order], vals[order], pid[order]
+        counts = torch.bincount(pid, minlength=P)
+        if self.world > 1:
+            keys, vals, pid = self._exchange(keys, vals, pid)
+            # received rows arrive grouped by sender, not by partition:
+            # restore pid order for the contiguous slicing below
+            order = torch.argsort(pid, stable=True)
+            keys, vals, pid = keys[order], vals[order], pid[order]
+            counts = torch.bincount(pid, minlength=P)
+        offs = torch.cumsum(counts, 0) - counts
+        counts_l = counts.tolist()
+        offs_l = offs.tolist()
+        for p in range(P):
+            if self.world > 1 and p % self.world != self.rank:
+                continue
+            n = counts_l[p]
+            if not n:
+                continue
+            o = offs_l[p]
+            k = keys[o:o + n].contiguous()
+            v = vals[o:o + n].contiguous()
+            if not already_sorted:
+                sk, sp = self.ops.sort_pairs(k)
+                k, v = sk, v[sp.to(torch.int64)]
+            run = DeviceRun(k, v, sorted=True)
+            store.setdefault(p, []).append(run)
+            self.pool.admit(run)
+        return store
+
+    def _exchange(self, keys, vals, pid):
+        """RCCL all-to-all: route rows to the partition's owning rank
+        (p % world); returns this rank's rows."""
+        from ..parallel.shuffle import exchange_columns
+        return exchange_columns(keys, vals, pid, self.world)
+
+    def _merged_partition(self, stores, p):
+        """All runs of partition p across input stores, merged key-sorted."""
+        ks, vs = [], []
+        for store in stores:
+            for run in store.get(p, []):
+                self.pool.touch(run, self.device)
+                ks.append(run.keys)
+                vs.append(run.vals)
+                self.pool.release(run)
+        if not ks:
+            return None, None
+        if len(ks) == 1:
+            return ks[0], vs[0]
+        keys = torch.cat(ks)
+        vals = torch.cat(vs)
+        sk, sp = self.ops.sort_pairs(keys)
+        return sk, vals[sp.to(torch.int64)]
+
+    def _parts(self, stores):
+        ps = set()
+        for s in stores:
+            ps.update(s.keys())
+        return sorted(ps)
+
+    # -- map stage ---------------------------------------------------------
+
+    def run_map(self, stage, ins):
+        spec = stage.options.get("device_map")
+        if spec is None:
+            return self._host_map(stage, ins)
+        kind = spec[0]
+        if kind == "kv":
+            if any(getattr(s, "keyed", False) for s in ins):
+                # tuple-valued records (keyed convention): column funcs
+                # don't apply — run the stage's Python mapper instead
+                return self._host_map(stage, ins)
+            # emit (keyfn(v), valfn(v)) per record — the group_by/count map
+            _kind, keyf, valf = spec
+            out = None
+            for p in self._parts(ins):
+                keys, vals = self._merged_partition(ins, p)
+                if keys is None:
+                    continue
+                nk = self._apply_colfunc(keyf, keys, vals)
+                nv = self._apply_colfunc(valf, keys, vals)
+                fkeys = nk.dtype == torch.float64
+                if fkeys:
+                    nk = _encode_f64_sortable(nk)
+                out_store = self._partition(nk, nv, fkeys=fkeys)
+                if out is None:
+                    out = out_store
+                else:
+                    for q, runs in out_store.items():
+                        out.setdefault(q, []).extend(runs)
+            return out if out is not None else PartStore()
+        if kind == "identity":
+            return self._merge_stores(ins)
+        raise ValueError("unknown device_map spec {!r}".format(spec))
+
+    def _apply_colfunc(self, name, keys, vals):
+        if name == "identity":
+            return vals
+        if name == "one":
+            return torch.ones_like(keys)
+        if name == "key":
+            return keys
+        raise ValueError("unknown column func {!r}".format(name))
+
+    def _merge_stores(self, stores):
+        if len(stores) == 1:
+            return stores[0]
+        out = PartStore(
+            keyed=any(getattr(s, "keyed", False) for s in stores),
+            fkeys=any(getattr(s, "fkeys", False) for s in stores))
+        for s in stores:
+            for p, runs in s.items():
+                out.setdefault(p, []).extend(runs)
+        return out
+
+    # -- reduce stage ------------------------------------------------------
+
+    def run_reduce(self, stage, ins):
+        spec = stage.options.get("device_reduce")
+        if spec is None or any(getattr(s, "keyed", False) for s in ins):
+            return self._host_reduce(stage, ins)
+        kind = spec[0]
+        in_fkeys = any(getattr(s, "fkeys", False) for s in ins)
+        out = PartStore(keyed=True, fkeys=in_fkeys)
+        if kind in ("sum", "min", "max"):
+            for p in self._parts(ins):
+                keys, vals = self._merged_partition(ins, p)
+                if keys is None:
+                    continue
+                uk, agg = self.ops.seg_reduce_sorted(keys, vals, kind)
+                run = DeviceRun(uk, agg, sorted=True)
+                out.setdefault(p, []).append(run)
+                self.pool.admit(run)
+            return out
+        if kind == "mean":
+            for p in self._parts(ins):
+                keys, vals = self._merged_partition(ins, p)
+                if keys is None:
+                    continue
+                uk, s = self.ops.seg_reduce_sorted(
+                    keys, vals.to(torch.float64), "sum")
+                _uk, c = self.ops.seg_reduce_sorted(
+                    keys, torch.ones_like(keys), "sum")
+                run = DeviceRun(uk, s / c.to(torch.float64), sorted=True)
+                out.setdefault(p, []).append(run)
+                self.pool.admit(run)
+            return out
+        if kind == "first":
+            # first value per key: stable sorts keep insertion order
+            # within equal keys, so the segment head is the first seen
+            for p in self._parts(ins):
+                keys, vals = self._merged_partition(ins, p)
+                if keys is None:
+                    continue
+                _seg, uk = self.ops.segment_ids(keys)
+                flags = torch.ones_like(keys)
+                flags[1:] = (keys[1:] != keys[:-1]).to(torch.int64)
+                first_idx = torch.nonzero(flags.bool()).flatten()
+                run = DeviceRun(uk, vals[first_idx], sorted=True)
+                out.setdefault(p, []).append(run)
+                self.pool.admit(run)
+            return out
+        if kind == "join":
+            how = spec[1]
+            assert len(ins) == 2, "join takes two inputs"
+            return self._device_join(ins[0], ins[1], how, stage)
+        raise ValueError("unknown device_reduce spec {!r}".format(spec))
+
+    def _device_join(self, left, right, how, stage):
+        """Per-partition device hash join (K8); emits matched value pairs.
+        The DSL's aggregate(left_vals, right_vals) runs per pair on the
+        host only when not a recognized columnar op; the default device
+        aggregate emits (key, (lv, rv)) as two columns folded into one
+        f64/i64 value via the stage's pair op, or keeps lv when
+        aggregate is 'left'."""
+        pair_op = stage.options.get("device_join_pair", "pair_host")
+        out = PartStore(
+            keyed=True,
+            fkeys=getattr(left, "fkeys", False)
+            or getattr(right, "fkeys", False))
+        for p in self._parts([left, right]):
+            lk, lv = self._merged_partition([left], p)
+            rk, rv = self._merged_partition([right], p)
+            if lk is None and rk is None:
+                continue
+            if lk is None:
+                lk = torch.zeros(0, dtype=torch.int64, device=self.device)
+                lv = torch.zeros(0, dtype=torch.int64, device=self.device)
+            if rk is None:
+                rk = torch.zeros(0, dtype=torch.int64, device=self.device)
+                rv = torch.zeros(0, dtype=torch.int64, device=self.device)
+            li, ri = self.ops.hash_join(lk, rk, how)
+            keys = torch.where(li >= 0, lk[torch.clamp(li, min=0)],
+                               rk[torch.clamp(ri, min=0)])
+            lvm = torch.where(li >= 0, lv[torch.clamp(li, min=0)],
+                              torch.zeros_like(li))
+            rvm = torch.where(ri >= 0, rv[torch.clamp(ri, min=0)],
+                              torch.zeros_like(ri))
+            valid_l = li >= 0
+            valid_r = ri >= 0
+            merged = self._apply_pair_op(pair_op, lvm, rvm, valid_l,
+                                         valid_r)
+            sk, sp = self.ops.sort_pairs(keys)
+            run = DeviceRun(sk, merged[sp.to(torch.int64)], sorted=True)
+            out.setdefault(p, []).append(run)
+            self.pool.admit(run)
+        return out
+
+    @staticmethod
+    def _apply_pair_op(op, lv, rv, valid_l, valid_r):
+        if op == "sum":
+            return lv + rv
+        if op == "left":
+            return lv
+        if op == "right":
+            return rv
+        if op == "mul":
+            return lv * rv
+        raise ValueError(
+            "join aggregate not columnar ({!r}); use funcs-recognized "
+            "aggregates or the host engine".format(op))
+
+    # -- sink --------------------------------------------------------------
+
+    def run_sink(self, stage, ins):
+        """Sink semantics match the host SinkWriter: one text line per
+        record, value only (keys dropped — reference: dataset.py:277-278).
+        Untagged sinks (pending format maps, e.g. sink_tsv's) run the
+        mapper on host over the decoded records."""
+        import os
+        path = stage.path
+        os.makedirs(path, exist_ok=True)
+        ds = self._collect(self._merge_stores(ins))
+        fname = os.path.join(path, "part-{}".format(self.rank))
+        if stage.options.get("device_sink") == "values":
+            with open(fname, "w") as fh:
+                for _kk, vv in ds.read():
+                    fh.write("{}\n".format(vv))
+        else:
+            from ..dataset import MemoryDataset
+            mem = MemoryDataset(list(ds.read()))
+            with open(fname, "w") as fh:
+                for _kk, vv in stage.mapper.map(mem):
+                    fh.write("{}\n".format(vv))
+        return PartStore()
+
+    # -- host fallback -----------------------------------------------------
+
+    def _decode_store(self, store):
+        records = []
+        keyed = getattr(store, "keyed", False)
+        fkeys = getattr(store, "fkeys", False)
+        for p in sorted(store):
+            keys, vals = self._merged_partition([store], p)
+            if keys is None:
+                continue
+            if fkeys:
+                keys = _decode_f64_sortable(keys)
+            kl = keys.cpu().tolist()
+            vl = vals.cpu().tolist()
+            if keyed:
+                records.extend((k, (k, v)) for k, v in zip(kl, vl))
+            else:
+                records.extend(zip(kl, vl))
+        return records
+
+    def _host_map(self, stage, ins):
+        """Decode -> run the stage's (opaque) mapper on host -> re-encode.
+        The shuffle core stays on device either side of this boundary.
+        Input 0 is the primary dataset; the rest ride as supplemental
+        dataset-lists (cross joins — reference: stagerunner.py:66-74)."""
+        from ..dataset import MemoryDataset
+        primary = MemoryDataset(self._decode_store(ins[0]))
+        supplemental = [[MemoryDataset(self._decode_store(s))]
+                        for s in ins[1:]]
+        out_records = list(stage.mapper.map(primary, *supplemental))
+        return self._encode_records(out_records)
+
+    def _host_reduce(self, stage, ins):
+        out_records = []
+        for p in self._parts(ins):
+            datasets = []
+            for store in ins:
+                keys, vals = self._merged_partition([store], p)
+                if keys is None:
+                    continue
+                datasets.append([ColumnDataset(keys.cpu(), vals.cpu())])
+            if not datasets:
+                continue
+            while len(datasets) < len(ins):
+                datasets.append([])
+            out_records.extend(stage.reducer.reduce(*datasets))
+        return self._encode_records(out_records)

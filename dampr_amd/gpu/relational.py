@@ -34,20 +34,30 @@ def radix_sort_pairs(keys, payload=None):
         return keys.clone(), payload.clone()
     RS_SPAN = 4096
     nblocks = (n + RS_SPAN - 1) // RS_SPAN
-    a_k, a_p = keys.contiguous(), payload.contiguous()
-    b_k = torch.empty_like(a_k)
-    b_p = torch.empty_like(a_p)
+    # The caller's tensors are read-only: the first executed pass scatters
+    # out of them into an owned buffer, and later passes ping-pong between
+    # two owned buffers (using the input as scratch would clobber it).
+    src_k, src_p = keys.contiguous(), payload.contiguous()
+    out_k, out_p = torch.empty_like(src_k), torch.empty_like(src_p)
+    n_done = 0
     for byte in range(8):
         shift = byte * 8
-        hist = ext.rs_hist(a_k, shift, nblocks).to(torch.int64)
+        hist = ext.rs_hist(src_k, shift, nblocks).to(torch.int64)
         per_bin = hist.view(256, nblocks).sum(1)
         if int((per_bin != 0).sum().item()) <= 1:
             continue                      # constant digit: skip pass
         scanned = torch.cumsum(hist, 0) - hist
-        ext.rs_scatter(a_k, a_p, scanned, shift, nblocks, b_k, b_p)
-        a_k, b_k = b_k, a_k
-        a_p, b_p = b_p, a_p
-    return a_k, a_p
+        ext.rs_scatter(src_k, src_p, scanned, shift, nblocks, out_k, out_p)
+        n_done += 1
+        if n_done == 1:
+            src_k, src_p = out_k, out_p
+            out_k, out_p = (torch.empty_like(src_k),
+                            torch.empty_like(src_p))
+        else:
+            src_k, src_p, out_k, out_p = out_k, out_p, src_k, src_p
+    if n_done == 0:
+        return src_k.clone(), src_p.clone()
+    return src_k, src_p
 
 
 def segment_ids(sorted_keys):

@@ -145,6 +145,47 @@ def exchange_keyed_payload(keys, vals, blob, lens, group=None):
             torch.cat(outs[2]), torch.cat(outs[3]))
 
 
+def exchange_columns(keys, vals, pids, world, group=None):
+    """All-to-all for the columnar engine: route rows to the partition's
+    owning rank (``pid % world``).  Returns (keys, vals, pids) owned by
+    this rank.  NCCL/RCCL path = three all_to_all_single calls; gloo path
+    (CPU tests) = all_gather_object emulation."""
+    if world == 1:
+        return keys, vals, pids
+    owner = torch.remainder(pids, world)
+    order = torch.argsort(owner, stable=True)
+    keys, vals, pids, owner = (keys[order], vals[order], pids[order],
+                               owner[order])
+    send_counts = torch.bincount(owner, minlength=world)
+
+    if dist.get_backend(group) == "nccl":
+        recv_counts = torch.empty_like(send_counts)
+        dist.all_to_all_single(recv_counts, send_counts, group=group)
+        in_sp = send_counts.tolist()
+        out_sp = recv_counts.tolist()
+        rk = keys.new_empty(sum(out_sp))
+        rv = vals.new_empty(sum(out_sp))
+        rp = pids.new_empty(sum(out_sp))
+        dist.all_to_all_single(rk, keys, out_sp, in_sp, group=group)
+        dist.all_to_all_single(rv, vals, out_sp, in_sp, group=group)
+        dist.all_to_all_single(rp, pids, out_sp, in_sp, group=group)
+        return rk, rv, rp
+
+    rank = dist.get_rank(group)
+    gathered = [None] * world
+    dist.all_gather_object(
+        gathered, (keys.cpu(), vals.cpu(), pids.cpu()), group=group)
+    ok, ov, op = [], [], []
+    for r in range(world):
+        k, v, p = gathered[r]
+        mine = torch.remainder(p, world) == rank
+        ok.append(k[mine])
+        ov.append(v[mine])
+        op.append(p[mine])
+    return (torch.cat(ok).to(keys.device), torch.cat(ov).to(vals.device),
+            torch.cat(op).to(pids.device))
+
+
 def all_reduce_scalar(x, group=None, device=None):
     """Sum an int across ranks (doc totals, C2 role)."""
     t = torch.tensor([x], dtype=torch.int64,

@@ -51,5 +51,14 @@ gpu_table_load = 0.50
 # Partitions per GPU for the device-side shuffle.
 gpu_partitions_per_rank = 8
 
+# Total partitions for the columnar engine's store (GpuRunner); sized so
+# one partition of a 288 GB-per-GPU job fits comfortably in the pool.
+gpu_partitions = 64
+
+# Columnar engine HBM pool capacity (MB) before runs spill to pinned host.
+# Default stays small enough for CPU test runs; bench/production set it to
+# ~0.9 * free HBM.
+hbm_pool_mb = int(os.environ.get("DAMPR_HBM_POOL_MB", "16384"))
+
 # Directory for host-side spill of device batches.
 spill_dir = os.environ.get("DAMPR_SPILL_DIR", "/tmp")

@@ -1,0 +1,208 @@
+"""Columnar (device) engine tests on the TorchOps CPU oracle.
+
+The same GpuRunner logic runs here and on the MI355X; only the ops backend
+differs (backend.TorchOps vs backend.HipOps).  GPU-side equivalence is in
+test_gpu_engine.py.
+"""
+import collections
+import multiprocessing
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from dampr_amd import Dampr, funcs
+from dampr_amd.gpu.engine import GpuRunner
+
+
+def by_key(pairs):
+    return sorted(pairs)
+
+
+# --------------------------------------------------------------- basics
+
+def test_columns_count():
+    rng = np.random.default_rng(0)
+    vals = rng.integers(0, 50, size=5000)
+    got = Dampr.columns(vals).count().run()
+    want = collections.Counter(int(v) for v in vals)
+    got_pairs = sorted(got.read())
+    assert got_pairs == sorted(want.items())
+
+
+def test_columns_fold_by_sum_min_max():
+    rng = np.random.default_rng(1)
+    vals = rng.integers(-100, 100, size=3000)
+    keys = rng.integers(0, 20, size=3000)
+
+    for binop, pyop in ((funcs.add, sum), (min, min), (max, max)):
+        got = Dampr.columns(vals, keys=keys).a_group_by(funcs.identity) \
+            .reduce(binop)
+        # a_group_by keys by key(value): identity groups equal values
+        res = dict(got.run().read())
+        groups = collections.defaultdict(list)
+        for v in vals:
+            groups[int(v)].append(int(v))
+        want = {k: pyop(vs) for k, vs in groups.items()}
+        assert res == want
+
+
+def test_columns_sum_and_first():
+    vals = np.array([5, 5, 7, 7, 7, 9], dtype=np.int64)
+    res = dict(Dampr.columns(vals).a_group_by().sum().run().read())
+    assert res == {5: 10, 7: 21, 9: 9}
+    res = dict(Dampr.columns(vals).a_group_by().first()
+               .run().read())
+    assert res == {5: 5, 7: 7, 9: 9}
+
+
+def test_float_values_sum():
+    rng = np.random.default_rng(3)
+    vals = rng.standard_normal(1000)
+    keys = rng.integers(0, 8, size=1000)
+    res = dict(Dampr.columns(vals, keys=keys)
+               .fold_by(funcs.identity, funcs.add).run().read())
+    groups = collections.defaultdict(float)
+    for v in vals:
+        groups[float(v)] += float(v)
+    # identity key on the float value: each distinct float its own group
+    assert len(res) == len(groups)
+
+
+def test_device_engine_selected():
+    """Columnar inputs must route to GpuRunner automatically."""
+    from dampr_amd.dampr import _pick_runner
+    pm = Dampr.columns(np.arange(10))
+    assert _pick_runner(pm.pmer.graph, None) is GpuRunner
+    pm2 = Dampr.memory(list(range(10)))
+    assert _pick_runner(pm2.pmer.graph, None) is not GpuRunner
+
+
+# ------------------------------------------------------- host fallback
+
+def test_opaque_map_falls_back_and_composes():
+    vals = np.arange(100)
+    # .map(lambda) is opaque -> host fallback stage; the count after it
+    # still runs on the columnar path
+    got = Dampr.columns(vals).map(lambda v: v % 7).count().run()
+    want = collections.Counter(int(v) % 7 for v in vals)
+    assert sorted(got.read()) == sorted(want.items())
+
+
+def test_opaque_filter_fallback():
+    vals = np.arange(1000)
+    got = Dampr.columns(vals).filter(lambda v: v % 3 == 0).count().run()
+    want = {int(v): 1 for v in range(0, 1000, 3)}
+    assert dict(got.read()) == want
+
+
+# --------------------------------------------------------------- joins
+
+def test_device_join_pair_product():
+    lk = np.array([1, 1, 2, 3], dtype=np.int64)
+    lv = np.array([10, 20, 30, 40], dtype=np.int64)
+    rk = np.array([1, 2, 2, 9], dtype=np.int64)
+    rv = np.array([2, 3, 4, 5], dtype=np.int64)
+    left = Dampr.columns(lv, keys=lk)
+    right = Dampr.columns(rv, keys=rk)
+    out = left.join(right).reduce(funcs.pair_product, many=True).run()
+    got = sorted(out.read())
+    want = []
+    for i, k in enumerate(lk):
+        for j, k2 in enumerate(rk):
+            if k == k2:
+                want.append((int(k), int(lv[i]) * int(rv[j])))
+    assert got == sorted(want)
+
+
+def test_join_opaque_aggregate_fallback():
+    lk = np.array([1, 2, 3], dtype=np.int64)
+    lv = np.array([10, 20, 30], dtype=np.int64)
+    rk = np.array([2, 3, 4], dtype=np.int64)
+    rv = np.array([1, 2, 3], dtype=np.int64)
+    out = Dampr.columns(lv, keys=lk).join(Dampr.columns(rv, keys=rk)) \
+        .reduce(lambda l, r: sum(l) + sum(r)).run()
+    got = sorted(out.read())
+    assert got == [(2, 21), (3, 32)]
+
+
+# ------------------------------------------------------------- spill
+
+def test_spill_watermark_same_result():
+    rng = np.random.default_rng(5)
+    vals = rng.integers(0, 200, size=20000)
+    pm = Dampr.columns(vals).count()
+    # tiny pool: force every run to spill and reload
+    got = pm.run(hbm_bytes=4096)
+    want = collections.Counter(int(v) for v in vals)
+    assert sorted(got.read()) == sorted(want.items())
+
+
+def test_spill_actually_spills():
+    from dampr_amd.gpu.engine import DeviceRun, HbmPool
+    pool = HbmPool(1024)
+    runs = []
+    for i in range(8):
+        k = torch.arange(64, dtype=torch.int64)
+        run = DeviceRun(k, k.clone(), sorted=True)
+        runs.append(run)
+        pool.admit(run)
+    assert any(not r.resident for r in runs)
+    # reload
+    for r in runs:
+        pool.touch(r, torch.device("cpu"))
+        assert r.resident
+        pool.release(r)
+
+
+# --------------------------------------------------------------- sinks
+
+def test_sink_values(tmp_path):
+    vals = np.array([3, 1, 2], dtype=np.int64)
+    path = str(tmp_path / "out")
+    Dampr.columns(vals).checkpoint(True).sink(path).run()
+    lines = []
+    for f in sorted(os.listdir(path)):
+        with open(os.path.join(path, f)) as fh:
+            lines.extend(ln.strip() for ln in fh if ln.strip())
+    assert sorted(lines) == ["1", "2", "3"]
+
+
+# -------------------------------------------------- multi-rank (gloo)
+
+def _engine_rank(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        vals = np.tile(np.arange(40), 50)       # same input on all ranks
+        got = Dampr.columns(vals).count().run()
+        pairs = sorted(got.read())
+        # every rank owns a disjoint subset; union checked via all_gather
+        gathered = [None] * world
+        dist.all_gather_object(gathered, pairs)
+        merged = sorted(p for lst in gathered for p in lst)
+        want = collections.Counter(int(v) for v in vals)
+        assert merged == sorted(want.items()), merged[:5]
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+def test_engine_gloo_world2():
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    world = 2
+    procs = [ctx.Process(target=_engine_rank, args=(r, world, 29517, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)

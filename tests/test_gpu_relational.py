@@ -146,3 +146,16 @@ def test_topk(rng):
     want = np.argsort(-vals_np)[:10]
     np.testing.assert_array_equal(np.sort(vals_np[idx])[::-1],
                                   np.sort(vals_np[want])[::-1])
+
+
+def test_radix_sort_input_not_clobbered(rng):
+    """Regression: multi-pass sort once used the caller's tensors as
+    ping-pong scratch, overwriting the input after two passes."""
+    from dampr_amd.gpu.relational import radix_sort_pairs
+    n = 100_000
+    keys_np = rng.integers(0, 1 << 63, size=n, dtype=np.int64)
+    keys = torch.from_numpy(keys_np).to(DEV)
+    sk, sp = radix_sort_pairs(keys)
+    torch.cuda.synchronize()
+    np.testing.assert_array_equal(keys.cpu().numpy(), keys_np)
+    assert torch.equal(keys[sp.to(torch.int64)], sk)
