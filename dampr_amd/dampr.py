@@ -160,10 +160,10 @@ class PMap(PBase):
         too."""
         return self.custom_mapper(StreamMapper(f), **options)
 
-    def partition_reduce(self, f):
+    def partition_reduce(self, f, **options):
         """Reduce over a whole partition's (key, values) iterator; runs on
         empty partitions too."""
-        return self.custom_reducer(StreamReducer(f))
+        return self.custom_reducer(StreamReducer(f), **options)
 
     # -- record-level operators ---------------------------------------------
 
@@ -277,17 +277,36 @@ class PMap(PBase):
         return self.a_group_by(key, funcs.one) \
                    .reduce(operator.add, **options)
 
-    def mean(self, key=lambda x: 1, value=lambda x: x, **options):
-        """Mean of value(v) grouped by key(v)."""
+    def mean(self, key=None, value=None, **options):
+        """Mean of value(v) grouped by key(v).
+
+        Recognized key/value funcs lower the whole chain to device: the
+        segmented-reduce kernel computes sum and count, the average fuses
+        into the same stage, and the trailing host ``_average`` map is
+        tagged device-identity (the division already happened on device)."""
+        key = funcs.one if key is None else key
+        value = funcs.identity if value is None else value
+
         def _mean_binop(x, y):
             return x[0] + y[0], x[1] + y[1]
 
         def _average(x):
             return (x[0], x[1][0] / float(x[1][1]))
 
-        return self.a_group_by(key, lambda v: (value(v), 1)) \
-                   .reduce(_mean_binop, **options) \
-                   .map(_average)
+        ar = self.a_group_by(key, lambda v: (value(v), 1))
+        kname = funcs.column_func_name(key)
+        vname = funcs.column_func_name(value)
+        dev = None
+        if kname and vname and ar._device_map is None and not self.agg:
+            # a_group_by saw the opaque tuple lambda; re-tag with the
+            # device form (plain value column; reduce computes the mean)
+            ar._device_map = ("kv", kname, vname)
+            dev = ("mean",)
+        pm = ar._run(_mean_binop, dev, 1000, options)
+        out = pm.map(_average)
+        if dev is not None:
+            return out.checkpoint(options={"device_map": ("identity",)})
+        return out
 
     def len(self):
         """Number of records in the collection."""
@@ -312,7 +331,12 @@ class PMap(PBase):
                    .map(lambda x: x[1])
 
     def topk(self, k, value=None):
-        """Top-k values ordered by value(x) (K11)."""
+        """Top-k values ordered by value(x) (K11).
+
+        ``value=None`` (natural order) on columnar inputs lowers to the
+        device path: per-partition radix top-k candidates, single-partition
+        final merge."""
+        device_ok = value is None and not self.agg
         if value is None:
             value = lambda x: x
         import heapq
@@ -330,9 +354,17 @@ class PMap(PBase):
             for _score, x in heapq.nlargest(k, candidates):
                 yield x, 1
 
-        return self.partition_map(map_topk) \
-                   .partition_reduce(reduce_topk) \
-                   .map(lambda x: x[0])
+        if not device_ok:
+            return self.partition_map(map_topk) \
+                       .partition_reduce(reduce_topk) \
+                       .map(lambda x: x[0])
+        me = self._add_mapper(StreamMapper(map_topk)) \
+            .checkpoint(options={"device_map": ("topk_local", k)})
+        source, pmer = me.pmer._add_reducer(
+            [me.source], StreamReducer(reduce_topk),
+            options={"device_reduce": ("topk_global", k)})
+        out = PMap(source, pmer).map(lambda x: x[0])
+        return out.checkpoint(options={"device_map": ("identity",)})
 
     # -- multi-graph operators ----------------------------------------------
 

@@ -91,6 +91,16 @@ class PartStore(dict):
         self.fkeys = fkeys
 
 
+class HostStore(list):
+    """Fallback record storage: a plain list of (k, v) Python records for
+    stages whose outputs don't fit typed columns.  Downstream stages run
+    on the host until records become numeric again (then they re-enter the
+    columnar path via _encode_or_host)."""
+
+    keyed = False
+    fkeys = False
+
+
 def _decode_f64_sortable(enc):
     """Inverse of relational.encode_f64_sortable."""
     sign_bit = -(1 << 63)
@@ -304,9 +314,15 @@ class GpuRunner(RunnerBase):
                 hi = n * (self.rank + 1) // self.world
                 keys, vals = keys[lo:hi], vals[lo:hi]
             return self._partition(keys, vals)
-        # host dataset / chunker: stream records, encode
+        # host dataset / chunker: stream records; numeric records become
+        # columns, object records stay host-side (HostStore)
         records = self._host_records_of_input(inp)
-        return self._encode_records(records)
+        if self.world > 1:
+            n = len(records)
+            lo = n * self.rank // self.world
+            hi = n * (self.rank + 1) // self.world
+            records = records[lo:hi]
+        return self._encode_or_host(records)
 
     @staticmethod
     def _host_records_of_input(inp):
@@ -357,12 +373,21 @@ class GpuRunner(RunnerBase):
         return self._partition(kt.to(self.device), vt.to(self.device),
                                keyed=keyed, fkeys=fkeys)
 
+    def _encode_or_host(self, records):
+        try:
+            return self._encode_records(records)
+        except TypeError:
+            return HostStore(records)
+
     def _collect(self, store):
         """Partition store -> one key-sorted ColumnDataset (the engine's
         MergeDataset analog: partitions are key-sorted, output is their
         merge)."""
         if isinstance(store, ColumnDataset):
             return store
+        if isinstance(store, HostStore):
+            from ..dataset import MemoryDataset
+            return MemoryDataset(sorted(store, key=lambda r: r[0]))
         keyed = getattr(store, "keyed", False)
         fkeys = getattr(store, "fkeys", False)
         ks, vs = [], []
@@ -377,8 +402,22 @@ class GpuRunner(RunnerBase):
             return ColumnDataset(z, z.clone(), keyed, fkeys)
         keys = torch.cat(ks)
         vals = torch.cat(vs)
-        sk, sp = self.ops.sort_pairs(keys)
+        sk, sp = self._sort(keys, fkeys=fkeys)
         return ColumnDataset(sk, vals[sp.to(torch.int64)], keyed, fkeys)
+
+    # -- ordering ----------------------------------------------------------
+
+    _SIGN = -(1 << 63)
+
+    def _sort(self, keys, payload=None, fkeys=False):
+        """Key sort in *host* ascending order: signed for i64 keys, float
+        ascending for f64-encoded keys (their encoding is already
+        unsigned-ascending).  The backend sort is unsigned (built for
+        hashes), so int keys are biased through the sign bit."""
+        if fkeys:
+            return self.ops.sort_pairs(keys, payload)
+        sk, sp = self.ops.sort_pairs(keys ^ self._SIGN, payload)
+        return sk ^ self._SIGN, sp
 
     # -- partitioning ------------------------------------------------------
 
@@ -414,7 +453,7 @@ class GpuRunner(RunnerBase):
             k = keys[o:o + n].contiguous()
             v = vals[o:o + n].contiguous()
             if not already_sorted:
-                sk, sp = self.ops.sort_pairs(k)
+                sk, sp = self._sort(k, fkeys=fkeys)
                 k, v = sk, v[sp.to(torch.int64)]
             run = DeviceRun(k, v, sorted=True)
             store.setdefault(p, []).append(run)
@@ -429,6 +468,7 @@ class GpuRunner(RunnerBase):
 
     def _merged_partition(self, stores, p):
         """All runs of partition p across input stores, merged key-sorted."""
+        fkeys = any(getattr(s, "fkeys", False) for s in stores)
         ks, vs = [], []
         for store in stores:
             for run in store.get(p, []):
@@ -442,7 +482,7 @@ class GpuRunner(RunnerBase):
             return ks[0], vs[0]
         keys = torch.cat(ks)
         vals = torch.cat(vs)
-        sk, sp = self.ops.sort_pairs(keys)
+        sk, sp = self._sort(keys, fkeys=fkeys)
         return sk, vals[sp.to(torch.int64)]
 
     def _parts(self, stores):
@@ -455,7 +495,7 @@ class GpuRunner(RunnerBase):
 
     def run_map(self, stage, ins):
         spec = stage.options.get("device_map")
-        if spec is None:
+        if spec is None or any(isinstance(s, HostStore) for s in ins):
             return self._host_map(stage, ins)
         kind = spec[0]
         if kind == "kv":
@@ -484,6 +524,43 @@ class GpuRunner(RunnerBase):
             return out if out is not None else PartStore()
         if kind == "identity":
             return self._merge_stores(ins)
+        if kind == "topk_local":
+            # per-partition top-k candidates by value (K11); all
+            # candidates meet in partition 0 for the global pass
+            if any(getattr(s, "keyed", False) for s in ins):
+                return self._host_map(stage, ins)
+            K = spec[1]
+            cand_k, cand_v = [], []
+            fkeys = False
+            for p in self._parts(ins):
+                keys, vals = self._merged_partition(ins, p)
+                if keys is None:
+                    continue
+                if vals.dtype == torch.float64:
+                    enc = _encode_f64_sortable(vals)
+                    fkeys = True
+                else:
+                    enc = vals
+                sk, sp = self._sort(enc, fkeys=vals.dtype
+                                    == torch.float64)
+                top = min(K, sk.numel())
+                cand_k.append(sk[-top:])
+                cand_v.append(vals[sp.to(torch.int64)[-top:]])
+            store = PartStore(fkeys=fkeys)
+            if not cand_k:
+                return store
+            ck = torch.cat(cand_k)
+            cv = torch.cat(cand_v)
+            if self.world > 1:
+                zeros = torch.zeros_like(ck)
+                ck, cv, _ = self._exchange(ck, cv, zeros)
+                if self.rank != 0 or ck.numel() == 0:
+                    return store
+            sk, sp = self._sort(ck, fkeys=fkeys)
+            run = DeviceRun(sk, cv[sp.to(torch.int64)], sorted=True)
+            store[0] = [run]
+            self.pool.admit(run)
+            return store
         raise ValueError("unknown device_map spec {!r}".format(spec))
 
     def _apply_colfunc(self, name, keys, vals):
@@ -498,6 +575,11 @@ class GpuRunner(RunnerBase):
     def _merge_stores(self, stores):
         if len(stores) == 1:
             return stores[0]
+        if any(isinstance(s, HostStore) for s in stores):
+            out = HostStore()
+            for s in stores:
+                out.extend(self._decode_store(s))
+            return out
         out = PartStore(
             keyed=any(getattr(s, "keyed", False) for s in stores),
             fkeys=any(getattr(s, "fkeys", False) for s in stores))
@@ -510,7 +592,8 @@ class GpuRunner(RunnerBase):
 
     def run_reduce(self, stage, ins):
         spec = stage.options.get("device_reduce")
-        if spec is None or any(getattr(s, "keyed", False) for s in ins):
+        if spec is None or any(getattr(s, "keyed", False) for s in ins) \
+                or any(isinstance(s, HostStore) for s in ins):
             return self._host_reduce(stage, ins)
         kind = spec[0]
         in_fkeys = any(getattr(s, "fkeys", False) for s in ins)
@@ -553,6 +636,28 @@ class GpuRunner(RunnerBase):
                 out.setdefault(p, []).append(run)
                 self.pool.admit(run)
             return out
+        if kind == "topk_global":
+            K = spec[1]
+            ks, vs = [], []
+            for p in self._parts(ins):
+                keys, vals = self._merged_partition(ins, p)
+                if keys is None:
+                    continue
+                ks.append(keys)
+                vs.append(vals)
+            out.keyed = False
+            if not ks:
+                return out
+            keys = torch.cat(ks)
+            vals = torch.cat(vs)
+            sk, sp = self._sort(keys, fkeys=in_fkeys)
+            top = min(K, sk.numel())
+            run = DeviceRun(sk[-top:].contiguous(),
+                            vals[sp.to(torch.int64)[-top:]].contiguous(),
+                            sorted=True)
+            out[0] = [run]
+            self.pool.admit(run)
+            return out
         if kind == "join":
             how = spec[1]
             assert len(ins) == 2, "join takes two inputs"
@@ -593,7 +698,7 @@ class GpuRunner(RunnerBase):
             valid_r = ri >= 0
             merged = self._apply_pair_op(pair_op, lvm, rvm, valid_l,
                                          valid_r)
-            sk, sp = self.ops.sort_pairs(keys)
+            sk, sp = self._sort(keys, fkeys=out.fkeys)
             run = DeviceRun(sk, merged[sp.to(torch.int64)], sorted=True)
             out.setdefault(p, []).append(run)
             self.pool.admit(run)
@@ -640,6 +745,8 @@ class GpuRunner(RunnerBase):
     # -- host fallback -----------------------------------------------------
 
     def _decode_store(self, store):
+        if isinstance(store, HostStore):
+            return list(store)
         records = []
         keyed = getattr(store, "keyed", False)
         fkeys = getattr(store, "fkeys", False)
@@ -667,20 +774,14 @@ class GpuRunner(RunnerBase):
         supplemental = [[MemoryDataset(self._decode_store(s))]
                         for s in ins[1:]]
         out_records = list(stage.mapper.map(primary, *supplemental))
-        return self._encode_records(out_records)
+        return self._encode_or_host(out_records)
 
     def _host_reduce(self, stage, ins):
-        out_records = []
-        for p in self._parts(ins):
-            datasets = []
-            for store in ins:
-                keys, vals = self._merged_partition([store], p)
-                if keys is None:
-                    continue
-                datasets.append([ColumnDataset(keys.cpu(), vals.cpu())])
-            if not datasets:
-                continue
-            while len(datasets) < len(ins):
-                datasets.append([])
-            out_records.extend(stage.reducer.reduce(*datasets))
-        return self._encode_records(out_records)
+        from ..dataset import MemoryDataset
+        datasets = []
+        for store in ins:
+            recs = self._decode_store(store)
+            recs.sort(key=lambda r: r[0])     # reducers need sorted streams
+            datasets.append([MemoryDataset(recs)])
+        out_records = list(stage.reducer.reduce(*datasets))
+        return self._encode_or_host(out_records)
