@@ -266,10 +266,17 @@ class PMap(PBase):
         """a_group_by(key, value).reduce(binop)."""
         return self.a_group_by(key, value).reduce(binop, **options)
 
-    def sort_by(self, key, **options):
-        """Totally order the collection by key(value)."""
+    def sort_by(self, key=None, **options):
+        """Totally order the collection by key(value).  Recognized key
+        funcs lower to the device radix sort (K3)."""
+        key = funcs.identity if key is None else key
+
         def _sort_by(_k, value):
             yield key(value), value
+        kname = funcs.column_func_name(key)
+        if kname and not self.agg:
+            options = dict(options)
+            options.setdefault("device_map", ("kv", kname, "identity"))
         return self._add_map(_sort_by).checkpoint(options=options)
 
     def count(self, key=None, **options):

@@ -123,3 +123,35 @@ def test_backend_cross_check():
     P = 64
     assert torch.equal(hip.partition_of(hk, P).cpu(),
                        cpu.partition_of(ck, P))
+
+
+def test_mean_device():
+    rng = np.random.default_rng(30)
+    vals = rng.integers(0, 500, size=500_000)
+    got = dict(_run_dev(Dampr.columns(vals).mean(funcs.identity)).read())
+    groups = collections.defaultdict(list)
+    for v in vals:
+        groups[int(v)].append(int(v))
+    want = {k: sum(vs) / float(len(vs)) for k, vs in groups.items()}
+    assert got == want
+
+
+def test_topk_device_gpu():
+    rng = np.random.default_rng(31)
+    vals = rng.integers(-(1 << 50), 1 << 50, size=1_000_000)
+    got = _run_dev(Dampr.columns(vals).topk(25)).read()
+    want = sorted(sorted((int(v) for v in vals), reverse=True)[:25])
+    assert sorted(got) == want
+
+
+def test_sort_by_device_gpu():
+    rng = np.random.default_rng(32)
+    vals = rng.integers(-(1 << 40), 1 << 40, size=500_000)
+    got = _run_dev(Dampr.columns(vals).sort_by()).read()
+    assert got == sorted(int(v) for v in vals)
+
+
+def test_negative_key_order_gpu():
+    vals = np.array([5, -3, 7, -3, 5], dtype=np.int64)
+    got = list(_run_dev(Dampr.columns(vals).count()).read())
+    assert got == [(-3, 2), (5, 2), (7, 1)]
