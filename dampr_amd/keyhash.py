@@ -31,6 +31,18 @@ def splitmix64(x: int) -> int:
     return x ^ (x >> 31)
 
 
+def tokmix64(data: bytes) -> int:
+    """Order-aware XOR-of-mixes token hash; must match tokmix_step() in
+    ops/hip/common.h (the GPU computes it with a wave-wide segmented XOR
+    scan; XOR's associativity makes serial and parallel answers equal).
+    Bytes are lowercased like the device tokenizer."""
+    h = 0
+    for j, b in enumerate(data):
+        c = b + 32 if 65 <= b <= 90 else b
+        h ^= splitmix64(((j << 8) | c) & _MASK)
+    return h
+
+
 def key_hash64(key) -> int:
     """Canonical u64 hash of a record key."""
     if isinstance(key, bool):          # bool before int: True is an int

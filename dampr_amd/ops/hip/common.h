@@ -26,6 +26,15 @@ __device__ __forceinline__ u64 splitmix64(u64 x) {
     return x ^ (x >> 31);
 }
 
+// Order-aware, *parallelizable* token hash: XOR over bytes of
+// splitmix64(pos << 8 | lower(byte)).  XOR is associative/commutative, so
+// a wave can compute it with a segmented scan while a serial loop gets the
+// identical value (mirrored in keyhash.tokmix64).  Positional mixing keeps
+// order sensitivity ("ab" != "ba").
+__device__ __forceinline__ u64 tokmix_step(u64 h, u32 pos, u8 lowered) {
+    return h ^ splitmix64(((u64)pos << 8) | (u64)lowered);
+}
+
 // ASCII '\w' classification with lowercasing, matching the reference's
 // tokenizer regex r'[^\w]+' (benchmarks/tf-idf-dampr.py:12) on ASCII text.
 __device__ __forceinline__ u8 lower_ascii(u8 c) {

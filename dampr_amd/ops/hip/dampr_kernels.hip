@@ -149,12 +149,14 @@ __global__ void tfidf_count_kernel(
          t += stride) {
         u32 start = tok_start[t];
         long p = start;
-        u64 h = FNV_OFFSET;
+        u64 h = 0;
+        u32 j = 0;
         while (p < n) {
             u8 c = text[p];
             if (!is_word(c)) break;
-            h = fnv1a64_step(h, lower_ascii(c));
+            h = tokmix_step(h, j, lower_ascii(c));
             ++p;
+            ++j;
         }
         u32 len = (u32)(p - start);
         // doc id = number of newlines strictly before `start`
@@ -237,7 +239,15 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                   u64 dict_mask, u64 pos_base,
                   u64* __restrict__ fb_seen, u64 fb_mask,
                   u32* __restrict__ err_flag, u32 ablate) {
-    __shared__ u8 stage[DOC_WAVES][STAGE_B];
+    // One wave per document; all 64 lanes active on every byte:
+    //  * 16 B/lane uint4 staging into LDS (aligned; no overlap restaging)
+    //  * 64-B windows, one BYTE per lane: ballot-derived token starts,
+    //    wave-wide segmented XOR scan builds every token's hash in
+    //    parallel (tokmix, common.h) — no serial per-token walks
+    //  * tokens crossing window/segment boundaries ride a wave-uniform
+    //    carry (partial hash, length, start), so arbitrarily long tokens
+    //    hash exactly with zero re-reads
+    __shared__ __align__(16) u8 stage[DOC_WAVES][STAGE_B + 16];
     __shared__ u64 dset[DOC_WAVES][DOC_SET];
     __shared__ u64 cck[CCACHE];
     __shared__ u32 ccv[CCACHE];
@@ -259,79 +269,122 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
         const long le = (d < n_nl) ? (long)nl_pos[d] : n;
         for (int s = lane; s < DOC_SET; s += WAVE) set[s] = 0;
 
-        for (long seg = ls; seg < le; seg += STAGE_B - SEG_OVERLAP) {
+        u32 carry_word = 0;            // wave-uniform token carry
+        u32 carry_len = 0;
+        u64 carry_g = 0;
+        long carry_start = ls;
+
+        for (long seg = ls; seg < le; seg += STAGE_B) {
             const long seg_end = min(seg + (long)STAGE_B, le);
             const int seg_len = (int)(seg_end - seg);
-            for (int i = lane; i < seg_len; i += WAVE)
-                st[i] = text[seg + i];
-            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            const long aseg = seg & ~15L;
+            const int soff = (int)(seg - aseg);
+            const int stage_bytes = soff + seg_len;
+            for (int i = lane * 16; i < stage_bytes; i += WAVE * 16) {
+                if (aseg + i + 16 <= n) {
+                    *reinterpret_cast<uint4*>(st + i) =
+                        *reinterpret_cast<const uint4*>(text + aseg + i);
+                } else {
+                    for (int j = i; j < stage_bytes; ++j)
+                        st[j] = (aseg + j < n) ? text[aseg + j] : (u8)0;
+                }
+            }
+            asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
             __builtin_amdgcn_wave_barrier();
 
-            const int start_lim = (seg_end == le)
-                ? seg_len : (STAGE_B - SEG_OVERLAP);
-            for (int base = 0; base < start_lim; base += WAVE) {
-                const int p = base + lane;
-                bool is_start = false;
-                if (p < start_lim) {
-                    const u8 c = st[p];
-                    bool pw;
-                    if (p > 0) pw = is_word(st[p - 1]);
-                    else pw = (seg > 0) ? is_word(text[seg - 1]) : false;
-                    is_start = is_word(c) && !pw;
+            const bool doc_continues = seg_end < le;
+            const int nwin = (seg_len + WAVE - 1) / WAVE;
+            for (int wdx = 0; wdx < nwin; ++wdx) {
+                const int wb = wdx * WAVE;
+                const int p = wb + lane;
+                const bool valid = p < seg_len;
+                const u8 c = valid ? st[soff + p] : (u8)0;
+                const bool w = valid && is_word(c);
+                const u64 wm = __ballot(w);
+                const u64 sm = wm & ~((wm << 1) | (u64)carry_word);
+                const u64 below_inc = (lane == 63)
+                    ? ~0ULL : ((1ULL << (lane + 1)) - 1ULL);
+                const u64 sm_le = sm & below_inc;
+                int s = WAVE;          // token start (window coords)
+                int pos = 0;           // byte position within token
+                u64 g = 0;
+                if (w) {
+                    s = sm_le ? (63 - __clzll(sm_le)) : -(int)carry_len;
+                    pos = lane - s;
+                    g = splitmix64(((u64)(u32)pos << 8)
+                                   | (u64)lower_ascii(c));
                 }
-                if (!is_start) continue;
-                int q = p;
-                u64 h = FNV_OFFSET;
-                while (q < seg_len) {
-                    const u8 c = st[q];
-                    if (!is_word(c)) break;
-                    h = fnv1a64_step(h, lower_ascii(c));
-                    ++q;
+                // segmented inclusive XOR scan along the wave
+                #pragma unroll
+                for (int dsh = 1; dsh < WAVE; dsh <<= 1) {
+                    const u64 g2 = __shfl_up(g, dsh, WAVE);
+                    if (w && lane >= dsh && (lane - dsh) >= s) g ^= g2;
                 }
-                u32 tl = (u32)(q - p);
-                if (q == seg_len && seg_end < le) {
-                    // pathological >2 KiB token: finish from global
-                    long g = seg + q;
-                    while (g < le && is_word(text[g])) {
-                        h = fnv1a64_step(h, lower_ascii(text[g]));
-                        ++g; ++tl;
-                    }
-                }
-                if (ablate & 1) {           // ablation: hash only
-                    if (h == 0xdeadbeefdeadbeefULL) err_flag[1] = 1;
-                    continue;
-                }
-                int fresh = lds_set_insert(set, h ? h : 1ULL);
-                if (fresh < 0) {
-                    // overflow: global (doc,hash) seen fallback
-                    u64 sk = splitmix64(h ^ ((u64)d
-                                             * 0x9E3779B97F4A7C15ULL));
-                    if (!sk) sk = 1;
-                    u64 slot = sk & fb_mask;
-                    fresh = 0;
-                    int probe = 0;
-                    while (true) {
-                        u64 prev = atomicCAS(&fb_seen[slot], 0ULL, sk);
-                        if (prev == 0ULL) { fresh = 1; break; }
-                        if (prev == sk) break;
-                        slot = (slot + 1) & fb_mask;
-                        if (++probe > FB_PROBE_CAP) {
-                            atomicOr(err_flag, 1u);
-                            break;
+                if (w && s < 0) g ^= carry_g;   // continuing token prefix
+
+                const int last_valid = min(seg_len - wb, WAVE) - 1;
+                bool at_end = w
+                    && (lane == 63 ? true : !((wm >> (lane + 1)) & 1));
+                if (lane == last_valid && w
+                    && (last_valid == 63 || doc_continues))
+                    at_end = false;    // token may continue: carry it
+
+                if (at_end && !(ablate & 1)) {
+                    const u64 key = g ? g : 1ULL;
+                    const u32 tl = (u32)(pos + 1);
+                    int fresh = lds_set_insert(set, key);
+                    if (fresh < 0) {
+                        // set overflow: global (doc,hash) seen fallback
+                        u64 sk = splitmix64(
+                            g ^ ((u64)d * 0x9E3779B97F4A7C15ULL));
+                        if (!sk) sk = 1;
+                        u64 slot = sk & fb_mask;
+                        fresh = 0;
+                        int probe = 0;
+                        while (true) {
+                            u64 prev = atomicCAS(&fb_seen[slot], 0ULL,
+                                                 sk);
+                            if (prev == 0ULL) { fresh = 1; break; }
+                            if (prev == sk) break;
+                            slot = (slot + 1) & fb_mask;
+                            if (++probe > FB_PROBE_CAP) {
+                                atomicOr(err_flag, 1u);
+                                break;
+                            }
                         }
                     }
+                    if (fresh == 1 && !(ablate & 2)) {
+                        block_count_add(cck, ccv, key, cnt_keys,
+                                        cnt_vals, cnt_mask);
+                        u64 slot;
+                        const long tstart = (s >= 0)
+                            ? (seg + wb + s) : carry_start;
+                        if (!(ablate & 4)
+                            && table_insert_u64(dict_keys, dict_mask,
+                                                key, &slot))
+                            dict_vals[slot] =
+                                ((pos_base + (u64)tstart) << 8)
+                                | (u64)min(tl, 255u);
+                    }
+                } else if (at_end) {   // ablation: consume the hash
+                    if (g == 0xdeadbeefdeadbeefULL) err_flag[1] = 1;
                 }
-                if (fresh == 1 && !(ablate & 2)) {
-                    const u64 key = h ? h : 1ULL;
-                    block_count_add(cck, ccv, key, cnt_keys, cnt_vals,
-                                    cnt_mask);
-                    u64 slot;
-                    if (!(ablate & 4)
-                        && table_insert_u64(dict_keys, dict_mask, key,
-                                            &slot))
-                        dict_vals[slot] =
-                            ((pos_base + (u64)(seg + p)) << 8)
-                            | (u64)min(tl, 255u);
+
+                // wave-uniform carry update from the tail lane
+                const int t_w = (int)((wm >> last_valid) & 1);
+                const int t_end = __shfl((int)at_end, last_valid, WAVE);
+                if (t_w && !t_end) {
+                    const int t_pos = __shfl(pos, last_valid, WAVE);
+                    const int t_s = __shfl(s, last_valid, WAVE);
+                    const u64 t_g = __shfl(g, last_valid, WAVE);
+                    carry_word = 1;
+                    carry_len = (u32)(t_pos + 1);
+                    carry_g = t_g;
+                    if (t_s >= 0) carry_start = seg + wb + t_s;
+                } else {
+                    carry_word = 0;
+                    carry_len = 0;
+                    carry_g = 0;
                 }
             }
             __builtin_amdgcn_wave_barrier();
