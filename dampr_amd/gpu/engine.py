@@ -277,21 +277,25 @@ class GpuRunner(RunnerBase):
     # -- plan walk ---------------------------------------------------------
 
     def run(self, outputs, cleanup=True):
+        from ..utils.trace import get_trace, trace_stage
+        get_trace().clear()
         data = {}
         for src, inp in self.graph.inputs.items():
-            data[src] = self._ingest(inp)
+            with trace_stage("ingest {}".format(src), self.device):
+                data[src] = self._ingest(inp)
         for stage_id, stage in enumerate(self.graph.stages):
             log.info("[device] Stage %s/%s: %r", stage_id + 1,
                      len(self.graph.stages), stage)
             ins = [data[i] for i in stage.inputs]
-            if isinstance(stage, GMap):
-                out = self.run_map(stage, ins)
-            elif isinstance(stage, GReduce):
-                out = self.run_reduce(stage, ins)
-            elif isinstance(stage, GSink):
-                out = self.run_sink(stage, ins)
-            else:
-                raise TypeError(stage)
+            with trace_stage(repr(stage), self.device):
+                if isinstance(stage, GMap):
+                    out = self.run_map(stage, ins)
+                elif isinstance(stage, GReduce):
+                    out = self.run_reduce(stage, ins)
+                elif isinstance(stage, GSink):
+                    out = self.run_sink(stage, ins)
+                else:
+                    raise TypeError(stage)
             data[stage.output] = out
         rets = []
         for source in outputs:

@@ -305,6 +305,8 @@ class RunnerBase(object):
         self.graph = graph
 
     def run(self, outputs, cleanup=True):
+        from .utils.trace import get_trace, trace_stage
+        get_trace().clear()
         data = dict(self.graph.inputs)
         to_delete = set()
         for stage_id, stage in enumerate(self.graph.stages):
@@ -312,15 +314,17 @@ class RunnerBase(object):
                      len(self.graph.stages), stage)
             input_data = [data[i] for i in stage.inputs]
             cleanup_stage = True
-            if isinstance(stage, GMap):
-                dm = self.run_map(stage_id, input_data, stage)
-            elif isinstance(stage, GReduce):
-                dm = self.run_reducer(stage_id, input_data, stage)
-            elif isinstance(stage, GSink):
-                dm = self.run_sink(stage_id, input_data, stage)
-                cleanup_stage = False
-            else:
-                raise TypeError("unknown stage type: {!r}".format(stage))
+            with trace_stage(repr(stage)):
+                if isinstance(stage, GMap):
+                    dm = self.run_map(stage_id, input_data, stage)
+                elif isinstance(stage, GReduce):
+                    dm = self.run_reducer(stage_id, input_data, stage)
+                elif isinstance(stage, GSink):
+                    dm = self.run_sink(stage_id, input_data, stage)
+                    cleanup_stage = False
+                else:
+                    raise TypeError(
+                        "unknown stage type: {!r}".format(stage))
             data[stage.output] = dm
             if cleanup_stage:
                 to_delete.add(stage.output)

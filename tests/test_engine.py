@@ -277,3 +277,13 @@ def test_object_records_host_store():
     words = ["b", "a", "b", "c", "a", "b"]
     got = Dampr.memory(words).count().run(runner=GpuRunner).read()
     assert sorted(got) == [("a", 2), ("b", 3), ("c", 1)]
+
+
+def test_trace_rows():
+    from dampr_amd.utils.trace import get_trace
+    vals = np.arange(1000)
+    Dampr.columns(vals).count().run()
+    tr = get_trace()
+    assert len(tr.rows) >= 3          # ingest + map + reduce
+    assert all(r["wall_ms"] >= 0 for r in tr.rows)
+    assert "stage" in tr.report()
