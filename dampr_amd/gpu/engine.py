@@ -115,29 +115,42 @@ def _encode_f64_sortable(x):
 
 
 class DeviceRun(object):
-    """One (keys, vals) run of a partition; spillable to host."""
+    """One (keys, vals) run of a partition, spillable down the tier
+    hierarchy HBM -> pinned host -> NVMe file (the reference's gzip-pickle
+    /tmp spill files, dataset.py:119-188, re-expressed for 288 GB HBM +
+    host DRAM + NVMe)."""
 
-    __slots__ = ("keys", "vals", "sorted", "_host")
+    __slots__ = ("keys", "vals", "sorted", "_host", "_disk", "_meta")
 
     def __init__(self, keys, vals, sorted=False):
         self.keys = keys
         self.vals = vals
         self.sorted = sorted
         self._host = None
+        self._disk = None
+        self._meta = None
 
     @property
     def nbytes(self):
-        if self.keys is None:
-            return 0
-        return self.keys.numel() * 8 + self.vals.element_size() * \
-            self.vals.numel()
+        if self.keys is not None:
+            return self.keys.numel() * 8 + self.vals.element_size() * \
+                self.vals.numel()
+        if self._meta is not None:
+            n, vdt = self._meta
+            return n * 8 + n * (8 if vdt in (torch.int64, torch.float64)
+                                else 8)
+        return 0
 
     @property
     def resident(self):
         return self.keys is not None
 
+    @property
+    def on_disk(self):
+        return self._disk is not None
+
     def spill(self):
-        """Move to (pinned) host memory; frees the HBM."""
+        """HBM -> (pinned) host memory."""
         if self._host is not None or self.keys is None:
             return
         pin = self.keys.device.type == "cuda"
@@ -145,12 +158,45 @@ class DeviceRun(object):
         hv = torch.empty_like(self.vals, device="cpu", pin_memory=pin)
         hk.copy_(self.keys)
         hv.copy_(self.vals)
+        self._meta = (self.keys.numel(), self.vals.dtype)
         self._host = (hk, hv)
         self.keys = None
         self.vals = None
 
+    def spill_to_disk(self, path):
+        """Host -> NVMe file (raw little-endian columns, no pickle)."""
+        if self._host is None:
+            return
+        hk, hv = self._host
+        with open(path, "wb") as fh:
+            fh.write(hk.numpy().tobytes())
+            fh.write(hv.numpy().tobytes())
+        self._disk = path
+        self._host = None
+
+    def _load_host(self):
+        if self._host is None and self._disk is not None:
+            import numpy as np
+            n, vdt = self._meta
+            with open(self._disk, "rb") as fh:
+                raw_k = fh.read(n * 8)
+                raw_v = fh.read()
+            hk = torch.from_numpy(
+                np.frombuffer(raw_k, dtype=np.int64).copy())
+            vnp = (np.float64 if vdt == torch.float64 else np.int64)
+            hv = torch.from_numpy(
+                np.frombuffer(raw_v, dtype=vnp).copy())
+            os_mod = __import__("os")
+            try:
+                os_mod.unlink(self._disk)
+            except OSError:
+                pass
+            self._disk = None
+            self._host = (hk, hv)
+
     def load(self, device):
         if self.keys is None:
+            self._load_host()
             hk, hv = self._host
             self.keys = hk.to(device, non_blocking=True)
             self.vals = hv.to(device, non_blocking=True)
@@ -159,14 +205,34 @@ class DeviceRun(object):
 
 
 class HbmPool(object):
-    """Watermark governor over device-resident run bytes (the reference's
-    MemoryChecker/MaxMemoryWriter analog, memory.py:72-113, at HBM
-    granularity)."""
+    """Two-watermark tier governor over run bytes (the reference's
+    MemoryChecker/MaxMemoryWriter analog, memory.py:72-113, re-expressed
+    over tiers): device-resident bytes above ``capacity`` spill to pinned
+    host; host-resident bytes above ``host_capacity`` spill on to raw
+    NVMe files under ``spill_dir``."""
 
-    def __init__(self, capacity_bytes):
+    def __init__(self, capacity_bytes, host_capacity=None, spill_dir=None):
+        import os
+        import uuid
         self.capacity = capacity_bytes
         self.used = 0
-        self._lru = []                 # DeviceRun insertion order
+        self._lru = []                 # device-resident, insertion order
+        self.host_capacity = (host_capacity
+                              if host_capacity is not None
+                              else float("inf"))
+        self.host_used = 0
+        self._host_lru = []
+        self.spill_dir = spill_dir or settings.spill_dir
+        self._run_tag = "dampr_amd_{}".format(uuid.uuid4().hex[:10])
+        self._file_ctr = 0
+        self._os = os
+
+    def _next_path(self):
+        self._os.makedirs(self.spill_dir, exist_ok=True)
+        self._file_ctr += 1
+        return self._os.path.join(
+            self.spill_dir,
+            "{}_{}.run".format(self._run_tag, self._file_ctr))
 
     def admit(self, run):
         self.used += run.nbytes
@@ -175,6 +241,9 @@ class HbmPool(object):
 
     def touch(self, run, device):
         if not run.resident:
+            if run in self._host_lru:
+                self._host_lru.remove(run)
+                self.host_used -= run.nbytes
             run.load(device)
             self.used += run.nbytes
             self._lru.append(run)
@@ -199,6 +268,12 @@ class HbmPool(object):
             self._lru.remove(victim)
             self.used -= victim.nbytes
             victim.spill()
+            self.host_used += victim.nbytes
+            self._host_lru.append(victim)
+        while self.host_used > self.host_capacity and self._host_lru:
+            v = self._host_lru.pop(0)
+            self.host_used -= v.nbytes
+            v.spill_to_disk(self._next_path())
 
 
 # --------------------------------------------------------------------------
@@ -256,7 +331,7 @@ class GpuRunner(RunnerBase):
     """
 
     def __init__(self, name, graph, device=None, n_partitions=None,
-                 hbm_bytes=None):
+                 hbm_bytes=None, host_bytes=None, spill_dir=None):
         # note: RunnerBase.__init__ builds a /tmp FileSystem we don't use;
         # keep it for interface parity (sinks reuse its naming).
         super(GpuRunner, self).__init__(name, graph)
@@ -266,7 +341,10 @@ class GpuRunner(RunnerBase):
         self.ops = ops_for(self.device)
         self.n_partitions = n_partitions or settings.gpu_partitions
         cap = hbm_bytes or settings.hbm_pool_mb * (1 << 20)
-        self.pool = HbmPool(cap)
+        host_cap = host_bytes if host_bytes is not None \
+            else settings.host_pool_mb * (1 << 20)
+        self.pool = HbmPool(cap, host_capacity=host_cap,
+                            spill_dir=spill_dir)
         if torch.distributed.is_available() and \
                 torch.distributed.is_initialized():
             self.world = torch.distributed.get_world_size()

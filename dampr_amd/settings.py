@@ -62,3 +62,7 @@ hbm_pool_mb = int(os.environ.get("DAMPR_HBM_POOL_MB", "16384"))
 
 # Directory for host-side spill of device batches.
 spill_dir = os.environ.get("DAMPR_SPILL_DIR", "/tmp")
+
+# Pinned-host spill tier capacity (MB); beyond it runs spill to raw files
+# under ``spill_dir`` (NVMe tier).
+host_pool_mb = int(os.environ.get("DAMPR_HOST_POOL_MB", "65536"))
