@@ -274,6 +274,45 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
         u64 carry_g = 0;
         long carry_start = ls;
 
+        // dedupe + count + dict insert for one finished token
+        auto emit_token = [&](u64 gh, u32 tl, long tstart) {
+            if (ablate & 1) {          // ablation: consume the hash
+                if (gh == 0xdeadbeefdeadbeefULL) err_flag[1] = 1;
+                return;
+            }
+            const u64 key = gh ? gh : 1ULL;
+            int fresh = lds_set_insert(set, key);
+            if (fresh < 0) {
+                // set overflow: global (doc,hash) seen fallback
+                u64 sk = splitmix64(gh ^ ((u64)d
+                                          * 0x9E3779B97F4A7C15ULL));
+                if (!sk) sk = 1;
+                u64 slot = sk & fb_mask;
+                fresh = 0;
+                int probe = 0;
+                while (true) {
+                    u64 prev = atomicCAS(&fb_seen[slot], 0ULL, sk);
+                    if (prev == 0ULL) { fresh = 1; break; }
+                    if (prev == sk) break;
+                    slot = (slot + 1) & fb_mask;
+                    if (++probe > FB_PROBE_CAP) {
+                        atomicOr(err_flag, 1u);
+                        break;
+                    }
+                }
+            }
+            if (fresh == 1 && !(ablate & 2)) {
+                block_count_add(cck, ccv, key, cnt_keys, cnt_vals,
+                                cnt_mask);
+                u64 slot;
+                if (!(ablate & 4)
+                    && table_insert_u64(dict_keys, dict_mask, key,
+                                        &slot))
+                    dict_vals[slot] = ((pos_base + (u64)tstart) << 8)
+                                      | (u64)min(tl, 255u);
+            }
+        };
+
         for (long seg = ls; seg < le; seg += STAGE_B) {
             const long seg_end = min(seg + (long)STAGE_B, le);
             const int seg_len = (int)(seg_end - seg);
@@ -301,6 +340,14 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                 const u8 c = valid ? st[soff + p] : (u8)0;
                 const bool w = valid && is_word(c);
                 const u64 wm = __ballot(w);
+                if (carry_word && !(wm & 1ULL)) {
+                    // carried token ended exactly at the window edge
+                    if (lane == 0)
+                        emit_token(carry_g, carry_len, carry_start);
+                    carry_word = 0;
+                    carry_len = 0;
+                    carry_g = 0;
+                }
                 const u64 sm = wm & ~((wm << 1) | (u64)carry_word);
                 const u64 below_inc = (lane == 63)
                     ? ~0ULL : ((1ULL << (lane + 1)) - 1ULL);
@@ -329,46 +376,9 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                     && (last_valid == 63 || doc_continues))
                     at_end = false;    // token may continue: carry it
 
-                if (at_end && !(ablate & 1)) {
-                    const u64 key = g ? g : 1ULL;
-                    const u32 tl = (u32)(pos + 1);
-                    int fresh = lds_set_insert(set, key);
-                    if (fresh < 0) {
-                        // set overflow: global (doc,hash) seen fallback
-                        u64 sk = splitmix64(
-                            g ^ ((u64)d * 0x9E3779B97F4A7C15ULL));
-                        if (!sk) sk = 1;
-                        u64 slot = sk & fb_mask;
-                        fresh = 0;
-                        int probe = 0;
-                        while (true) {
-                            u64 prev = atomicCAS(&fb_seen[slot], 0ULL,
-                                                 sk);
-                            if (prev == 0ULL) { fresh = 1; break; }
-                            if (prev == sk) break;
-                            slot = (slot + 1) & fb_mask;
-                            if (++probe > FB_PROBE_CAP) {
-                                atomicOr(err_flag, 1u);
-                                break;
-                            }
-                        }
-                    }
-                    if (fresh == 1 && !(ablate & 2)) {
-                        block_count_add(cck, ccv, key, cnt_keys,
-                                        cnt_vals, cnt_mask);
-                        u64 slot;
-                        const long tstart = (s >= 0)
-                            ? (seg + wb + s) : carry_start;
-                        if (!(ablate & 4)
-                            && table_insert_u64(dict_keys, dict_mask,
-                                                key, &slot))
-                            dict_vals[slot] =
-                                ((pos_base + (u64)tstart) << 8)
-                                | (u64)min(tl, 255u);
-                    }
-                } else if (at_end) {   // ablation: consume the hash
-                    if (g == 0xdeadbeefdeadbeefULL) err_flag[1] = 1;
-                }
+                if (at_end)
+                    emit_token(g, (u32)(pos + 1),
+                               (s >= 0) ? (seg + wb + s) : carry_start);
 
                 // wave-uniform carry update from the tail lane
                 const int t_w = (int)((wm >> last_valid) & 1);
@@ -388,6 +398,11 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                 }
             }
             __builtin_amdgcn_wave_barrier();
+        }
+        if (carry_word) {
+            // doc ended exactly at a window edge with a live token
+            if (lane == 0)
+                emit_token(carry_g, carry_len, carry_start);
         }
     }
 

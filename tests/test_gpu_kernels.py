@@ -158,3 +158,23 @@ def test_tfidf_handles_irregular_text():
     got = run_tfidf(arr, device=DEV)
     want = oracle_df(arr)
     assert {t: v[0] for t, v in got.items()} == want
+
+
+def test_tfidf_window_edge_tokens():
+    # Regression: tokens ending exactly at a 64-byte window edge (and a
+    # doc ending at a window edge with a live token) must not drop.
+    from dampr_amd.gpu.corpus import oracle_df
+    from dampr_amd.gpu.tfidf import run_tfidf
+    w60 = "y" * 60                      # bytes 0..59
+    lines = []
+    # "zzz" occupies bytes 61..63: ends exactly at byte 63, space at 64
+    lines.append(w60 + " zzz qqq")
+    # doc whose length is exactly 128 with a word running to the end
+    doc = ("a" * 63 + " " + "b" * 64)   # 128 bytes, token ends at byte 127
+    lines.append(doc)
+    # the same zzz in a second doc (df must be 2)
+    lines.append("zzz tail")
+    text = ("\n".join(lines) + "\n").encode()
+    arr = np.frombuffer(text, dtype=np.uint8).copy()
+    got = {t: v[0] for t, v in run_tfidf(arr, device=DEV).items()}
+    assert got == oracle_df(arr)
