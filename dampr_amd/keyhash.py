@@ -31,16 +31,23 @@ def splitmix64(x: int) -> int:
     return x ^ (x >> 31)
 
 
+_TOKTAB_SEED = 0x7A0BDCAF
+
+
 def tokmix64(data: bytes) -> int:
-    """Order-aware XOR-of-mixes token hash; must match tokmix_step() in
-    ops/hip/common.h (the GPU computes it with a wave-wide segmented XOR
-    scan; XOR's associativity makes serial and parallel answers equal).
-    Bytes are lowercased like the device tokenizer."""
+    """Order-aware tabulation-rotate token hash; must match
+    tokmix_step()/tokmix_final() in ops/hip/common.h (the GPU computes it
+    with a wave-wide segmented XOR scan; XOR's associativity makes serial
+    and parallel answers equal).  Bytes are lowercased like the device
+    tokenizer; the length finalizer disambiguates >64-byte rotation
+    wraps."""
     h = 0
     for j, b in enumerate(data):
         c = b + 32 if 65 <= b <= 90 else b
-        h ^= splitmix64(((j << 8) | c) & _MASK)
-    return h
+        t = splitmix64(c + _TOKTAB_SEED)
+        r = j & 63
+        h ^= ((t << r) | (t >> (64 - r))) & _MASK if r else t
+    return h ^ splitmix64(len(data))
 
 
 def key_hash64(key) -> int:

@@ -26,13 +26,31 @@ __device__ __forceinline__ u64 splitmix64(u64 x) {
     return x ^ (x >> 31);
 }
 
-// Order-aware, *parallelizable* token hash: XOR over bytes of
-// splitmix64(pos << 8 | lower(byte)).  XOR is associative/commutative, so
-// a wave can compute it with a segmented scan while a serial loop gets the
-// identical value (mirrored in keyhash.tokmix64).  Positional mixing keeps
-// order sensitivity ("ab" != "ba").
+// Order-aware, *parallelizable* token hash (tabulation-rotate): XOR over
+// bytes of rotl64(TAB[lower(byte)], pos & 63), finalized with the token
+// length.  XOR's associativity lets a wave compute it with a segmented
+// scan while a serial loop gets the identical value (mirrored in
+// keyhash.tokmix64).  TAB[c] = splitmix64(c + TOKTAB_SEED) — kernels
+// materialize it in LDS once per block; serial paths compute entries on
+// the fly.  The rotation wraps at 64, so the length finalizer
+// disambiguates the (rare) >64-byte tokens' wrapped positions.
+#define TOKTAB_SEED 0x7A0BDCAFULL
+
+__device__ __forceinline__ u64 toktab_entry(u8 lowered) {
+    return splitmix64((u64)lowered + TOKTAB_SEED);
+}
+
+__device__ __forceinline__ u64 rotl64(u64 x, u32 r) {
+    r &= 63u;
+    return r ? ((x << r) | (x >> (64 - r))) : x;
+}
+
 __device__ __forceinline__ u64 tokmix_step(u64 h, u32 pos, u8 lowered) {
-    return h ^ splitmix64(((u64)pos << 8) | (u64)lowered);
+    return h ^ rotl64(toktab_entry(lowered), pos);
+}
+
+__device__ __forceinline__ u64 tokmix_final(u64 h, u32 len) {
+    return h ^ splitmix64((u64)len);
 }
 
 // ASCII '\w' classification with lowercasing, matching the reference's

@@ -98,13 +98,34 @@ __global__ void write_marks_kernel(const u8* __restrict__ text, long n,
 
     for (int it = 0; it < iters; ++it) {
         long off = base + (long)it * TILE + (long)tid * VBYTES;
-        // gather this lane's marks (positions within its 16 bytes)
+        // gather this lane's marks (positions within its 16 bytes);
+        // vectorized uint4 fast path like count_marks_kernel
         u32 cnt = 0;
         u8 rel[VBYTES];
         if (off < n) {
-            long lim = min((long)VBYTES, n - off);
-            for (long j = 0; j < lim; ++j)
-                if (mark_at(mode, text, off + j, n)) rel[cnt++] = (u8)j;
+            if (off + VBYTES <= n) {
+                uint4 v = *reinterpret_cast<const uint4*>(text + off);
+                const u8* b = reinterpret_cast<const u8*>(&v);
+                #pragma unroll
+                for (int j = 0; j < VBYTES; ++j) {
+                    u8 c = b[j];
+                    bool m;
+                    if (mode == MODE_NEWLINE) {
+                        m = (c == '\n');
+                    } else {
+                        bool prev_word = (j > 0)
+                            ? is_word(b[j - 1])
+                            : (off > 0 ? is_word(text[off - 1]) : false);
+                        m = is_word(c) && !prev_word;
+                    }
+                    if (m) rel[cnt++] = (u8)j;
+                }
+            } else {
+                long lim = n - off;
+                for (long j = 0; j < lim; ++j)
+                    if (mark_at(mode, text, off + j, n))
+                        rel[cnt++] = (u8)j;
+            }
         }
         // block exclusive scan of cnt
         u32 scan = cnt;
@@ -159,6 +180,7 @@ __global__ void tfidf_count_kernel(
             ++j;
         }
         u32 len = (u32)(p - start);
+        h = tokmix_final(h, len);
         // doc id = number of newlines strictly before `start`
         long lo = 0, hi = n_nl;
         while (lo < hi) {
@@ -203,7 +225,7 @@ __device__ __forceinline__ void block_count_add(
         u64* __restrict__ cck, u32* __restrict__ ccv, u64 key,
         u64* __restrict__ gk, u64* __restrict__ gv, u64 gmask) {
     u32 slot = (u32)(key & (CCACHE - 1));
-    for (int probe = 0; probe < 16; ++probe) {
+    for (int probe = 0; probe < 4; ++probe) {
         u64 cur = cck[slot];
         if (cur == key) { atomicAdd(&ccv[slot], 1u); return; }
         if (cur == 0ULL) {
@@ -251,6 +273,7 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
     __shared__ u64 dset[DOC_WAVES][DOC_SET];
     __shared__ u64 cck[CCACHE];
     __shared__ u32 ccv[CCACHE];
+    __shared__ u64 toktab[256];
     const int wid = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
     const long gwave = (long)blockIdx.x * DOC_WAVES + wid;
@@ -262,6 +285,8 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
         cck[i] = 0;
         ccv[i] = 0;
     }
+    for (int i = threadIdx.x; i < 256; i += blockDim.x)
+        toktab[i] = toktab_entry((u8)i);
     __syncthreads();
 
     for (long d = gwave; d < n_docs; d += nwaves) {
@@ -280,11 +305,12 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                 if (gh == 0xdeadbeefdeadbeefULL) err_flag[1] = 1;
                 return;
             }
-            const u64 key = gh ? gh : 1ULL;
+            const u64 hh = tokmix_final(gh, tl);
+            const u64 key = hh ? hh : 1ULL;
             int fresh = lds_set_insert(set, key);
             if (fresh < 0) {
                 // set overflow: global (doc,hash) seen fallback
-                u64 sk = splitmix64(gh ^ ((u64)d
+                u64 sk = splitmix64(hh ^ ((u64)d
                                           * 0x9E3779B97F4A7C15ULL));
                 if (!sk) sk = 1;
                 u64 slot = sk & fb_mask;
@@ -358,14 +384,23 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                 if (w) {
                     s = sm_le ? (63 - __clzll(sm_le)) : -(int)carry_len;
                     pos = lane - s;
-                    g = splitmix64(((u64)(u32)pos << 8)
-                                   | (u64)lower_ascii(c));
+                    g = rotl64(toktab[lower_ascii(c)], (u32)pos);
                 }
-                // segmented inclusive XOR scan along the wave
+                // segmented inclusive XOR scan along the wave; tokens
+                // shorter than 8 bytes (the common case) need only the
+                // first three steps
                 #pragma unroll
-                for (int dsh = 1; dsh < WAVE; dsh <<= 1) {
+                for (int dsh = 1; dsh <= 4; dsh <<= 1) {
                     const u64 g2 = __shfl_up(g, dsh, WAVE);
                     if (w && lane >= dsh && (lane - dsh) >= s) g ^= g2;
+                }
+                if (__ballot(w && pos >= 8)) {
+                    #pragma unroll
+                    for (int dsh = 8; dsh < WAVE; dsh <<= 1) {
+                        const u64 g2 = __shfl_up(g, dsh, WAVE);
+                        if (w && lane >= dsh && (lane - dsh) >= s)
+                            g ^= g2;
+                    }
                 }
                 if (w && s < 0) g ^= carry_g;   // continuing token prefix
 
