@@ -20,6 +20,22 @@ from .keyhash import partition_of
 
 
 class Splitter(object):
+    """Partition assignment for the host engine's shuffle.
+
+    Uses the builtin ``hash`` (C speed): workers are *forked*, so the
+    interpreter's per-process hash salt is shared and placement is
+    consistent within a run — the same property the reference relies on
+    (reference: base.py:6-8).  The device engine never sees these
+    partitions; where CPU/GPU-stable keys are required (string dict,
+    cross-rank exchange) the canonical ``keyhash`` functions are used
+    instead.  ``StableSplitter`` is the opt-in device-reproducible form.
+    """
+
+    def partition(self, key, n_partitions):
+        return hash(key) % n_partitions
+
+
+class StableSplitter(object):
     def partition(self, key, n_partitions):
         return partition_of(key, n_partitions)
 

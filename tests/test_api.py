@@ -467,3 +467,17 @@ def test_uncompressed_spill(items):
 def test_inspect_passthrough(items, capsys):
     res = items.inspect("dbg").read()
     assert res == list(range(10, 20))
+
+
+def test_indexer(tmp_path):
+    from dampr_amd.utils.indexer import Indexer
+    f = tmp_path / "data.txt"
+    f.write_text("apple red\nbanana yellow\ncherry red\nkiwi green\n")
+    idx = Indexer(str(f))
+    idx.build(lambda line: line.split())
+    red = sorted(idx.union(["red"]).read())
+    assert red == ["apple red\n", "cherry red\n"]
+    both = idx.intersect(["red", "apple"]).read()
+    assert both == ["apple red\n"]
+    either = sorted(idx.intersect(["red", "banana"], min_match=1).read())
+    assert either == ["apple red\n", "banana yellow\n", "cherry red\n"]
