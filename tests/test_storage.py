@@ -70,7 +70,7 @@ def test_partitioned_spiller_sorted_within_partition(fs):
         for run in runs:
             ks = [k for k, _ in run.read()]
             assert ks == sorted(ks)
-            assert all(partition_of(k, 7) == p for k in ks)
+            assert all(hash(k) % 7 == p for k in ks)
             seen.extend(ks)
     assert sorted(seen) == sorted(keys)
 
