@@ -215,6 +215,7 @@ __global__ void tfidf_count_kernel(
 #define SEG_OVERLAP 272              // > max dict token length (255)
 #define DOC_SET 256                  // per-wave dedupe set slots (pow2)
 #define FB_PROBE_CAP 512
+#define DOC_BLK 64                   // contiguous docs per wave block
 #define CCACHE 1024                  // block-level LDS count cache slots
 
 // Two-level counting: Zipf-hot keys would serialize ~50M same-address L2
@@ -289,9 +290,24 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
         toktab[i] = toktab_entry((u8)i);
     __syncthreads();
 
-    for (long d = gwave; d < n_docs; d += nwaves) {
-        const long ls = d ? (long)nl_pos[d - 1] + 1 : 0;
-        const long le = (d < n_nl) ? (long)nl_pos[d] : n;
+    // Contiguous blocks of DOC_BLK docs per wave: one coalesced nl_pos
+    // vector load supplies 64 doc boundaries (shfl per doc), and one
+    // staged LDS window covers many consecutive short docs — the per-doc
+    // global-latency chain of the naive one-doc-per-iteration layout
+    // amortizes ~20x for ~100 B documents.
+    for (long dbase = gwave * DOC_BLK; dbase < n_docs;
+         dbase += nwaves * DOC_BLK) {
+        const long dlim = min(dbase + (long)DOC_BLK, n_docs);
+        const long di = dbase + lane;
+        const u32 nl_lane = (di < n_nl) ? nl_pos[di] : (u32)n;
+        long prev_le = dbase ? (long)nl_pos[dbase - 1] : -1;
+        long win_lo = -1, win_hi = -1, aseg = 0;
+
+      for (long d = dbase; d < dlim; ++d) {
+        const long ls = prev_le + 1;
+        const long le =
+            (long)(u32)__shfl((int)nl_lane, (int)(d - dbase), WAVE);
+        prev_le = le;
         for (int s = lane; s < DOC_SET; s += WAVE) set[s] = 0;
 
         u32 carry_word = 0;            // wave-uniform token carry
@@ -339,24 +355,34 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
             }
         };
 
-        for (long seg = ls; seg < le; seg += STAGE_B) {
-            const long seg_end = min(seg + (long)STAGE_B, le);
-            const int seg_len = (int)(seg_end - seg);
-            const long aseg = seg & ~15L;
-            const int soff = (int)(seg - aseg);
-            const int stage_bytes = soff + seg_len;
-            for (int i = lane * 16; i < stage_bytes; i += WAVE * 16) {
-                if (aseg + i + 16 <= n) {
-                    *reinterpret_cast<uint4*>(st + i) =
-                        *reinterpret_cast<const uint4*>(text + aseg + i);
-                } else {
-                    for (int j = i; j < stage_bytes; ++j)
-                        st[j] = (aseg + j < n) ? text[aseg + j] : (u8)0;
+        for (long seg = ls; seg < le; ) {
+            if (seg < win_lo || seg >= win_hi) {
+                // (re)stage an aligned window starting at seg; bytes
+                // beyond this doc belong to following docs and are
+                // reused without restaging
+                aseg = seg & ~15L;
+                const int stage_bytes =
+                    (int)min((long)(STAGE_B + 16), n - aseg);
+                for (int i = lane * 16; i < stage_bytes; i += WAVE * 16) {
+                    if (aseg + i + 16 <= n) {
+                        *reinterpret_cast<uint4*>(st + i) =
+                            *reinterpret_cast<const uint4*>(
+                                text + aseg + i);
+                    } else {
+                        for (int j = i; j < stage_bytes; ++j)
+                            st[j] = (aseg + j < n) ? text[aseg + j]
+                                                   : (u8)0;
+                    }
                 }
+                asm volatile(
+                    "s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+                __builtin_amdgcn_wave_barrier();
+                win_lo = seg;
+                win_hi = aseg + stage_bytes;
             }
-            asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
-            __builtin_amdgcn_wave_barrier();
-
+            const long seg_end = min(le, win_hi);
+            const int seg_len = (int)(seg_end - seg);
+            const int soff = (int)(seg - aseg);
             const bool doc_continues = seg_end < le;
             const int nwin = (seg_len + WAVE - 1) / WAVE;
             for (int wdx = 0; wdx < nwin; ++wdx) {
@@ -433,12 +459,14 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                 }
             }
             __builtin_amdgcn_wave_barrier();
+            seg = seg_end;
         }
         if (carry_word) {
             // doc ended exactly at a window edge with a live token
             if (lane == 0)
                 emit_token(carry_g, carry_len, carry_start);
         }
+      }
     }
 
     // flush the block's count cache
@@ -690,7 +718,7 @@ long tfidf_count_docs(torch::Tensor text, torch::Tensor nl_pos,
                       long ablate) {
     check_u8(text);
     if (n_docs == 0) return 0;
-    long waves_needed = n_docs;
+    long waves_needed = (n_docs + 63) / 64;     // DOC_BLK docs per wave
     long blocks = std::min<long>((waves_needed + DOC_WAVES - 1) / DOC_WAVES,
                                  8192);
     hipLaunchKernelGGL(tfidf_docs_kernel, dim3((u32)blocks),
