@@ -216,6 +216,7 @@ __global__ void tfidf_count_kernel(
 #define DOC_SET 256                  // per-wave dedupe set slots (pow2)
 #define FB_PROBE_CAP 512
 #define DOC_BLK 64                   // contiguous docs per wave block
+#define GROUP_BYTES 1024             // target bytes per doc group
 #define CCACHE 1024                  // block-level LDS count cache slots
 
 // Two-level counting: Zipf-hot keys would serialize ~50M same-address L2
@@ -262,14 +263,17 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
                   u64 dict_mask, u64 pos_base,
                   u64* __restrict__ fb_seen, u64 fb_mask,
                   u32* __restrict__ err_flag, u32 ablate) {
-    // One wave per document; all 64 lanes active on every byte:
-    //  * 16 B/lane uint4 staging into LDS (aligned; no overlap restaging)
-    //  * 64-B windows, one BYTE per lane: ballot-derived token starts,
-    //    wave-wide segmented XOR scan builds every token's hash in
-    //    parallel (tokmix, common.h) — no serial per-token walks
-    //  * tokens crossing window/segment boundaries ride a wave-uniform
-    //    carry (partial hash, length, start), so arbitrarily long tokens
-    //    hash exactly with zero re-reads
+    // v3: group-of-docs stream processing, all 64 lanes on every byte.
+    //  * a wave owns DOC_BLK contiguous docs; one coalesced nl_pos vector
+    //    load gives all their boundaries (shfl per query)
+    //  * docs are processed in *groups* (~GROUP_BYTES of consecutive
+    //    docs): windows run continuously across doc boundaries, so short
+    //    docs no longer waste partial windows or per-doc framing
+    //  * doc identity inside a window comes from the newline ballot;
+    //    dedupe keys are salted with the doc id, so the per-wave LDS set
+    //    only needs clearing once per group (always at a doc boundary)
+    //  * token hashes by wave-wide segmented XOR scan (tokmix); tokens
+    //    crossing window/stage edges ride a wave-uniform carry
     __shared__ __align__(16) u8 stage[DOC_WAVES][STAGE_B + 16];
     __shared__ u64 dset[DOC_WAVES][DOC_SET];
     __shared__ u64 cck[CCACHE];
@@ -290,183 +294,211 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
         toktab[i] = toktab_entry((u8)i);
     __syncthreads();
 
-    // Contiguous blocks of DOC_BLK docs per wave: one coalesced nl_pos
-    // vector load supplies 64 doc boundaries (shfl per doc), and one
-    // staged LDS window covers many consecutive short docs — the per-doc
-    // global-latency chain of the naive one-doc-per-iteration layout
-    // amortizes ~20x for ~100 B documents.
     for (long dbase = gwave * DOC_BLK; dbase < n_docs;
          dbase += nwaves * DOC_BLK) {
         const long dlim = min(dbase + (long)DOC_BLK, n_docs);
         const long di = dbase + lane;
+        // lane i holds the END byte (newline pos, or n) of doc dbase+i
         const u32 nl_lane = (di < n_nl) ? nl_pos[di] : (u32)n;
-        long prev_le = dbase ? (long)nl_pos[dbase - 1] : -1;
         long win_lo = -1, win_hi = -1, aseg = 0;
+        long gd = dbase;                       // first doc of the group
+        long gs = dbase ? (long)nl_pos[dbase - 1] + 1 : 0;
 
-      for (long d = dbase; d < dlim; ++d) {
-        const long ls = prev_le + 1;
-        const long le =
-            (long)(u32)__shfl((int)nl_lane, (int)(d - dbase), WAVE);
-        prev_le = le;
-        for (int s = lane; s < DOC_SET; s += WAVE) set[s] = 0;
-
-        u32 carry_word = 0;            // wave-uniform token carry
-        u32 carry_len = 0;
-        u64 carry_g = 0;
-        long carry_start = ls;
-
-        // dedupe + count + dict insert for one finished token
-        auto emit_token = [&](u64 gh, u32 tl, long tstart) {
-            if (ablate & 1) {          // ablation: consume the hash
-                if (gh == 0xdeadbeefdeadbeefULL) err_flag[1] = 1;
-                return;
+        while (gd < dlim) {
+            // group = docs [gd, ge): consecutive docs spanning at most
+            // GROUP_BYTES (a longer single doc forms its own group).
+            // ends[lane] = end byte of doc dbase+lane (exclusive \n).
+            const u64 fit = __ballot(
+                di < dlim && di >= gd
+                && (long)nl_lane <= gs + (long)GROUP_BYTES);
+            long ge;
+            if (fit) {
+                ge = dbase + 63 - __clzll(fit) + 1;
+            } else {
+                ge = gd + 1;                   // oversized doc: alone
             }
-            const u64 hh = tokmix_final(gh, tl);
-            const u64 key = hh ? hh : 1ULL;
-            int fresh = lds_set_insert(set, key);
-            if (fresh < 0) {
-                // set overflow: global (doc,hash) seen fallback
-                u64 sk = splitmix64(hh ^ ((u64)d
-                                          * 0x9E3779B97F4A7C15ULL));
-                if (!sk) sk = 1;
-                u64 slot = sk & fb_mask;
-                fresh = 0;
-                int probe = 0;
-                while (true) {
-                    u64 prev = atomicCAS(&fb_seen[slot], 0ULL, sk);
-                    if (prev == 0ULL) { fresh = 1; break; }
-                    if (prev == sk) break;
-                    slot = (slot + 1) & fb_mask;
-                    if (++probe > FB_PROBE_CAP) {
-                        atomicOr(err_flag, 1u);
-                        break;
+            const long gend = (ge - 1 < n_nl)
+                ? (long)(u32)__shfl((int)nl_lane, (int)(ge - 1 - dbase),
+                                    WAVE)
+                : n;
+
+            for (int s = lane; s < DOC_SET; s += WAVE) set[s] = 0;
+            u32 carry_word = 0;                // wave-uniform token carry
+            u32 carry_len = 0;
+            u64 carry_g = 0;
+            long carry_start = gs;
+            u32 carry_lines = 0;               // newlines seen in group
+
+            // dedupe + count + dict insert for one finished token
+            auto emit_token = [&](u64 gh, u32 tl, long tstart,
+                                  long doc_id) {
+                if (ablate & 1) {              // ablation: consume hash
+                    if (gh == 0xdeadbeefdeadbeefULL) err_flag[1] = 1;
+                    return;
+                }
+                const u64 hh = tokmix_final(gh, tl);
+                const u64 key = hh ? hh : 1ULL;
+                u64 dk = hh ^ splitmix64((u64)doc_id + 0x5bd1e995ULL);
+                if (!dk) dk = 1;
+                int fresh = lds_set_insert(set, dk);
+                if (fresh < 0) {
+                    // set overflow: global (doc,hash) seen fallback
+                    u64 sk = splitmix64(hh ^ ((u64)doc_id
+                                              * 0x9E3779B97F4A7C15ULL));
+                    if (!sk) sk = 1;
+                    u64 slot = sk & fb_mask;
+                    fresh = 0;
+                    int probe = 0;
+                    while (true) {
+                        u64 prev = atomicCAS(&fb_seen[slot], 0ULL, sk);
+                        if (prev == 0ULL) { fresh = 1; break; }
+                        if (prev == sk) break;
+                        slot = (slot + 1) & fb_mask;
+                        if (++probe > FB_PROBE_CAP) {
+                            atomicOr(err_flag, 1u);
+                            break;
+                        }
                     }
                 }
-            }
-            if (fresh == 1 && !(ablate & 2)) {
-                block_count_add(cck, ccv, key, cnt_keys, cnt_vals,
-                                cnt_mask);
-                u64 slot;
-                if (!(ablate & 4)
-                    && table_insert_u64(dict_keys, dict_mask, key,
-                                        &slot))
-                    dict_vals[slot] = ((pos_base + (u64)tstart) << 8)
-                                      | (u64)min(tl, 255u);
-            }
-        };
+                if (fresh == 1 && !(ablate & 2)) {
+                    block_count_add(cck, ccv, key, cnt_keys, cnt_vals,
+                                    cnt_mask);
+                    u64 slot;
+                    if (!(ablate & 4)
+                        && table_insert_u64(dict_keys, dict_mask, key,
+                                            &slot))
+                        dict_vals[slot] =
+                            ((pos_base + (u64)tstart) << 8)
+                            | (u64)min(tl, 255u);
+                }
+            };
 
-        for (long seg = ls; seg < le; ) {
-            if (seg < win_lo || seg >= win_hi) {
-                // (re)stage an aligned window starting at seg; bytes
-                // beyond this doc belong to following docs and are
-                // reused without restaging
-                aseg = seg & ~15L;
-                const int stage_bytes =
-                    (int)min((long)(STAGE_B + 16), n - aseg);
-                for (int i = lane * 16; i < stage_bytes; i += WAVE * 16) {
-                    if (aseg + i + 16 <= n) {
-                        *reinterpret_cast<uint4*>(st + i) =
-                            *reinterpret_cast<const uint4*>(
-                                text + aseg + i);
-                    } else {
-                        for (int j = i; j < stage_bytes; ++j)
-                            st[j] = (aseg + j < n) ? text[aseg + j]
-                                                   : (u8)0;
+            for (long seg = gs; seg < gend; ) {
+                if (seg < win_lo || seg >= win_hi) {
+                    // (re)stage an aligned window from seg; bytes beyond
+                    // this group belong to following docs and are reused
+                    aseg = seg & ~15L;
+                    const int stage_bytes =
+                        (int)min((long)(STAGE_B + 16), n - aseg);
+                    for (int i = lane * 16; i < stage_bytes;
+                         i += WAVE * 16) {
+                        if (aseg + i + 16 <= n) {
+                            *reinterpret_cast<uint4*>(st + i) =
+                                *reinterpret_cast<const uint4*>(
+                                    text + aseg + i);
+                        } else {
+                            for (int j = i; j < stage_bytes; ++j)
+                                st[j] = (aseg + j < n) ? text[aseg + j]
+                                                       : (u8)0;
+                        }
                     }
+                    asm volatile(
+                        "s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+                    __builtin_amdgcn_wave_barrier();
+                    win_lo = seg;
+                    win_hi = aseg + stage_bytes;
                 }
-                asm volatile(
-                    "s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
-                __builtin_amdgcn_wave_barrier();
-                win_lo = seg;
-                win_hi = aseg + stage_bytes;
-            }
-            const long seg_end = min(le, win_hi);
-            const int seg_len = (int)(seg_end - seg);
-            const int soff = (int)(seg - aseg);
-            const bool doc_continues = seg_end < le;
-            const int nwin = (seg_len + WAVE - 1) / WAVE;
-            for (int wdx = 0; wdx < nwin; ++wdx) {
-                const int wb = wdx * WAVE;
-                const int p = wb + lane;
-                const bool valid = p < seg_len;
-                const u8 c = valid ? st[soff + p] : (u8)0;
-                const bool w = valid && is_word(c);
-                const u64 wm = __ballot(w);
-                if (carry_word && !(wm & 1ULL)) {
-                    // carried token ended exactly at the window edge
-                    if (lane == 0)
-                        emit_token(carry_g, carry_len, carry_start);
-                    carry_word = 0;
-                    carry_len = 0;
-                    carry_g = 0;
-                }
-                const u64 sm = wm & ~((wm << 1) | (u64)carry_word);
-                const u64 below_inc = (lane == 63)
-                    ? ~0ULL : ((1ULL << (lane + 1)) - 1ULL);
-                const u64 sm_le = sm & below_inc;
-                int s = WAVE;          // token start (window coords)
-                int pos = 0;           // byte position within token
-                u64 g = 0;
-                if (w) {
-                    s = sm_le ? (63 - __clzll(sm_le)) : -(int)carry_len;
-                    pos = lane - s;
-                    g = rotl64(toktab[lower_ascii(c)], (u32)pos);
-                }
-                // segmented inclusive XOR scan along the wave; tokens
-                // shorter than 8 bytes (the common case) need only the
-                // first three steps
-                #pragma unroll
-                for (int dsh = 1; dsh <= 4; dsh <<= 1) {
-                    const u64 g2 = __shfl_up(g, dsh, WAVE);
-                    if (w && lane >= dsh && (lane - dsh) >= s) g ^= g2;
-                }
-                if (__ballot(w && pos >= 8)) {
+                const long seg_end = min(gend, win_hi);
+                const int seg_len = (int)(seg_end - seg);
+                const int soff = (int)(seg - aseg);
+                const bool group_continues = seg_end < gend;
+                const int nwin = (seg_len + WAVE - 1) / WAVE;
+                for (int wdx = 0; wdx < nwin; ++wdx) {
+                    const int wb = wdx * WAVE;
+                    const int p = wb + lane;
+                    const bool valid = p < seg_len;
+                    const u8 c = valid ? st[soff + p] : (u8)0;
+                    const bool w = valid && is_word(c);
+                    const u64 wm = __ballot(w);
+                    const u64 nlm = __ballot(valid && c == '\n');
+                    if (carry_word && !(wm & 1ULL)) {
+                        // carried token ended exactly at the window edge
+                        if (lane == 0)
+                            emit_token(carry_g, carry_len, carry_start,
+                                       gd + carry_lines);
+                        carry_word = 0;
+                        carry_len = 0;
+                        carry_g = 0;
+                    }
+                    const u64 sm = wm & ~((wm << 1) | (u64)carry_word);
+                    const u64 below_inc = (lane == 63)
+                        ? ~0ULL : ((1ULL << (lane + 1)) - 1ULL);
+                    const u64 sm_le = sm & below_inc;
+                    int s = WAVE;          // token start (window coords)
+                    int pos = 0;           // byte position within token
+                    u64 g = 0;
+                    if (w) {
+                        s = sm_le ? (63 - __clzll(sm_le))
+                                  : -(int)carry_len;
+                        pos = lane - s;
+                        g = rotl64(toktab[lower_ascii(c)], (u32)pos);
+                    }
+                    // segmented inclusive XOR scan; <8-byte tokens (the
+                    // common case) need only the first three steps
                     #pragma unroll
-                    for (int dsh = 8; dsh < WAVE; dsh <<= 1) {
+                    for (int dsh = 1; dsh <= 4; dsh <<= 1) {
                         const u64 g2 = __shfl_up(g, dsh, WAVE);
                         if (w && lane >= dsh && (lane - dsh) >= s)
                             g ^= g2;
                     }
+                    if (__ballot(w && pos >= 8)) {
+                        #pragma unroll
+                        for (int dsh = 8; dsh < WAVE; dsh <<= 1) {
+                            const u64 g2 = __shfl_up(g, dsh, WAVE);
+                            if (w && lane >= dsh && (lane - dsh) >= s)
+                                g ^= g2;
+                        }
+                    }
+                    if (w && s < 0) g ^= carry_g;  // continuing prefix
+
+                    const int last_valid = min(seg_len - wb, WAVE) - 1;
+                    bool at_end = w
+                        && (lane == 63 ? true
+                                       : !((wm >> (lane + 1)) & 1));
+                    if (lane == last_valid && w
+                        && (last_valid == 63 || group_continues))
+                        at_end = false;    // may continue: carry it
+
+                    if (at_end) {
+                        const u64 below = (1ULL << lane) - 1ULL;
+                        const long doc_id = gd + carry_lines
+                            + __popcll(nlm & below);
+                        emit_token(g, (u32)(pos + 1),
+                                   (s >= 0) ? (seg + wb + s)
+                                            : carry_start, doc_id);
+                    }
+
+                    // wave-uniform carry update from the tail lane
+                    const int t_w = (int)((wm >> last_valid) & 1);
+                    const int t_end = __shfl((int)at_end, last_valid,
+                                             WAVE);
+                    if (t_w && !t_end) {
+                        const int t_pos = __shfl(pos, last_valid, WAVE);
+                        const int t_s = __shfl(s, last_valid, WAVE);
+                        const u64 t_g = __shfl(g, last_valid, WAVE);
+                        carry_word = 1;
+                        carry_len = (u32)(t_pos + 1);
+                        carry_g = t_g;
+                        if (t_s >= 0) carry_start = seg + wb + t_s;
+                    } else {
+                        carry_word = 0;
+                        carry_len = 0;
+                        carry_g = 0;
+                    }
+                    carry_lines += (u32)__popcll(nlm);
                 }
-                if (w && s < 0) g ^= carry_g;   // continuing token prefix
-
-                const int last_valid = min(seg_len - wb, WAVE) - 1;
-                bool at_end = w
-                    && (lane == 63 ? true : !((wm >> (lane + 1)) & 1));
-                if (lane == last_valid && w
-                    && (last_valid == 63 || doc_continues))
-                    at_end = false;    // token may continue: carry it
-
-                if (at_end)
-                    emit_token(g, (u32)(pos + 1),
-                               (s >= 0) ? (seg + wb + s) : carry_start);
-
-                // wave-uniform carry update from the tail lane
-                const int t_w = (int)((wm >> last_valid) & 1);
-                const int t_end = __shfl((int)at_end, last_valid, WAVE);
-                if (t_w && !t_end) {
-                    const int t_pos = __shfl(pos, last_valid, WAVE);
-                    const int t_s = __shfl(s, last_valid, WAVE);
-                    const u64 t_g = __shfl(g, last_valid, WAVE);
-                    carry_word = 1;
-                    carry_len = (u32)(t_pos + 1);
-                    carry_g = t_g;
-                    if (t_s >= 0) carry_start = seg + wb + t_s;
-                } else {
-                    carry_word = 0;
-                    carry_len = 0;
-                    carry_g = 0;
-                }
+                __builtin_amdgcn_wave_barrier();
+                seg = seg_end;
             }
-            __builtin_amdgcn_wave_barrier();
-            seg = seg_end;
+            if (carry_word) {
+                // group ended at a window edge with a live token
+                if (lane == 0)
+                    emit_token(carry_g, carry_len, carry_start,
+                               gd + carry_lines);
+            }
+            gd = ge;
+            gs = gend + 1;                     // past the newline
         }
-        if (carry_word) {
-            // doc ended exactly at a window edge with a live token
-            if (lane == 0)
-                emit_token(carry_g, carry_len, carry_start);
-        }
-      }
     }
 
     // flush the block's count cache
