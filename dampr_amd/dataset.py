@@ -321,6 +321,21 @@ class Spiller(DatasetWriter):
         if not self.memory and self.governor.over_watermark():
             self.flush()
 
+    def add_records(self, it):
+        """Bulk add: batches of records with one watermark check per
+        batch — the per-record method-call overhead dominates pure-Python
+        hot loops otherwise."""
+        from itertools import islice
+        buf_extend = None
+        while True:
+            batch = list(islice(it, 16384))
+            if not batch:
+                return
+            self.buf.extend(batch)
+            if not self.memory and \
+                    self.governor.over_watermark_bulk(len(batch)):
+                self.flush()
+
     def flush(self):
         if not self.buf:
             return
@@ -359,6 +374,17 @@ class PartitionedSpiller(DatasetWriter):
         self.buf.append((key, value))
         if not self.memory and self.governor.over_watermark():
             self.flush()
+
+    def add_records(self, it):
+        from itertools import islice
+        while True:
+            batch = list(islice(it, 16384))
+            if not batch:
+                return
+            self.buf.extend(batch)
+            if not self.memory and \
+                    self.governor.over_watermark_bulk(len(batch)):
+                self.flush()
 
     def flush(self):
         if not self.buf:
@@ -413,6 +439,27 @@ class CombineSpiller(DatasetWriter):
                 return
         if not self.memory and self.governor.over_watermark():
             self.flush()
+
+    def add_records(self, it):
+        """Bulk add: tight upsert loop over record batches, one
+        watermark / max-keys check per batch."""
+        from itertools import islice
+        binop = self.binop
+        while True:
+            batch = list(islice(it, 16384))
+            if not batch:
+                return
+            t = self.table
+            for key, value in batch:
+                if key in t:
+                    t[key] = binop(t[key], value)
+                else:
+                    t[key] = value
+            if self.max_keys is not None and len(t) >= self.max_keys:
+                self.flush()
+            elif not self.memory and \
+                    self.governor.over_watermark_bulk(len(batch)):
+                self.flush()
 
     def flush(self):
         if not self.table:

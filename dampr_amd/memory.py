@@ -40,6 +40,11 @@ class MemoryGovernor(object):
         self._last_count = 0
         self._last_mb = self.base_mb
 
+    def over_watermark_bulk(self, n) -> bool:
+        """Amortized form: account n records at once."""
+        self._count += n - 1
+        return self.over_watermark()
+
     def over_watermark(self) -> bool:
         """Call once per record added; True when the buffer should spill."""
         self._count += 1

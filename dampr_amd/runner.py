@@ -173,8 +173,7 @@ class _MapWorker(StageWorker):
 
     def process(self, job):
         _t_id, chunk, supplemental = job
-        for k, v in self.mapper.map(chunk, *supplemental):
-            self.writer.add_record(k, v)
+        self.writer.add_records(self.mapper.map(chunk, *supplemental))
 
     def finish(self):
         return self.writer.finished()
@@ -205,8 +204,7 @@ class _CombineMapWorker(StageWorker):
 
     def process(self, job):
         _t_id, chunk, supplemental = job
-        for k, v in self.mapper.map(chunk, *supplemental):
-            self.writer.add_record(k, v)
+        self.writer.add_records(self.mapper.map(chunk, *supplemental))
 
     def finish(self):
         runs = self.writer.finished()[0]
