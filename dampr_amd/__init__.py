@@ -17,9 +17,9 @@ from .dampr import Dampr, PMap, PReduce, PJoin, ARReduce, ValueEmitter
 from .base import BlockMapper, BlockReducer, Map, StreamMapper, \
     StreamReducer, Reduce
 from .dataset import Dataset, Chunker
-from . import settings
+from . import funcs, settings
 
-__all__ = ["Dampr", "PMap", "PReduce", "PJoin", "ARReduce", "ValueEmitter",
+__all__ = ["Dampr", "PMap", "PReduce", "PJoin", "ARReduce", "ValueEmitter", "funcs",
            "BlockMapper", "BlockReducer", "Dataset", "Chunker", "settings",
            "setup_logging"]
 

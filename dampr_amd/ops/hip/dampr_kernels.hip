@@ -81,9 +81,43 @@ __global__ void count_marks_kernel(const u8* __restrict__ text, long n,
 }
 
 // ---------------------------------------------------------------- scan pass 2
-// Re-scan and write mark positions in ascending order, starting at the
-// block's exclusive offset.  In-block order: per-iteration block-wide
-// exclusive scan of per-lane counts (wave shfl scan + LDS across waves).
+// Re-scan and write mark positions in ascending order.  Two-phase within
+// the block: per-(wave, tile) counts into LDS, ONE barrier, then a
+// sync-free scatter pass where every wave derives its own bases from the
+// count matrix (the old per-tile cursor + double barrier serialized the
+// block on memory latency every 4 KiB).
+#define SCAN_MAX_ITERS 16
+
+__device__ __forceinline__ u32 gather_marks(const u8* __restrict__ text,
+                                            long n, int mode, long off,
+                                            u8* rel) {
+    u32 cnt = 0;
+    if (off >= n) return 0;
+    if (off + VBYTES <= n) {
+        uint4 v = *reinterpret_cast<const uint4*>(text + off);
+        const u8* b = reinterpret_cast<const u8*>(&v);
+        #pragma unroll
+        for (int j = 0; j < VBYTES; ++j) {
+            u8 c = b[j];
+            bool m;
+            if (mode == MODE_NEWLINE) {
+                m = (c == '\n');
+            } else {
+                bool prev_word = (j > 0)
+                    ? is_word(b[j - 1])
+                    : (off > 0 ? is_word(text[off - 1]) : false);
+                m = is_word(c) && !prev_word;
+            }
+            if (m) rel[cnt++] = (u8)j;
+        }
+    } else {
+        long lim = n - off;
+        for (long j = 0; j < lim; ++j)
+            if (mark_at(mode, text, off + j, n)) rel[cnt++] = (u8)j;
+    }
+    return cnt;
+}
+
 __global__ void write_marks_kernel(const u8* __restrict__ text, long n,
                                    int mode, int iters,
                                    const u32* __restrict__ block_offsets,
@@ -91,62 +125,41 @@ __global__ void write_marks_kernel(const u8* __restrict__ text, long n,
     long base = (long)blockIdx.x * iters * TILE;
     int tid = threadIdx.x;
     int lane = tid & (WAVE - 1), wid = tid / WAVE;
-    __shared__ u32 cursor;
-    __shared__ u32 wtot[BLOCK / WAVE];
-    if (tid == 0) cursor = block_offsets[blockIdx.x];
-    __syncthreads();
+    __shared__ u32 wtile[BLOCK / WAVE][SCAN_MAX_ITERS];
 
+    // phase 1: per-(wave, tile) mark counts
     for (int it = 0; it < iters; ++it) {
         long off = base + (long)it * TILE + (long)tid * VBYTES;
-        // gather this lane's marks (positions within its 16 bytes);
-        // vectorized uint4 fast path like count_marks_kernel
-        u32 cnt = 0;
         u8 rel[VBYTES];
-        if (off < n) {
-            if (off + VBYTES <= n) {
-                uint4 v = *reinterpret_cast<const uint4*>(text + off);
-                const u8* b = reinterpret_cast<const u8*>(&v);
-                #pragma unroll
-                for (int j = 0; j < VBYTES; ++j) {
-                    u8 c = b[j];
-                    bool m;
-                    if (mode == MODE_NEWLINE) {
-                        m = (c == '\n');
-                    } else {
-                        bool prev_word = (j > 0)
-                            ? is_word(b[j - 1])
-                            : (off > 0 ? is_word(text[off - 1]) : false);
-                        m = is_word(c) && !prev_word;
-                    }
-                    if (m) rel[cnt++] = (u8)j;
-                }
-            } else {
-                long lim = n - off;
-                for (long j = 0; j < lim; ++j)
-                    if (mark_at(mode, text, off + j, n))
-                        rel[cnt++] = (u8)j;
-            }
+        u32 cnt = gather_marks(text, n, mode, off, rel);
+        u32 tot = cnt;
+        for (int d = WAVE / 2; d > 0; d >>= 1)
+            tot += __shfl_down(tot, d, WAVE);
+        if (lane == 0) wtile[wid][it] = tot;
+    }
+    __syncthreads();
+
+    // phase 2: rescan (L2-hot) and scatter; bases derived per wave
+    u32 run = block_offsets[blockIdx.x];
+    for (int it = 0; it < iters; ++it) {
+        u32 wb = run;
+        u32 tile_tot = 0;
+        for (int w = 0; w < BLOCK / WAVE; ++w) {
+            if (w < wid) wb += wtile[w][it];
+            tile_tot += wtile[w][it];
         }
-        // block exclusive scan of cnt
+        long off = base + (long)it * TILE + (long)tid * VBYTES;
+        u8 rel[VBYTES];
+        u32 cnt = gather_marks(text, n, mode, off, rel);
         u32 scan = cnt;
         for (int d = 1; d < WAVE; d <<= 1) {
             u32 x = __shfl_up(scan, d, WAVE);
             if (lane >= d) scan += x;
         }
-        if (lane == WAVE - 1) wtot[wid] = scan;
-        __syncthreads();
-        u32 wbase = 0;
-        for (int w = 0; w < wid; ++w) wbase += wtot[w];
-        u32 excl = cursor + wbase + scan - cnt;
+        u32 excl = wb + scan - cnt;
         for (u32 j = 0; j < cnt; ++j)
             out[excl + j] = (u32)(off + rel[j]);
-        __syncthreads();
-        if (tid == 0) {
-            u32 tile_total = 0;
-            for (int w = 0; w < BLOCK / WAVE; ++w) tile_total += wtot[w];
-            cursor += tile_total;
-        }
-        __syncthreads();
+        run += tile_tot;
     }
 }
 
@@ -210,14 +223,28 @@ __global__ void tfidf_count_kernel(
 // (doc,hash)-keyed seen table; if THAT overflows its probe bound the
 // kernel sets an error flag and the host reruns the chunk on the fully
 // general token-centric kernel above.
+#ifndef DOC_WAVES
 #define DOC_WAVES 4                  // waves per block
+#endif
+#ifndef STAGE_B
 #define STAGE_B 2048                 // staged line segment bytes
+#endif
 #define SEG_OVERLAP 272              // > max dict token length (255)
+#ifndef DOC_SET
 #define DOC_SET 256                  // per-wave dedupe set slots (pow2)
+#endif
+#ifndef FB_PROBE_CAP
 #define FB_PROBE_CAP 512
+#endif
+#ifndef DOC_BLK
 #define DOC_BLK 64                   // contiguous docs per wave block
+#endif
+#ifndef GROUP_BYTES
 #define GROUP_BYTES 1024             // target bytes per doc group
+#endif
+#ifndef CCACHE
 #define CCACHE 1024                  // block-level LDS count cache slots
+#endif
 
 // Two-level counting: Zipf-hot keys would serialize ~50M same-address L2
 // atomics; the block-level LDS cache turns that into one global add per

@@ -17,7 +17,10 @@ _ERR = None
 _HERE = os.path.dirname(os.path.abspath(__file__))
 SOURCES = [os.path.join(_HERE, "hip", "dampr_kernels.hip"),
            os.path.join(_HERE, "hip", "dampr_sort.hip")]
-BUILD_DIR = os.path.join(_HERE, "_build")
+# Overridable for kernel-parameter sweeps (scripts/sweep_tfidf.py): each
+# variant builds into its own in-tree dir and is selected per process.
+BUILD_DIR = os.environ.get(
+    "DAMPR_HIP_BUILD_DIR", os.path.join(_HERE, "_build"))
 
 
 def build(verbose=False):
@@ -27,12 +30,14 @@ def build(verbose=False):
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     os.makedirs(BUILD_DIR, exist_ok=True)
     from torch.utils import cpp_extension
+    defines = [d for d in os.environ.get("DAMPR_HIP_DEFINES",
+                                         "").split(",") if d]
     _EXT = cpp_extension.load(
         name="dampr_hip",
         sources=SOURCES,
         build_directory=BUILD_DIR,
-        extra_cflags=["-O3"],
-        extra_cuda_cflags=["-O3"],
+        extra_cflags=["-O3"] + ["-D" + d for d in defines],
+        extra_cuda_cflags=["-O3"] + ["-D" + d for d in defines],
         verbose=verbose,
     )
     _ERR = None
