@@ -87,11 +87,14 @@ def _identity(k, v):
 
 def _pick_runner(graph, explicit):
     """Engine selection: explicit wins; columnar inputs get the device
-    engine; everything else the multi-process host engine."""
+    engine; everything else the multi-process host engine.  The check is
+    a sentinel attribute, NOT an isinstance against gpu.engine — importing
+    torch into the parent process would bloat every forked CPU worker."""
     if explicit is not None:
         return explicit
-    from .gpu.engine import ColumnSource, GpuRunner
-    if any(isinstance(d, ColumnSource) for d in graph.inputs.values()):
+    if any(getattr(d, "dampr_columnar", False)
+           for d in graph.inputs.values()):
+        from .gpu.engine import GpuRunner
         return GpuRunner
     return MTRunner
 

@@ -59,6 +59,8 @@ def _as_column(x, device, dtype=None):
 class ColumnSource(object):
     """A device-columnar input: (keys i64, vals i64|f64)."""
 
+    dampr_columnar = True          # engine-selection sentinel (dampr.py)
+
     def __init__(self, keys, vals):
         assert keys.dtype == torch.int64
         assert vals.dtype in _VAL_DTYPES
