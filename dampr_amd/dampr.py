@@ -102,9 +102,12 @@ def _pick_runner(graph, explicit):
 class PMap(PBase):
     """A (possibly fused) chain of map operations."""
 
-    def __init__(self, source, pmer, agg=None):
+    def __init__(self, source, pmer, agg=None, agg_specs=None):
         super(PMap, self).__init__(source, pmer)
         self.agg = agg or []
+        # device specs of pending (fusable) maps, parallel to ``agg``;
+        # None for opaque lambdas
+        self.agg_specs = agg_specs or []
 
     def run(self, name=None, **kwargs):
         if self.agg:
@@ -113,12 +116,13 @@ class PMap(PBase):
 
     # -- fusion ------------------------------------------------------------
 
-    def _add_mapper(self, mapper):
+    def _add_mapper(self, mapper, spec=None):
         assert isinstance(mapper, Streamable)
-        return PMap(self.source, self.pmer, self.agg + [mapper])
+        return PMap(self.source, self.pmer, self.agg + [mapper],
+                    self.agg_specs + [spec])
 
-    def _add_map(self, f):
-        return self._add_mapper(Map(f))
+    def _add_map(self, f, spec=None):
+        return self._add_mapper(Map(f), spec)
 
     def checkpoint(self, force=False, combiner=None, options=None):
         """Fuse the pending maps into one stage; shares the materialized
@@ -208,11 +212,14 @@ class PMap(PBase):
         return self._add_map(_filter)
 
     def flat_map(self, f):
-        """Map each value to an iterable and flatten."""
+        """Map each value to an iterable and flatten.  Recognized
+        ``funcs.tokenize_set`` marks the stage for the fused text
+        document-frequency lowering on the device engine."""
         def _flat_map(k, v):
             for vi in f(v):
                 yield k, vi
-        return self._add_map(_flat_map)
+        spec = "flat_tokenize_set" if f is funcs.tokenize_set else None
+        return self._add_map(_flat_map, spec)
 
     def sample(self, prob):
         """Uniformly keep each record with probability ``prob``."""
@@ -262,6 +269,11 @@ class PMap(PBase):
         dev_map = None
         if kname and vname and not self.agg:
             dev_map = ("kv", kname, vname)
+        elif (kname == "identity" and vname == "one"
+              and self.agg_specs == ["flat_tokenize_set"]):
+            # text df idiom: flat_map(tokenize_set).count() fuses into
+            # the one-pass doc-frequency kernel on TextSource inputs
+            dev_map = ("text_df",)
         # No checkpoint: ARReduce attaches the combiner to this stage.
         return ARReduce(self._add_map(_a_group_by), device_map=dev_map)
 
@@ -621,6 +633,17 @@ class Dampr(object):
         """Pipeline over an in-memory list."""
         mi = MemoryInput(list(enumerate(items)), partitions)
         source, ng = Graph().add_input(mi)
+        return PMap(source, cls(ng))
+
+    @classmethod
+    def device_text(cls, data):
+        """Newline-delimited text for the device engine: a path, bytes,
+        or a u8 numpy array.  ``device_text(p).flat_map(funcs.tokenize_set)
+        .count()`` runs as the fused single-pass document-frequency kernel
+        (the flagship TF-IDF path) on an MI355X; the same pipeline runs on
+        the host fallback elsewhere."""
+        from .gpu.engine import TextSource
+        source, ng = Graph().add_input(TextSource(data))
         return PMap(source, cls(ng))
 
     @classmethod

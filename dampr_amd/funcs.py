@@ -50,6 +50,15 @@ COLUMN_FUNCS = {
 }
 
 
+def tokenize_set(line):
+    """Distinct ASCII word-tokens of a line, lowercased — the host mirror
+    of the device tokenizer (ops/hip: is_word/lower_ascii).  Built-in so
+    ``device_text(...).flat_map(funcs.tokenize_set).count()`` lowers to
+    the fused single-pass document-frequency kernel."""
+    import re
+    return set(re.findall(r"[a-z0-9_]+", line.lower()))
+
+
 # --- join pair aggregates ---------------------------------------------------
 # PJoin.reduce(aggregate, many=True) aggregates the cartesian product of
 # each key's (left values, right values).  These named per-pair forms let

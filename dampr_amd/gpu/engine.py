@@ -78,6 +78,77 @@ class ColumnSource(object):
         return cls(k, v)
 
 
+class TextSource(object):
+    """Newline-delimited text input for the device engine: the
+    ``device_text(...).flat_map(funcs.tokenize_set).count()`` idiom lowers
+    onto the fused single-pass document-frequency kernel (gpu/tfidf.py)."""
+
+    dampr_columnar = True
+
+    def __init__(self, data):
+        import numpy as np
+        if isinstance(data, str):
+            with open(data, "rb") as fh:
+                raw = fh.read()
+            self.text = np.frombuffer(raw, dtype=np.uint8).copy()
+        elif isinstance(data, bytes):
+            self.text = np.frombuffer(data, dtype=np.uint8).copy()
+        else:
+            self.text = np.asarray(data, dtype=np.uint8)
+
+
+class TokenStore(object):
+    """Result of the fused text document-frequency stage: token-hash keys
+    + counts, with the string dictionary needed to materialize tokens."""
+
+    keyed = False
+    fkeys = False
+
+    def __init__(self, engine, keys, vals, text_dev):
+        self.engine = engine
+        self.keys = keys
+        self.vals = vals
+        self.text_dev = text_dev
+
+
+class TokenColumnDataset(object):
+    """ColumnDataset analog whose keys decode to token strings via the
+    device string dictionary."""
+
+    def __init__(self, store, keyed):
+        self.store = store
+        self.keyed = keyed
+
+    def read(self):
+        st = self.store
+        blob, lens = st.engine.token_strings(st.keys, st.text_dev)
+        vals = st.vals.cpu().tolist()
+        out = []
+        pos = 0
+        b = blob.tobytes()
+        for i, ln in enumerate(lens):
+            tok = b[pos:pos + int(ln)].decode("ascii")
+            pos += int(ln)
+            if self.keyed:
+                out.append((tok, (tok, vals[i])))
+            else:
+                out.append((tok, vals[i]))
+        out.sort(key=lambda r: r[0])
+        return iter(out)
+
+    def grouped_read(self):
+        import itertools
+        for key, group in itertools.groupby(self.read(),
+                                            key=lambda p: p[0]):
+            yield key, (v for _k, v in group)
+
+    def delete(self):
+        self.store = None
+
+    def __iter__(self):
+        return self.read()
+
+
 class PartStore(dict):
     """{partition -> [DeviceRun]} plus column metadata:
 
@@ -388,6 +459,13 @@ class GpuRunner(RunnerBase):
     def _ingest(self, inp):
         """Input -> partition store.  ColumnSource goes straight to device;
         host Datasets/Chunkers decode through the numeric encoder."""
+        if isinstance(inp, TextSource):
+            if self.world > 1:
+                raise NotImplementedError(
+                    "device_text on multi-rank runs: use the explicit "
+                    "tfidf pipeline (bench.py) which exchanges partials "
+                    "over RCCL")
+            return inp
         if isinstance(inp, ColumnSource):
             keys = inp.keys.to(self.device)
             vals = inp.vals.to(self.device)
@@ -472,6 +550,9 @@ class GpuRunner(RunnerBase):
         if isinstance(store, HostStore):
             from ..dataset import MemoryDataset
             return MemoryDataset(sorted(store, key=lambda r: r[0]))
+        if isinstance(store, TokenStore):
+            return TokenColumnDataset(store, getattr(store, "keyed",
+                                                     False))
         keyed = getattr(store, "keyed", False)
         fkeys = getattr(store, "fkeys", False)
         ks, vs = [], []
@@ -608,6 +689,20 @@ class GpuRunner(RunnerBase):
             return out if out is not None else PartStore()
         if kind == "identity":
             return self._merge_stores(ins)
+        if kind == "text_df":
+            src = ins[0]
+            if not isinstance(src, TextSource) or \
+                    self.device.type != "cuda":
+                return self._host_map(stage, ins)
+            from .tfidf import TfidfEngine
+            text = torch.from_numpy(src.text).to(self.device)
+            eng = TfidfEngine(self.device)
+            eng.reset()
+            assert text.numel() < (1 << 31), \
+                "chunk device_text inputs above 2 GiB"
+            eng.count_chunk(text)
+            keys, df = eng.extract()
+            return TokenStore(eng, keys, df, text)
         if kind == "topk_local":
             # per-partition top-k candidates by value (K11); all
             # candidates meet in partition 0 for the global pass
@@ -676,6 +771,10 @@ class GpuRunner(RunnerBase):
 
     def run_reduce(self, stage, ins):
         spec = stage.options.get("device_reduce")
+        if len(ins) == 1 and isinstance(ins[0], TokenStore) \
+                and spec == ("sum",):
+            ins[0].keyed = True        # keyed-reducer output convention
+            return ins[0]
         if spec is None or any(getattr(s, "keyed", False) for s in ins) \
                 or any(isinstance(s, HostStore) for s in ins):
             return self._host_reduce(stage, ins)
@@ -831,6 +930,21 @@ class GpuRunner(RunnerBase):
     def _decode_store(self, store):
         if isinstance(store, HostStore):
             return list(store)
+        if isinstance(store, TextSource):
+            records = []
+            pos = 0
+            data = store.text.tobytes()
+            for line in data.split(b"\n"):
+                if line or pos + len(line) < len(data):
+                    records.append(
+                        (pos, line.decode("utf-8", "replace")))
+                pos += len(line) + 1
+            if records and pos - 1 >= len(data) and not records[-1][1]:
+                records.pop()
+            return records
+        if isinstance(store, TokenStore):
+            return list(TokenColumnDataset(
+                store, getattr(store, "keyed", False)).read())
         records = []
         keyed = getattr(store, "keyed", False)
         fkeys = getattr(store, "fkeys", False)

@@ -314,3 +314,17 @@ def test_disk_spill_files_created(tmp_path):
         pool.touch(r, torch.device("cpu"))
         assert torch.equal(r.keys, torch.arange(64, dtype=torch.int64) + i)
         pool.release(r)
+
+
+def test_device_text_df_cpu_fallback():
+    """device_text + tokenize_set + count: on CPU the engine decodes the
+    text and runs the host operators; the GPU path fuses into the df
+    kernel (tests/test_gpu_engine.py)."""
+    text = b"the cat sat\nthe dog the dog ran\ncat!\n"
+    got = dict(Dampr.device_text(text)
+               .flat_map(funcs.tokenize_set).count().run().read())
+    assert got == {"the": 2, "cat": 2, "sat": 1, "dog": 1, "ran": 1}
+
+
+def test_tokenize_set_matches_kernel_charset():
+    assert funcs.tokenize_set("A_b9 c-d") == {"a_b9", "c", "d"}

@@ -155,3 +155,14 @@ def test_negative_key_order_gpu():
     vals = np.array([5, -3, 7, -3, 5], dtype=np.int64)
     got = list(_run_dev(Dampr.columns(vals).count()).read())
     assert got == [(-3, 2), (5, 2), (7, 1)]
+
+
+def test_device_text_df_fused():
+    """The reference's own TF-IDF DSL idiom runs on the fused one-pass
+    document-frequency kernel."""
+    from dampr_amd.gpu.corpus import synth_corpus, oracle_df
+    arr = synth_corpus(1 << 20, vocab=5000, seed=21)
+    got = dict(_run_dev(Dampr.device_text(arr)
+                        .flat_map(funcs.tokenize_set).count()).read())
+    want = oracle_df(arr)
+    assert got == want
