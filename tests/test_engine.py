@@ -328,3 +328,18 @@ def test_device_text_df_cpu_fallback():
 
 def test_tokenize_set_matches_kernel_charset():
     assert funcs.tokenize_set("A_b9 c-d") == {"a_b9", "c", "d"}
+
+
+def test_multi_output_shared_subgraph_device():
+    """Dampr.run over the device engine: two outputs sharing one columnar
+    subgraph (the word-stats idiom)."""
+    rng = np.random.default_rng(41)
+    vals = rng.integers(0, 30, size=3000)
+    base = Dampr.columns(vals).checkpoint(True)
+    counts = base.count()
+    sums = base.fold_by(funcs.identity, funcs.add)
+    ec, es = Dampr.run(counts, sums)
+    want_c = collections.Counter(int(v) for v in vals)
+    assert sorted(ec.read()) == sorted(want_c.items())
+    want_s = {k: k * c for k, c in want_c.items()}
+    assert sorted(es.read()) == sorted(want_s.items())
