@@ -412,7 +412,6 @@ class GpuRunner(RunnerBase):
             device = "cuda" if torch.cuda.is_available() else "cpu"
         self.device = torch.device(device)
         self.ops = ops_for(self.device)
-        self.n_partitions = n_partitions or settings.gpu_partitions
         cap = hbm_bytes or settings.hbm_pool_mb * (1 << 20)
         host_cap = host_bytes if host_bytes is not None \
             else settings.host_pool_mb * (1 << 20)
@@ -424,6 +423,28 @@ class GpuRunner(RunnerBase):
             self.rank = torch.distributed.get_rank()
         else:
             self.world, self.rank = 1, 0
+        if n_partitions:
+            self.n_partitions = n_partitions
+        elif self.world == 1 and self._inputs_fit(cap):
+            # single rank, everything resident: one partition means one
+            # full-device sort/reduce per stage instead of 64 small ones
+            # (the kernels are whole-device parallel; partitions exist
+            # for out-of-core paging and the cross-rank exchange)
+            self.n_partitions = 1
+        else:
+            self.n_partitions = settings.gpu_partitions
+
+    def _inputs_fit(self, cap):
+        total = 0
+        for inp in self.graph.inputs.values():
+            if isinstance(inp, ColumnSource):
+                total += inp.keys.numel() * 8 + \
+                    inp.vals.element_size() * inp.vals.numel()
+            elif isinstance(inp, TextSource):
+                total += inp.text.nbytes
+            else:
+                return False             # unknown size: keep partitions
+        return total * 4 < cap           # headroom for intermediates
 
     # -- plan walk ---------------------------------------------------------
 
