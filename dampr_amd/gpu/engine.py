@@ -615,6 +615,17 @@ class GpuRunner(RunnerBase):
         store = PartStore(keyed=keyed, fkeys=fkeys)
         if keys.numel() == 0:
             return store
+        if P == 1 and self.world == 1:
+            # no routing needed: one resident partition
+            if not already_sorted:
+                run = DeviceRun(keys.contiguous(), vals.contiguous(),
+                                sorted=False)
+            else:
+                run = DeviceRun(keys.contiguous(), vals.contiguous(),
+                                sorted=True)
+            store[0] = [run]
+            self.pool.admit(run)
+            return store
         pid = self.ops.partition_of(keys, P)
         order = torch.argsort(pid, stable=True)
         keys, vals, pid = keys[order], vals[order], pid[order]
@@ -652,22 +663,28 @@ class GpuRunner(RunnerBase):
         from ..parallel.shuffle import exchange_columns
         return exchange_columns(keys, vals, pid, self.world)
 
-    def _merged_partition(self, stores, p):
-        """All runs of partition p across input stores, merged key-sorted."""
+    def _merged_partition(self, stores, p, need_sorted=True):
+        """All runs of partition p across input stores, merged
+        key-sorted (``need_sorted=False`` skips the sort for consumers
+        that re-key anyway, e.g. the kv map)."""
         fkeys = any(getattr(s, "fkeys", False) for s in stores)
         ks, vs = [], []
+        all_sorted = True
         for store in stores:
             for run in store.get(p, []):
                 self.pool.touch(run, self.device)
                 ks.append(run.keys)
                 vs.append(run.vals)
+                all_sorted = all_sorted and run.sorted
                 self.pool.release(run)
         if not ks:
             return None, None
-        if len(ks) == 1:
+        if len(ks) == 1 and (all_sorted or not need_sorted):
             return ks[0], vs[0]
         keys = torch.cat(ks)
         vals = torch.cat(vs)
+        if not need_sorted:
+            return keys, vals
         sk, sp = self._sort(keys, fkeys=fkeys)
         return sk, vals[sp.to(torch.int64)]
 
@@ -693,7 +710,8 @@ class GpuRunner(RunnerBase):
             _kind, keyf, valf = spec
             out = None
             for p in self._parts(ins):
-                keys, vals = self._merged_partition(ins, p)
+                keys, vals = self._merged_partition(ins, p,
+                                                    need_sorted=False)
                 if keys is None:
                     continue
                 nk = self._apply_colfunc(keyf, keys, vals)
@@ -733,7 +751,8 @@ class GpuRunner(RunnerBase):
             cand_k, cand_v = [], []
             fkeys = False
             for p in self._parts(ins):
-                keys, vals = self._merged_partition(ins, p)
+                keys, vals = self._merged_partition(ins, p,
+                                                    need_sorted=False)
                 if keys is None:
                     continue
                 if vals.dtype == torch.float64:
@@ -881,8 +900,9 @@ class GpuRunner(RunnerBase):
             fkeys=getattr(left, "fkeys", False)
             or getattr(right, "fkeys", False))
         for p in self._parts([left, right]):
-            lk, lv = self._merged_partition([left], p)
-            rk, rv = self._merged_partition([right], p)
+            lk, lv = self._merged_partition([left], p, need_sorted=False)
+            rk, rv = self._merged_partition([right], p,
+                                            need_sorted=False)
             if lk is None and rk is None:
                 continue
             if lk is None:
