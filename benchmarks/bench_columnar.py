@@ -28,7 +28,9 @@ def bench_group_sum_values(rows, card, device, iters, warmup):
     """Group-by-sum over a value column (fold_by(identity, add)):
     device path = partition + radix sort + segmented reduce."""
     rng = np.random.default_rng(0)
-    vals = rng.integers(0, card, size=rows)
+    vals = torch.from_numpy(rng.integers(0, card, size=rows))
+    if device.startswith("cuda"):
+        vals = vals.to(device)          # resident in HBM (like bench.py)
     times = []
     for i in range(warmup + iters):
         t0 = time.perf_counter()
@@ -48,10 +50,13 @@ def bench_group_sum_values(rows, card, device, iters, warmup):
 
 def bench_join(rows_l, rows_r, card, device, iters, warmup):
     rng = np.random.default_rng(1)
-    lk = rng.integers(0, card, size=rows_l)
-    lv = rng.integers(0, 1000, size=rows_l)
-    rk = rng.integers(0, card, size=rows_r)
-    rv = rng.integers(0, 1000, size=rows_r)
+    def col(a):
+        t = torch.from_numpy(a)
+        return t.to(device) if device.startswith("cuda") else t
+    lk = col(rng.integers(0, card, size=rows_l))
+    lv = col(rng.integers(0, 1000, size=rows_l))
+    rk = col(rng.integers(0, card, size=rows_r))
+    rv = col(rng.integers(0, 1000, size=rows_r))
     times = []
     for i in range(warmup + iters):
         t0 = time.perf_counter()
