@@ -343,3 +343,12 @@ def test_multi_output_shared_subgraph_device():
     assert sorted(ec.read()) == sorted(want_c.items())
     want_s = {k: k * c for k, c in want_c.items()}
     assert sorted(es.read()) == sorted(want_s.items())
+
+
+def test_device_text_df_compose_downstream():
+    """TokenStore results compose with downstream host-fallback stages."""
+    text = b"aa bb\naa cc\n"
+    got = sorted(Dampr.device_text(text)
+                 .flat_map(funcs.tokenize_set).count()
+                 .filter(lambda kv: kv[1] > 1).run().read())
+    assert got == [("aa", 2)]
