@@ -82,19 +82,26 @@ def main():
     ap.add_argument("--card", type=int, default=1_000_000)
     ap.add_argument("--iters", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--trace", action="store_true",
+                    help="print per-stage timings of the last run")
     args = ap.parse_args()
     device = "cuda:0" if torch.cuda.is_available() else "cpu"
     if device == "cpu":
         args.rows = min(args.rows, 1_000_000)
 
+    from dampr_amd.utils.trace import get_trace
     r = bench_group_sum_values(args.rows, args.card, device, args.iters,
                                args.warmup)
     r["device"] = device
     print(json.dumps(r))
+    if args.trace:
+        print(get_trace().report())
     r = bench_join(args.rows, args.rows // 8, max(args.rows, 1), device,
                    args.iters, args.warmup)
     r["device"] = device
     print(json.dumps(r))
+    if args.trace:
+        print(get_trace().report())
 
 
 if __name__ == "__main__":

@@ -100,33 +100,59 @@ __global__ void marks_lookback_kernel(const u8* __restrict__ text, long n,
     u32 agg = 0;
     for (int w = 0; w < BLOCK / WAVE; ++w) agg += wtot[w];
 
-    if (tid == 0) {
-        // publish aggregate, then resolve the exclusive prefix
+    // Publish aggregate, then resolve the exclusive prefix with a
+    // wave-parallel lookback: wave 0's 64 lanes read 64 predecessor
+    // status words at once (the naive one-lane walk serialized the whole
+    // grid on load latency).
+    if (wid == 0) {
         if (b == 0) {
-            __atomic_store_n((u64*)&status[0], MK_FLAG_INC | (u64)agg,
-                             __ATOMIC_RELEASE);
-            s_prefix = 0;
-        } else {
-            __atomic_store_n((u64*)&status[b], MK_FLAG_AGG | (u64)agg,
-                             __ATOMIC_RELEASE);
-            u64 prefix = 0;
-            long j = b - 1;
-            while (j >= 0) {
-                u64 st;
-                do {
-                    st = __atomic_load_n((u64*)&status[j],
-                                         __ATOMIC_ACQUIRE);
-                } while (!(st >> 62));
-                prefix += MK_VAL(st);
-                if (st & MK_FLAG_INC) break;
-                --j;
+            if (lane == 0) {
+                __atomic_store_n((u64*)&status[0],
+                                 MK_FLAG_INC | (u64)agg,
+                                 __ATOMIC_RELEASE);
+                s_prefix = 0;
             }
-            __atomic_store_n((u64*)&status[b],
-                             MK_FLAG_INC | (u64)(prefix + agg),
-                             __ATOMIC_RELEASE);
-            s_prefix = prefix;
+        } else {
+            if (lane == 0)
+                __atomic_store_n((u64*)&status[b], MK_FLAG_AGG | (u64)agg,
+                                 __ATOMIC_RELEASE);
+            u64 prefix = 0;
+            long base = b;             // window is [base-64, base)
+            while (true) {
+                const long j = base - WAVE + lane;
+                u64 st = 0;
+                bool ready = false;
+                while (!ready) {
+                    st = (j >= 0)
+                        ? __atomic_load_n((u64*)&status[j],
+                                          __ATOMIC_ACQUIRE)
+                        : MK_FLAG_INC;
+                    // wait until the whole window is resolved
+                    ready = (__ballot((st >> 62) != 0)
+                             == ~0ULL);
+                }
+                const u64 inc_m = __ballot((st & MK_FLAG_INC) != 0
+                                           && j >= 0);
+                int cut = -1;          // highest lane holding INC
+                if (inc_m) cut = 63 - __clzll(inc_m);
+                // sum contributions from lanes > cut (exclusive of
+                // earlier blocks already folded into the INC)
+                u64 contrib = (j >= 0 && lane >= cut) ? MK_VAL(st) : 0;
+                for (int d = WAVE / 2; d > 0; d >>= 1)
+                    contrib += __shfl_down(contrib, d, WAVE);
+                prefix += __shfl(contrib, 0, WAVE);
+                if (inc_m || base - WAVE <= 0)
+                    break;
+                base -= WAVE;
+            }
+            if (lane == 0) {
+                __atomic_store_n((u64*)&status[b],
+                                 MK_FLAG_INC | (u64)(prefix + agg),
+                                 __ATOMIC_RELEASE);
+                s_prefix = prefix;
+            }
         }
-        if (off0 + TILE >= n)          // last block: publish the total
+        if (lane == 0 && off0 + TILE >= n)   // last block: the total
             *total = s_prefix + agg;
     }
     __syncthreads();
