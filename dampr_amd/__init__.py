@@ -11,7 +11,6 @@ Execution is tiered:
   RCCL over xGMI for the multi-GPU exchange.
 """
 import logging
-import sys
 
 from .dampr import Dampr, PMap, PReduce, PJoin, ARReduce, ValueEmitter
 from .base import BlockMapper, BlockReducer, Map, StreamMapper, \

@@ -19,7 +19,6 @@ dataset.py:43-629) with a different design:
   module is the host/out-of-core tier.
 """
 import heapq
-import io
 import itertools
 import os
 import pickle
@@ -326,7 +325,6 @@ class Spiller(DatasetWriter):
         batch — the per-record method-call overhead dominates pure-Python
         hot loops otherwise."""
         from itertools import islice
-        buf_extend = None
         while True:
             batch = list(islice(it, 16384))
             if not batch:

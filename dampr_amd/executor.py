@@ -15,7 +15,6 @@ boundary; only job descriptors and {partition: [dataset-handle]} results do.
 """
 import logging
 import multiprocessing
-import os
 import queue as queue_mod
 import traceback
 

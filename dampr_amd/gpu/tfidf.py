@@ -15,12 +15,10 @@ No sort is needed anywhere on the hot path — grouping is the device hash
 table, exactly where the reference uses its in-memory combine dict
 (dataset.py:84-117).
 """
-import math
 import os
 
 import torch
 
-from .. import settings
 from ..ops import native
 
 MODE_NEWLINE = 0
