@@ -43,7 +43,9 @@ log = logging.getLogger("dampr_amd")
 _VAL_DTYPES = (torch.int64, torch.float64)
 
 
-def _as_column(x, device, dtype=None):
+def _as_column(x, device=None, dtype=None):
+    """device=None keeps a torch input where it lives (an HBM-resident
+    tensor must not bounce through the host)."""
     if isinstance(x, torch.Tensor):
         t = x
     else:
@@ -53,7 +55,7 @@ def _as_column(x, device, dtype=None):
         t = t.to(torch.float64 if t.is_floating_point() else torch.int64)
     if dtype is not None:
         t = t.to(dtype)
-    return t.to(device)
+    return t if device is None else t.to(device)
 
 
 class ColumnSource(object):
@@ -69,12 +71,14 @@ class ColumnSource(object):
         self.vals = vals
 
     @classmethod
-    def from_data(cls, vals, keys=None, device="cpu"):
+    def from_data(cls, vals, keys=None, device=None):
         v = _as_column(vals, device)
         if keys is None:
             k = torch.arange(v.numel(), dtype=torch.int64, device=v.device)
         else:
             k = _as_column(keys, device, torch.int64)
+            if k.device != v.device:
+                k = k.to(v.device)
         return cls(k, v)
 
 

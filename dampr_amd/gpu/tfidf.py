@@ -54,7 +54,7 @@ class TfidfEngine(object):
 
     # -- scans ---------------------------------------------------------------
 
-    _MK_TILE = 4096                 # must match TILE in dampr_kernels.hip
+    _MK_TILE = 65536            # MK_ITERS * TILE in dampr_kernels.hip
 
     def positions(self, text, mode):
         """Ordered mark positions in ONE pass (decoupled-lookback kernel):
