@@ -73,7 +73,7 @@ __device__ __forceinline__ u32 gather_marks(const u8* __restrict__ text,
     return cnt;
 }
 
-#define MK_ITERS 16                   // tiles per block (64 KiB)
+#define MK_ITERS 64                   // tiles per block (256 KiB)
 
 __global__ void marks_lookback_kernel(const u8* __restrict__ text, long n,
                                       int mode,
