@@ -54,31 +54,13 @@ class TfidfEngine(object):
 
     # -- scans ---------------------------------------------------------------
 
-    _MK_TILE = 65536            # MK_ITERS * TILE in dampr_kernels.hip
-
     def positions(self, text, mode):
-        """Ordered mark positions in ONE pass (decoupled-lookback kernel):
-        no count pre-pass, no host cumsum.  Buffers persist across chunks
-        (position capacity = worst case one mark per byte)."""
-        n = text.numel()
-        nblocks = (n + self._MK_TILE - 1) // self._MK_TILE
-        if getattr(self, "_pos_buf", None) is None \
-                or self._pos_buf.numel() < n + 1:
-            self._pos_buf = torch.empty(n + 1, dtype=torch.int32,
-                                        device=self.device)
-        if getattr(self, "_mk_status", None) is None \
-                or self._mk_status.numel() < nblocks:
-            self._mk_status = torch.zeros(max(nblocks, 1),
-                                          dtype=torch.int64,
-                                          device=self.device)
-            self._mk_total = torch.zeros(1, dtype=torch.int64,
-                                         device=self.device)
-        else:
-            self._mk_status[:nblocks].zero_()
-        self.ext.marks(text, mode, self._pos_buf, self._mk_status,
-                       self._mk_total)
-        total = int(self._mk_total.item())
-        return self._pos_buf[:total], total
+        counts = self.ext.mark_counts(text, mode).to(torch.int64)
+        offsets = torch.cumsum(counts, 0) - counts
+        total = int(offsets[-1].item() + counts[-1].item())
+        out = torch.empty(total, dtype=torch.int32, device=self.device)
+        self.ext.mark_positions(text, mode, offsets.to(torch.int32), out)
+        return out, total
 
     # -- one chunk -----------------------------------------------------------
 
@@ -127,9 +109,6 @@ class TfidfEngine(object):
         return n_docs
 
     def _count_chunk_general(self, text, nl, pos_base, doc_base):
-        # ``nl`` may be a view of the shared positions buffer; the next
-        # positions() call below would overwrite it
-        nl = nl.clone()
         ts, n_tok = self.positions(text, MODE_TOKEN_START)
         seen_cap = _pow2_at_least(2 * max(n_tok, 1))
         if self.seen is None or self.seen.numel() < seen_cap:

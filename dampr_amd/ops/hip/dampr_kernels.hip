@@ -32,16 +32,61 @@ __device__ __forceinline__ bool mark_at(int mode, const u8* text,
     return i == 0 || !is_word(text[i - 1]);
 }
 
-// ------------------------------------------------- single-pass positions
-// Decoupled-lookback stream compaction (rocprim-style): one pass over the
-// text produces the ordered mark-position array and the total, replacing
-// the old count+write two-pass (two full text reads + host cumsum).
-// status[b] packs {flag:2, value:62}: 1 = aggregate ready, 2 = inclusive
-// prefix ready.
+// ---------------------------------------------------------------- scan pass 1
+// Each block owns a contiguous [base, base + iters*TILE) byte range and
+// counts its marks.
+__global__ void count_marks_kernel(const u8* __restrict__ text, long n,
+                                   int mode, int iters,
+                                   u32* __restrict__ counts) {
+    long base = (long)blockIdx.x * iters * TILE;
+    int tid = threadIdx.x;
+    u32 local = 0;
+    for (int it = 0; it < iters; ++it) {
+        long off = base + (long)it * TILE + (long)tid * VBYTES;
+        if (off >= n) break;
+        if (off + VBYTES <= n) {
+            uint4 v = *reinterpret_cast<const uint4*>(text + off);
+            const u8* b = reinterpret_cast<const u8*>(&v);
+            #pragma unroll
+            for (int j = 0; j < VBYTES; ++j) {
+                u8 c = b[j];
+                bool m;
+                if (mode == MODE_NEWLINE) {
+                    m = (c == '\n');
+                } else {
+                    bool prev_word = (j > 0)
+                        ? is_word(b[j - 1])
+                        : (off > 0 ? is_word(text[off - 1]) : false);
+                    m = is_word(c) && !prev_word;
+                }
+                local += m;
+            }
+        } else {
+            for (long i = off; i < n; ++i)
+                local += mark_at(mode, text, i, n);
+        }
+    }
+    // block reduction: wave reduce then LDS
+    __shared__ u32 wsum[BLOCK / WAVE];
+    for (int d = WAVE / 2; d > 0; d >>= 1)
+        local += __shfl_down(local, d, WAVE);
+    int lane = tid & (WAVE - 1), wid = tid / WAVE;
+    if (lane == 0) wsum[wid] = local;
+    __syncthreads();
+    if (tid == 0) {
+        u32 total = 0;
+        for (int w = 0; w < BLOCK / WAVE; ++w) total += wsum[w];
+        counts[blockIdx.x] = total;
+    }
+}
 
-#define MK_FLAG_AGG  (1ULL << 62)
-#define MK_FLAG_INC  (2ULL << 62)
-#define MK_VAL(x)    ((x) & ((1ULL << 62) - 1))
+// ---------------------------------------------------------------- scan pass 2
+// Re-scan and write mark positions in ascending order.  Two-phase within
+// the block: per-(wave, tile) counts into LDS, ONE barrier, then a
+// sync-free scatter pass where every wave derives its own bases from the
+// count matrix (the old per-tile cursor + double barrier serialized the
+// block on memory latency every 4 KiB).
+#define SCAN_MAX_ITERS 16
 
 __device__ __forceinline__ u32 gather_marks(const u8* __restrict__ text,
                                             long n, int mode, long off,
@@ -73,24 +118,19 @@ __device__ __forceinline__ u32 gather_marks(const u8* __restrict__ text,
     return cnt;
 }
 
-#define MK_ITERS 64                   // tiles per block (256 KiB)
+__global__ void write_marks_kernel(const u8* __restrict__ text, long n,
+                                   int mode, int iters,
+                                   const u32* __restrict__ block_offsets,
+                                   u32* __restrict__ out) {
+    long base = (long)blockIdx.x * iters * TILE;
+    int tid = threadIdx.x;
+    int lane = tid & (WAVE - 1), wid = tid / WAVE;
+    __shared__ u32 wtile[BLOCK / WAVE][SCAN_MAX_ITERS];
 
-__global__ void marks_lookback_kernel(const u8* __restrict__ text, long n,
-                                      int mode,
-                                      volatile u64* __restrict__ status,
-                                      u32* __restrict__ out,
-                                      u64* __restrict__ total) {
-    const long b = blockIdx.x;
-    const long off0 = b * (long)MK_ITERS * TILE;
-    const int tid = threadIdx.x;
-    const int lane = tid & (WAVE - 1), wid = tid / WAVE;
-    __shared__ u32 wtile[BLOCK / WAVE][MK_ITERS];
-    __shared__ u64 s_prefix;
-
-    // phase 1: per-(wave, tile) counts
-    u8 rel[VBYTES];
-    for (int it = 0; it < MK_ITERS; ++it) {
-        const long off = off0 + (long)it * TILE + (long)tid * VBYTES;
+    // phase 1: per-(wave, tile) mark counts
+    for (int it = 0; it < iters; ++it) {
+        long off = base + (long)it * TILE + (long)tid * VBYTES;
+        u8 rel[VBYTES];
         u32 cnt = gather_marks(text, n, mode, off, rel);
         u32 tot = cnt;
         for (int d = WAVE / 2; d > 0; d >>= 1)
@@ -98,78 +138,25 @@ __global__ void marks_lookback_kernel(const u8* __restrict__ text, long n,
         if (lane == 0) wtile[wid][it] = tot;
     }
     __syncthreads();
-    u32 agg = 0;
-    for (int w = 0; w < BLOCK / WAVE; ++w)
-        for (int it = 0; it < MK_ITERS; ++it) agg += wtile[w][it];
 
-    // publish aggregate; wave-parallel decoupled lookback (64 status
-    // words per probe) resolves the exclusive prefix
-    if (wid == 0) {
-        if (b == 0) {
-            if (lane == 0) {
-                __atomic_store_n((u64*)&status[0],
-                                 MK_FLAG_INC | (u64)agg,
-                                 __ATOMIC_RELEASE);
-                s_prefix = 0;
-            }
-        } else {
-            if (lane == 0)
-                __atomic_store_n((u64*)&status[b], MK_FLAG_AGG | (u64)agg,
-                                 __ATOMIC_RELEASE);
-            u64 prefix = 0;
-            long base = b;             // window is [base-64, base)
-            while (true) {
-                const long j = base - WAVE + lane;
-                u64 st = 0;
-                bool ready = false;
-                while (!ready) {
-                    st = (j >= 0)
-                        ? __atomic_load_n((u64*)&status[j],
-                                          __ATOMIC_ACQUIRE)
-                        : MK_FLAG_INC;
-                    ready = (__ballot((st >> 62) != 0) == ~0ULL);
-                }
-                const u64 inc_m = __ballot((st & MK_FLAG_INC) != 0
-                                           && j >= 0);
-                int cut = -1;
-                if (inc_m) cut = 63 - __clzll(inc_m);
-                u64 contrib = (j >= 0 && lane >= cut) ? MK_VAL(st) : 0;
-                for (int d = WAVE / 2; d > 0; d >>= 1)
-                    contrib += __shfl_down(contrib, d, WAVE);
-                prefix += __shfl(contrib, 0, WAVE);
-                if (inc_m || base - WAVE <= 0)
-                    break;
-                base -= WAVE;
-            }
-            if (lane == 0) {
-                __atomic_store_n((u64*)&status[b],
-                                 MK_FLAG_INC | (u64)(prefix + agg),
-                                 __ATOMIC_RELEASE);
-                s_prefix = prefix;
-            }
-        }
-        if (lane == 0 && off0 + (long)MK_ITERS * TILE >= n)
-            *total = s_prefix + agg;
-    }
-    __syncthreads();
-
-    // phase 2: rescan (L2-hot) and scatter at resolved offsets
-    u32 run = (u32)s_prefix;
-    for (int it = 0; it < MK_ITERS; ++it) {
+    // phase 2: rescan (L2-hot) and scatter; bases derived per wave
+    u32 run = block_offsets[blockIdx.x];
+    for (int it = 0; it < iters; ++it) {
         u32 wb = run;
         u32 tile_tot = 0;
         for (int w = 0; w < BLOCK / WAVE; ++w) {
             if (w < wid) wb += wtile[w][it];
             tile_tot += wtile[w][it];
         }
-        const long off = off0 + (long)it * TILE + (long)tid * VBYTES;
-        const u32 cnt = gather_marks(text, n, mode, off, rel);
+        long off = base + (long)it * TILE + (long)tid * VBYTES;
+        u8 rel[VBYTES];
+        u32 cnt = gather_marks(text, n, mode, off, rel);
         u32 scan = cnt;
         for (int d = 1; d < WAVE; d <<= 1) {
             u32 x = __shfl_up(scan, d, WAVE);
             if (lane >= d) scan += x;
         }
-        const u32 excl = wb + scan - cnt;
+        u32 excl = wb + scan - cnt;
         for (u32 j = 0; j < cnt; ++j)
             out[excl + j] = (u32)(off + rel[j]);
         run += tile_tot;
@@ -728,31 +715,36 @@ void check_u8(const torch::Tensor& t) {
                 t.is_contiguous(), "expected contiguous u8 device tensor");
 }
 
+constexpr int SCAN_ITERS = 8;   // 32 KiB per block
 
 }  // namespace
 
 // Returns per-block mark counts; python computes exclusive offsets.
-std::vector<torch::Tensor> marks(torch::Tensor text, long mode,
-                                 torch::Tensor out_buf,
-                                 torch::Tensor status_buf,
-                                 torch::Tensor total_buf) {
+torch::Tensor mark_counts(torch::Tensor text, long mode) {
     check_u8(text);
     long n = text.numel();
-    long nblocks = (n + (long)MK_ITERS * TILE - 1)
-                   / ((long)MK_ITERS * TILE);
-    TORCH_CHECK(status_buf.numel() >= std::max(nblocks, 1L),
-                "status buffer too small");
-    if (n == 0) {
-        total_buf.zero_();
-        return {out_buf, total_buf};
-    }
-    hipLaunchKernelGGL(marks_lookback_kernel, dim3((u32)nblocks),
-                       dim3(BLOCK), 0, cur_stream(),
-                       text.data_ptr<u8>(), n, (int)mode,
-                       (volatile u64*)status_buf.data_ptr(),
-                       (u32*)out_buf.data_ptr(),
-                       (u64*)total_buf.data_ptr());
-    return {out_buf, total_buf};
+    long span = (long)SCAN_ITERS * TILE;
+    long nblocks = (n + span - 1) / span;
+    auto counts = torch::empty({std::max(nblocks, 1L)},
+        torch::TensorOptions().dtype(torch::kUInt32).device(text.device()));
+    if (n == 0) { counts.zero_(); return counts; }
+    hipLaunchKernelGGL(count_marks_kernel, dim3((u32)nblocks), dim3(BLOCK),
+                       0, cur_stream(), text.data_ptr<u8>(), n, (int)mode,
+                       SCAN_ITERS, (u32*)counts.data_ptr());
+    return counts;
+}
+
+void mark_positions(torch::Tensor text, long mode,
+                    torch::Tensor block_offsets, torch::Tensor out) {
+    check_u8(text);
+    long n = text.numel();
+    if (n == 0 || out.numel() == 0) return;
+    long span = (long)SCAN_ITERS * TILE;
+    long nblocks = (n + span - 1) / span;
+    hipLaunchKernelGGL(write_marks_kernel, dim3((u32)nblocks), dim3(BLOCK),
+                       0, cur_stream(), text.data_ptr<u8>(), n, (int)mode,
+                       SCAN_ITERS, (u32*)block_offsets.data_ptr(),
+                       (u32*)out.data_ptr());
 }
 
 void tfidf_count(torch::Tensor text, torch::Tensor nl_pos,
@@ -935,9 +927,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("hj_emit", &hj_emit, "hash-join emit (l,r) row-index pairs");
     m.def("tsv_sizes", &tsv_sizes, "per-row TSV byte sizes");
     m.def("tsv_format", &tsv_format, "format token/df/idf rows as TSV");
-    m.def("marks", &marks,
-          "single-pass ordered mark positions + total (decoupled "
-          "lookback; mode 0=newline, 1=token start)");
+    m.def("mark_counts", &mark_counts,
+          "per-block counts of marks (mode 0=newline, 1=token start)");
+    m.def("mark_positions", &mark_positions,
+          "write ascending mark positions given exclusive block offsets");
     m.def("tfidf_count", &tfidf_count,
           "tokenize+hash+per-doc-dedupe+df-count in one pass");
     m.def("tfidf_count_docs", &tfidf_count_docs,
