@@ -302,14 +302,27 @@ class HbmPool(object):
         self.spill_dir = spill_dir or settings.spill_dir
         self._run_tag = "dampr_amd_{}".format(uuid.uuid4().hex[:10])
         self._file_ctr = 0
+        self._disk_paths = []
         self._os = os
 
     def _next_path(self):
         self._os.makedirs(self.spill_dir, exist_ok=True)
         self._file_ctr += 1
-        return self._os.path.join(
+        p = self._os.path.join(
             self.spill_dir,
             "{}_{}.run".format(self._run_tag, self._file_ctr))
+        self._disk_paths.append(p)
+        return p
+
+    def cleanup(self):
+        """Unlink any spill files still on disk (runs never paged back
+        before the job finished)."""
+        for p in self._disk_paths:
+            try:
+                self._os.unlink(p)
+            except OSError:
+                pass
+        self._disk_paths = []
 
     def admit(self, run):
         self.used += run.nbytes
@@ -477,6 +490,7 @@ class GpuRunner(RunnerBase):
         for source in outputs:
             store = data[source]
             rets.append(self._collect(store))
+        self.pool.cleanup()
         return rets
 
     # -- ingest / collect --------------------------------------------------
