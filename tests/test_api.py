@@ -481,3 +481,19 @@ def test_indexer(tmp_path):
     assert both == ["apple red\n"]
     either = sorted(idx.intersect(["red", "banana"], min_match=1).read())
     assert either == ["apple red\n", "banana yellow\n", "cherry red\n"]
+
+
+def test_compaction_fan_in():
+    """Force the file-count compaction pass: more map workers than
+    max_files_per_stage means every partition exceeds the fan-in cap
+    (reference: runner.py:293-320)."""
+    from dampr_amd import settings as st
+    old = st.max_files_per_stage
+    st.max_files_per_stage = 3
+    try:
+        res = Dampr.memory(list(range(200)), partitions=16) \
+            .count(lambda x: x % 5).run(n_maps=8)
+        got = dict(res.read())
+        assert got == {k: 40 for k in range(5)}
+    finally:
+        st.max_files_per_stage = old
