@@ -62,9 +62,23 @@ def main():
     eng = TfidfEngine(device)
     shutil.rmtree(args.sink_dir, ignore_errors=True)
 
+    # chunk bounds on newline boundaries (count_chunk takes < 2 GiB)
+    import numpy as np
+    cb = 1 << 30
+    bounds = [0]
+    while bounds[-1] < n:
+        e = min(bounds[-1] + cb, n)
+        if e < n:
+            nl = np.flatnonzero(text_np[e - 1:min(e + (1 << 16), n)]
+                                == ord("\n"))
+            e = (e - 1 + int(nl[0]) + 1) if len(nl) else n
+        bounds.append(e)
+    chunks = [(s, e) for s, e in zip(bounds, bounds[1:])]
+
     def step():
         eng.reset()
-        eng.count_chunk(text)
+        for s, e in chunks:
+            eng.count_chunk(text[s:e].contiguous(), pos_base=s)
         keys, df = eng.extract()
         if dist_mode:
             blob, lens = eng.token_strings_dev(keys, text)
