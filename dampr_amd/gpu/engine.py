@@ -91,14 +91,31 @@ class TextSource(object):
 
     def __init__(self, data):
         import numpy as np
-        if isinstance(data, str):
+        self.text_t = None             # device-resident u8 tensor, if given
+        self._text_np = None
+        if isinstance(data, torch.Tensor):
+            assert data.dtype == torch.uint8
+            self.text_t = data
+        elif isinstance(data, str):
             with open(data, "rb") as fh:
                 raw = fh.read()
-            self.text = np.frombuffer(raw, dtype=np.uint8).copy()
+            self._text_np = np.frombuffer(raw, dtype=np.uint8).copy()
         elif isinstance(data, bytes):
-            self.text = np.frombuffer(data, dtype=np.uint8).copy()
+            self._text_np = np.frombuffer(data, dtype=np.uint8).copy()
         else:
-            self.text = np.asarray(data, dtype=np.uint8)
+            self._text_np = np.asarray(data, dtype=np.uint8)
+
+    @property
+    def nbytes(self):
+        return (self.text_t.numel() if self.text_t is not None
+                else self._text_np.nbytes)
+
+    @property
+    def text(self):
+        """Host view (materialized on demand for fallback paths)."""
+        if self._text_np is None:
+            self._text_np = self.text_t.cpu().numpy()
+        return self._text_np
 
 
 class TokenStore(object):
@@ -458,7 +475,7 @@ class GpuRunner(RunnerBase):
                 total += inp.keys.numel() * 8 + \
                     inp.vals.element_size() * inp.vals.numel()
             elif isinstance(inp, TextSource):
-                total += inp.text.nbytes
+                total += inp.nbytes
             else:
                 return False             # unknown size: keep partitions
         return total * 4 < cap           # headroom for intermediates
@@ -751,9 +768,11 @@ class GpuRunner(RunnerBase):
             if not isinstance(src, TextSource) or \
                     self.device.type != "cuda":
                 return self._host_map(stage, ins)
-            import numpy as np
             from .tfidf import TfidfEngine
-            text = torch.from_numpy(src.text).to(self.device)
+            if src.text_t is not None:
+                text = src.text_t.to(self.device)
+            else:
+                text = torch.from_numpy(src.text).to(self.device)
             eng = TfidfEngine(self.device)
             eng.reset()
             n = text.numel()
@@ -762,10 +781,10 @@ class GpuRunner(RunnerBase):
             while bounds[-1] < n:
                 e = min(bounds[-1] + cb, n)
                 if e < n:
-                    nl = np.flatnonzero(
-                        src.text[e - 1:min(e + (1 << 16), n)]
-                        == ord("\n"))
-                    e = (e - 1 + int(nl[0]) + 1) if len(nl) else n
+                    seg = text[e - 1:min(e + (1 << 16), n)]
+                    nl = torch.nonzero(seg == ord("\n")).flatten()
+                    e = (e - 1 + int(nl[0].item()) + 1) if nl.numel() \
+                        else n
                 bounds.append(e)
             for s0, e0 in zip(bounds, bounds[1:]):
                 eng.count_chunk(text[s0:e0].contiguous(), pos_base=s0)
