@@ -51,8 +51,31 @@ def check_columnar_spill():
     print("columnar spill ok: {} rows, {} groups".format(rows, len(got)))
 
 
+def bench_out_of_core(rows=750_000_000):
+    """Out-of-core group-by at ~1.5x the HBM pool: 12 GB of columns
+    through a 6 GB pool (runs spill to pinned host and page back)."""
+    import time
+    rng = np.random.default_rng(9)
+    vals = torch.from_numpy(
+        rng.integers(0, 1_000_000, size=rows)).cuda()
+    t0 = time.perf_counter()
+    out = Dampr.columns(vals).fold_by(funcs.identity, funcs.add) \
+        .run(device="cuda:0", hbm_bytes=6 << 30)
+    k, v = out.dataset.columns()
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    assert int(v.sum().item()) == int(vals.sum().item())
+    print("out-of-core groupby: {} rows ({}.{} GB) through 6 GB pool: "
+          "{:.0f} ms, {:.0f}M rows/s, {} groups".format(
+              rows, rows * 16 // (1 << 30),
+              (rows * 16 % (1 << 30)) // 100000000, dt * 1000,
+              rows / dt / 1e6, k.numel()))
+
+
 if __name__ == "__main__":
     assert torch.cuda.is_available()
     check_columnar_spill()
     check_text(int(sys.argv[1]) if len(sys.argv) > 1 else 8)
+    if len(sys.argv) > 2 and sys.argv[2] == "big":
+        bench_out_of_core()
     print("scale checks passed")
