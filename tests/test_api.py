@@ -497,3 +497,29 @@ def test_compaction_fan_in():
         assert got == {k: 40 for k in range(5)}
     finally:
         st.max_files_per_stage = old
+
+
+def test_worker_exception_fails_fast():
+    """A raising UDF must surface as WorkerCrash with the traceback, not
+    hang the run (the reference deadlocks: stagerunner.py:35-37)."""
+    from dampr_amd.executor import WorkerCrash
+
+    def boom(x):
+        raise ValueError("intentional-test-boom")
+
+    with pytest.raises(WorkerCrash) as ei:
+        Dampr.memory(list(range(100))).map(boom).count().run()
+    assert "intentional-test-boom" in str(ei.value)
+
+
+def test_worker_hard_death_detected():
+    """A worker dying without reporting (os._exit) is detected instead of
+    deadlocking."""
+    import os as _os
+    from dampr_amd.executor import WorkerCrash
+
+    def die(x):
+        _os._exit(13)
+
+    with pytest.raises(WorkerCrash):
+        Dampr.memory(list(range(100))).map(die).count().run()
