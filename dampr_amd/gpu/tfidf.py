@@ -5,8 +5,9 @@ reference's ``flat_map(set(tokenize)).count().cross_right(docs.len(), idf)
 .sink_tsv`` (benchmarks/tf-idf-dampr.py:9-21) re-built MI355X-native:
 
   text (resident in HBM)
-    -> newline scan + token-start scan        (hand-written HIP, K1-adjacent)
-    -> tokenize+hash+per-doc-dedupe+df-count  (one fused kernel: K1+K6)
+    -> newline position scan                  (hand-written HIP, K1-adjacent)
+    -> tokenize+hash+per-doc-dedupe+df-count  (one fused group-stream
+       kernel: wave-wide segmented-XOR token hashing, K1+K6)
     -> [multi-GPU: RCCL all-to-all exchange of (key, df) partials]
     -> idf epilogue (K9's scalar broadcast-apply, fused elementwise)
     -> token-string gather + TSV sink
@@ -69,10 +70,11 @@ class TfidfEngine(object):
         is the chunk's absolute byte offset in the job's corpus (dict
         entries store absolute positions so multi-chunk gathers work).
 
-        Fast path: the wave-per-doc kernel (LDS staging + LDS-set dedupe).
-        Documents whose distinct-token count overflows both the LDS set and
-        the small global fallback seen-table trip an error flag and the
-        chunk reruns on the fully general token-centric kernel."""
+        Fast path: the group-stream kernel (uint4-staged LDS windows,
+        ballot token starts, segmented-XOR hash scan, doc-salted LDS-set
+        dedupe).  Documents whose distinct-token count overflows both the
+        LDS set and the global fallback seen-table trip an error flag and
+        the chunk reruns on the fully general token-centric kernel."""
         assert text.numel() < (1 << 31), "chunk must be < 2 GiB"
         nl, n_nl = self.positions(text, MODE_NEWLINE)
         # docs = newlines (+1 unterminated tail line)
