@@ -352,3 +352,60 @@ def test_device_text_df_compose_downstream():
                  .flat_map(funcs.tokenize_set).count()
                  .filter(lambda kv: kv[1] > 1).run().read())
     assert got == [("aa", 2)]
+
+
+def _engine_rank2(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+
+        # device join across ranks: both sides co-partitioned by the
+        # exchange; union of per-rank outputs == single-rank oracle
+        lk = np.arange(1000) % 37
+        lv = np.arange(1000)
+        rk = np.arange(200) % 37
+        rv = np.arange(200) * 2
+        out = Dampr.columns(lv, keys=lk).join(
+            Dampr.columns(rv, keys=rk)) \
+            .reduce(funcs.pair_sum, many=True).run()
+        pairs = sorted(out.read())
+        gathered = [None] * world
+        dist.all_gather_object(gathered, pairs)
+        merged = sorted(p for lst in gathered for p in lst)
+        want = []
+        for i in range(1000):
+            for j in range(200):
+                if lk[i] == rk[j]:
+                    want.append((int(lk[i]), int(lv[i] + rv[j])))
+        assert merged == sorted(want), (len(merged), len(want))
+
+        # device topk across ranks: candidates meet on rank 0
+        vals = np.arange(5000)
+        got = Dampr.columns(vals).topk(7).run().read()
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got)
+        merged = sorted(p for lst in gathered for p in lst)
+        # every rank ingests a slice; global top-7 of the union
+        assert merged == list(range(4993, 5000)), merged
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+def test_engine_gloo_join_topk_world2():
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    world = 2
+    procs = [ctx.Process(target=_engine_rank2, args=(r, world, 29519, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
