@@ -179,10 +179,14 @@ class PartStore(dict):
       float64 keys (relational.encode_f64_sortable); decoded on read.
     """
 
-    def __init__(self, keyed=False, fkeys=False):
+    def __init__(self, keyed=False, fkeys=False, partitioned=True):
         super(PartStore, self).__init__()
         self.keyed = keyed
         self.fkeys = fkeys
+        # False = runs in pseudo-partition 0, not yet routed by key hash
+        # (lazy ingest: stages that need co-partitioned data call
+        # _ensure_partitioned; record-wise maps stream run by run)
+        self.partitioned = partitioned
 
 
 class HostStore(list):
@@ -531,7 +535,23 @@ class GpuRunner(RunnerBase):
                 lo = n * self.rank // self.world
                 hi = n * (self.rank + 1) // self.world
                 keys, vals = keys[lo:hi], vals[lo:hi]
-            return self._partition(keys, vals)
+                return self._partition(keys, vals)
+            if self.n_partitions == 1:
+                return self._partition(keys, vals)
+            # lazy ingest: batched unpartitioned runs — partitioning by
+            # the input keys is wasted work (and a wasted spill round
+            # trip) when the first stage re-keys anyway
+            store = PartStore(partitioned=False)
+            store[0] = []
+            n = keys.numel()
+            step = max(1, settings.gpu_batch_records)
+            for lo in range(0, n, step):
+                hi = min(lo + step, n)
+                run = DeviceRun(keys[lo:hi].contiguous(),
+                                vals[lo:hi].contiguous(), sorted=False)
+                store[0].append(run)
+                self.pool.admit(run)
+            return store
         # host dataset / chunker: stream records; numeric records become
         # columns, object records stay host-side (HostStore)
         records = self._host_records_of_input(inp)
@@ -692,6 +712,28 @@ class GpuRunner(RunnerBase):
             self.pool.admit(run)
         return store
 
+    def _ensure_partitioned(self, store):
+        """Route an unpartitioned store's runs to hash partitions, one
+        run at a time (each run is bounded, so this streams through the
+        pool)."""
+        if not isinstance(store, PartStore) or store.partitioned:
+            return store
+        out = None
+        for run in store.get(0, []):
+            self.pool.touch(run, self.device)
+            k, v = run.keys, run.vals
+            self.pool.release(run)
+            part = self._partition(k, v, keyed=store.keyed,
+                                   fkeys=store.fkeys)
+            if out is None:
+                out = part
+            else:
+                for q, runs in part.items():
+                    out.setdefault(q, []).extend(runs)
+        if out is None:
+            out = PartStore(keyed=store.keyed, fkeys=store.fkeys)
+        return out
+
     def _exchange(self, keys, vals, pid):
         """RCCL all-to-all: route rows to the partition's owning rank
         (p % world); returns this rank's rows."""
@@ -744,22 +786,37 @@ class GpuRunner(RunnerBase):
             # emit (keyfn(v), valfn(v)) per record — the group_by/count map
             _kind, keyf, valf = spec
             out = None
-            for p in self._parts(ins):
-                keys, vals = self._merged_partition(ins, p,
-                                                    need_sorted=False)
-                if keys is None:
-                    continue
+
+            def kv_batch(keys, vals):
                 nk = self._apply_colfunc(keyf, keys, vals)
                 nv = self._apply_colfunc(valf, keys, vals)
                 fkeys = nk.dtype == torch.float64
                 if fkeys:
                     nk = _encode_f64_sortable(nk)
-                out_store = self._partition(nk, nv, fkeys=fkeys)
+                return self._partition(nk, nv, fkeys=fkeys)
+
+            def fold(out, part):
                 if out is None:
-                    out = out_store
+                    return part
+                for q, runs in part.items():
+                    out.setdefault(q, []).extend(runs)
+                return out
+
+            for store in ins:
+                if isinstance(store, PartStore) and not store.partitioned:
+                    # record-wise op: stream run by run, no merge needed
+                    for run in store.get(0, []):
+                        self.pool.touch(run, self.device)
+                        k, v = run.keys, run.vals
+                        self.pool.release(run)
+                        out = fold(out, kv_batch(k, v))
                 else:
-                    for q, runs in out_store.items():
-                        out.setdefault(q, []).extend(runs)
+                    for p in self._parts([store]):
+                        keys, vals = self._merged_partition(
+                            [store], p, need_sorted=False)
+                        if keys is None:
+                            continue
+                        out = fold(out, kv_batch(keys, vals))
             return out if out is not None else PartStore()
         if kind == "identity":
             return self._merge_stores(ins)
@@ -798,11 +855,9 @@ class GpuRunner(RunnerBase):
             K = spec[1]
             cand_k, cand_v = [], []
             fkeys = False
-            for p in self._parts(ins):
-                keys, vals = self._merged_partition(ins, p,
-                                                    need_sorted=False)
-                if keys is None:
-                    continue
+
+            def topk_batch(vals):
+                nonlocal fkeys
                 if vals.dtype == torch.float64:
                     enc = _encode_f64_sortable(vals)
                     fkeys = True
@@ -813,6 +868,22 @@ class GpuRunner(RunnerBase):
                 top = min(K, sk.numel())
                 cand_k.append(sk[-top:])
                 cand_v.append(vals[sp.to(torch.int64)[-top:]])
+
+            for store in ins:
+                if isinstance(store, PartStore) \
+                        and not store.partitioned:
+                    for run in store.get(0, []):
+                        self.pool.touch(run, self.device)
+                        v = run.vals
+                        self.pool.release(run)
+                        topk_batch(v)
+                    continue
+                for p in self._parts([store]):
+                    keys, vals = self._merged_partition(
+                        [store], p, need_sorted=False)
+                    if keys is None:
+                        continue
+                    topk_batch(vals)
             store = PartStore(fkeys=fkeys)
             if not cand_k:
                 return store
@@ -847,9 +918,15 @@ class GpuRunner(RunnerBase):
             for s in stores:
                 out.extend(self._decode_store(s))
             return out
+        flags = [getattr(s, "partitioned", True) for s in stores]
+        if not all(flags) and any(flags):
+            # mixing hashed and unrouted partition-0 runs would corrupt
+            # co-location: route everything first
+            stores = [self._ensure_partitioned(s) for s in stores]
         out = PartStore(
             keyed=any(getattr(s, "keyed", False) for s in stores),
-            fkeys=any(getattr(s, "fkeys", False) for s in stores))
+            fkeys=any(getattr(s, "fkeys", False) for s in stores),
+            partitioned=all(flags))
         for s in stores:
             for p, runs in s.items():
                 out.setdefault(p, []).extend(runs)
@@ -859,6 +936,7 @@ class GpuRunner(RunnerBase):
 
     def run_reduce(self, stage, ins):
         spec = stage.options.get("device_reduce")
+        ins = [self._ensure_partitioned(s) for s in ins]
         if len(ins) == 1 and isinstance(ins[0], TokenStore) \
                 and spec == ("sum",):
             ins[0].keyed = True        # keyed-reducer output convention

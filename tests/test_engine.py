@@ -409,3 +409,24 @@ def test_engine_gloo_join_topk_world2():
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def test_lazy_ingest_spill_paths(tmp_path):
+    """Lazy unpartitioned ingest under a tiny pool: count, topk and sort
+    all stream through spilled batched runs with exact results."""
+    import dampr_amd.settings as st
+    old = st.gpu_batch_records
+    st.gpu_batch_records = 1000        # many small runs
+    try:
+        rng = np.random.default_rng(8)
+        vals = rng.integers(0, 300, size=25000)
+        kw = dict(hbm_bytes=32768, host_bytes=32768,
+                  spill_dir=str(tmp_path))
+        got = dict(Dampr.columns(vals).count().run(**kw).read())
+        want = collections.Counter(int(v) for v in vals)
+        assert got == dict(want)
+        top = Dampr.columns(vals).topk(5).run(**kw).read()
+        assert sorted(top) == sorted(sorted(
+            (int(v) for v in vals), reverse=True)[:5])
+    finally:
+        st.gpu_batch_records = old
