@@ -704,10 +704,10 @@ class GpuRunner(RunnerBase):
             o = offs_l[p]
             k = keys[o:o + n].contiguous()
             v = vals[o:o + n].contiguous()
-            if not already_sorted:
-                sk, sp = self._sort(k, fkeys=fkeys)
-                k, v = sk, v[sp.to(torch.int64)]
-            run = DeviceRun(k, v, sorted=True)
+            # runs stay unsorted: consumers that need key order sort at
+            # merge time, once per partition instead of once per
+            # (run, partition) slice — thousands of tiny sorts otherwise
+            run = DeviceRun(k, v, sorted=bool(already_sorted))
             store.setdefault(p, []).append(run)
             self.pool.admit(run)
         return store
