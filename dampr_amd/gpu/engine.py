@@ -949,10 +949,9 @@ class GpuRunner(RunnerBase):
         out = PartStore(keyed=True, fkeys=in_fkeys)
         if kind in ("sum", "min", "max"):
             for p in self._parts(ins):
-                keys, vals = self._merged_partition(ins, p)
-                if keys is None:
+                uk, agg = self._reduce_partition(ins, p, kind)
+                if uk is None:
                     continue
-                uk, agg = self.ops.seg_reduce_sorted(keys, vals, kind)
                 run = DeviceRun(uk, agg, sorted=True)
                 out.setdefault(p, []).append(run)
                 self.pool.admit(run)
@@ -1012,6 +1011,37 @@ class GpuRunner(RunnerBase):
             assert len(ins) == 2, "join takes two inputs"
             return self._device_join(ins[0], ins[1], how, stage)
         raise ValueError("unknown device_reduce spec {!r}".format(spec))
+
+    def _reduce_partition(self, ins, p, kind):
+        """Segmented reduce of one partition.  A skewed partition whose
+        runs exceed half the pool is reduced run-by-run and the partial
+        aggregates re-reduced (associativity) — bounded memory instead of
+        a giant merge (SURVEY §7 "skewed keys")."""
+        runs = [r for store in ins for r in store.get(p, [])]
+        if not runs:
+            return None, None
+        total = sum(r.nbytes for r in runs)
+        if total <= self.pool.capacity // 2 or len(runs) == 1:
+            keys, vals = self._merged_partition(ins, p)
+            return self.ops.seg_reduce_sorted(keys, vals, kind)
+        pk, pv = [], []
+        for run in runs:
+            self.pool.touch(run, self.device)
+            k, v = run.keys, run.vals
+            self.pool.release(run)
+            if not run.sorted:
+                fkeys = any(getattr(st, "fkeys", False) for st in ins)
+                k, sp = self._sort(k, fkeys=fkeys)
+                v = v[sp.to(torch.int64)]
+            uk, agg = self.ops.seg_reduce_sorted(k, v, kind)
+            pk.append(uk)
+            pv.append(agg)
+        keys = torch.cat(pk)
+        vals = torch.cat(pv)
+        fkeys = any(getattr(st, "fkeys", False) for st in ins)
+        sk, sp = self._sort(keys, fkeys=fkeys)
+        return self.ops.seg_reduce_sorted(sk, vals[sp.to(torch.int64)],
+                                          kind)
 
     def _device_join(self, left, right, how, stage):
         """Per-partition device hash join (K8); emits matched value pairs.

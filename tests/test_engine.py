@@ -430,3 +430,28 @@ def test_lazy_ingest_spill_paths(tmp_path):
             (int(v) for v in vals), reverse=True)[:5])
     finally:
         st.gpu_batch_records = old
+
+
+def test_skewed_partition_batched_reduce(tmp_path):
+    """A partition whose runs exceed half the pool reduces run-by-run
+    (associative re-reduce) instead of one giant merge."""
+    import dampr_amd.settings as st
+    old = st.gpu_batch_records
+    st.gpu_batch_records = 2000
+    try:
+        rng = np.random.default_rng(13)
+        vals = rng.integers(0, 50, size=30000)   # heavy duplication
+        got = dict(Dampr.columns(vals)
+                   .fold_by(funcs.identity, funcs.add)
+                   .run(hbm_bytes=16384, host_bytes=16384,
+                        spill_dir=str(tmp_path)).read())
+        groups = collections.defaultdict(int)
+        for v in vals:
+            groups[int(v)] += int(v)
+        assert got == dict(groups)
+        # min/max through the same path
+        got = dict(Dampr.columns(vals).a_group_by().reduce(min)
+                   .run(hbm_bytes=16384).read())
+        assert got == {int(v): int(v) for v in set(vals.tolist())}
+    finally:
+        st.gpu_batch_records = old
