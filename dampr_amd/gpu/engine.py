@@ -16,11 +16,16 @@ pools):
   host operators with decode/encode at the boundary (SURVEY.md §7 "hard
   parts": opaque UDFs run on the host, the shuffle/sort/combine core stays
   on device).
-* Out-of-core: partitions live in an ``HbmPool`` with a watermark; least
-  recently used runs spill to (pinned) host memory and are paged back per
-  partition at reduce time — the reference's RSS-watermark spill files
-  (dampr/memory.py, dataset.py:190-262) re-expressed at HBM granularity.
-  Only the active partition must be resident, so jobs scale past 288 GB.
+* Out-of-core: runs live in an ``HbmPool`` with two watermarks; least
+  recently used runs spill to pinned host memory and on to raw NVMe
+  files, paging back on touch — the reference's RSS-watermark spill
+  files (dampr/memory.py, dataset.py:190-262) re-expressed over HBM →
+  host → disk tiers.  Single-rank columnar ingest is *lazy*: batched
+  unrouted runs, streamed run-by-run through record-wise stages; only
+  key-colocating stages (reduce/join) route to hash partitions, and a
+  skewed partition larger than half the pool reduces run-by-run with an
+  associative re-reduce.  Only the active working set must be resident,
+  so jobs scale past device memory.
 * Multi-GPU: after every partitioning map stage the engine exchanges
   partitions to their owning rank over RCCL all-to-all (xGMI); partition p
   is owned by rank ``p % world`` (parallel/shuffle.py).
