@@ -552,8 +552,10 @@ class GpuRunner(RunnerBase):
             step = max(1, settings.gpu_batch_records)
             for lo in range(0, n, step):
                 hi = min(lo + step, n)
-                run = DeviceRun(keys[lo:hi].contiguous(),
-                                vals[lo:hi].contiguous(), sorted=False)
+                # clone, not view: a spilled run must actually release
+                # its HBM (views pin the whole parent tensor)
+                run = DeviceRun(keys[lo:hi].clone(),
+                                vals[lo:hi].clone(), sorted=False)
                 store[0].append(run)
                 self.pool.admit(run)
             return store
@@ -707,8 +709,10 @@ class GpuRunner(RunnerBase):
             if not n:
                 continue
             o = offs_l[p]
-            k = keys[o:o + n].contiguous()
-            v = vals[o:o + n].contiguous()
+            # clone, not view: spilled runs must release their HBM
+            # (views pin the whole routed batch)
+            k = keys[o:o + n].clone()
+            v = vals[o:o + n].clone()
             # runs stay unsorted: consumers that need key order sort at
             # merge time, once per partition instead of once per
             # (run, partition) slice — thousands of tiny sorts otherwise
