@@ -166,3 +166,33 @@ def test_device_text_df_fused():
                         .flat_map(funcs.tokenize_set).count()).read())
     want = oracle_df(arr)
     assert got == want
+
+
+def test_cross_engine_agreement_on_hardware():
+    """Host engine (Python operators) vs device engine (HipOps kernels)
+    on identical pipelines — the end-to-end numerics oracle on real
+    hardware."""
+    from dampr_amd.runner import MTRunner
+    rng = np.random.default_rng(77)
+    vals = rng.integers(-500, 500, size=3000)
+
+    def pipelines(pm):
+        return {
+            "count": pm.count(),
+            "sum": pm.a_group_by().sum(),
+            "min": pm.a_group_by().reduce(min),
+            "first": pm.a_group_by().first(),
+            "topk": pm.topk(9),
+            "sort": pm.sort_by(),
+            "mean": pm.mean(funcs.identity),
+        }
+
+    dev = {name: sorted(p.run(device="cuda:0").read())
+           for name, p in pipelines(Dampr.columns(vals)).items()}
+    host = {name: sorted(p.run(runner=MTRunner, n_maps=2,
+                               n_reducers=2).read())
+            for name, p in pipelines(
+                Dampr.memory(vals.tolist())).items()}
+    for name in dev:
+        assert dev[name] == host[name], (name, dev[name][:4],
+                                         host[name][:4])
