@@ -184,10 +184,17 @@ class PartStore(dict):
       float64 keys (relational.encode_f64_sortable); decoded on read.
     """
 
-    def __init__(self, keyed=False, fkeys=False, partitioned=True):
+    def __init__(self, keyed=False, fkeys=False, partitioned=True,
+                 str_table=None):
         super(PartStore, self).__init__()
         self.keyed = keyed
         self.fkeys = fkeys
+        # dictionary-encoded string keys: key column holds ranks into
+        # this sorted tuple (rank order == lexicographic order, so sorted
+        # device output decodes to host-ordered strings).  Stores with
+        # tables only combine with themselves; cross-store combinations
+        # fall back to host records (_decode_store).
+        self.str_table = str_table
         # False = runs in pseudo-partition 0, not yet routed by key hash
         # (lazy ingest: stages that need co-partitioned data call
         # _ensure_partitioned; record-wise maps stream run by run)
@@ -402,11 +409,13 @@ class ColumnDataset(object):
     composition.  ``keyed`` reproduces the host reducers' value
     convention (k, (k, v)); ``fkeys`` decodes float64 keys."""
 
-    def __init__(self, keys, vals, keyed=False, fkeys=False):
+    def __init__(self, keys, vals, keyed=False, fkeys=False,
+                 str_table=None):
         self.keys_t = keys
         self.vals_t = vals
         self.keyed = keyed
         self.fkeys = fkeys
+        self.str_table = str_table
 
     def columns(self):
         return self.keys_t, self.vals_t
@@ -416,6 +425,9 @@ class ColumnDataset(object):
         if self.fkeys:
             kt = _decode_f64_sortable(kt)
         k = kt.cpu().tolist()
+        if self.str_table is not None:
+            tbl = self.str_table
+            k = [tbl[i] for i in k]
         v = self.vals_t.cpu().tolist()
         if self.keyed:
             return iter((kk, (kk, vv)) for kk, vv in zip(k, v))
@@ -605,18 +617,28 @@ class GpuRunner(RunnerBase):
                 "device engine requires numeric values; use the host "
                 "engine for object records")
         fkeys = False
+        str_table = None
         if all(isinstance(k, (int, bool)) for k in ks):
             kt = torch.from_numpy(np.asarray(ks, dtype=np.int64))
         elif all(isinstance(k, (int, float, bool)) for k in ks):
             kt = _encode_f64_sortable(
                 torch.from_numpy(np.asarray(ks, dtype=np.float64)))
             fkeys = True
+        elif all(isinstance(k, str) for k in ks):
+            # dictionary encoding: rank ids preserve lexicographic order
+            str_table = tuple(sorted(set(ks)))
+            rank = {t: i for i, t in enumerate(str_table)}
+            kt = torch.from_numpy(
+                np.fromiter((rank[k] for k in ks), dtype=np.int64,
+                            count=len(ks)))
         else:
             raise TypeError(
-                "device engine requires numeric keys; use the host engine "
-                "for object records")
-        return self._partition(kt.to(self.device), vt.to(self.device),
-                               keyed=keyed, fkeys=fkeys)
+                "device engine requires numeric or string keys; use the "
+                "host engine for object records")
+        store = self._partition(kt.to(self.device), vt.to(self.device),
+                                keyed=keyed, fkeys=fkeys)
+        store.str_table = str_table
+        return store
 
     def _encode_or_host(self, records):
         try:
@@ -638,6 +660,7 @@ class GpuRunner(RunnerBase):
                                                      False))
         keyed = getattr(store, "keyed", False)
         fkeys = getattr(store, "fkeys", False)
+        tbl = getattr(store, "str_table", None)
         ks, vs = [], []
         for p in sorted(store):
             for run in store[p]:
@@ -647,11 +670,12 @@ class GpuRunner(RunnerBase):
                 self.pool.release(run)
         if not ks:
             z = torch.zeros(0, dtype=torch.int64)
-            return ColumnDataset(z, z.clone(), keyed, fkeys)
+            return ColumnDataset(z, z.clone(), keyed, fkeys, tbl)
         keys = torch.cat(ks)
         vals = torch.cat(vs)
         sk, sp = self._sort(keys, fkeys=fkeys)
-        return ColumnDataset(sk, vals[sp.to(torch.int64)], keyed, fkeys)
+        return ColumnDataset(sk, vals[sp.to(torch.int64)], keyed, fkeys,
+                             tbl)
 
     # -- ordering ----------------------------------------------------------
 
@@ -741,6 +765,7 @@ class GpuRunner(RunnerBase):
                     out.setdefault(q, []).extend(runs)
         if out is None:
             out = PartStore(keyed=store.keyed, fkeys=store.fkeys)
+        out.str_table = getattr(store, "str_table", None)
         return out
 
     def _exchange(self, keys, vals, pid):
@@ -927,6 +952,15 @@ class GpuRunner(RunnerBase):
             for s in stores:
                 out.extend(self._decode_store(s))
             return out
+        if len(stores) > 1 and any(
+                getattr(s, "str_table", None) is not None
+                for s in stores):
+            # rank ids from different encodes are incompatible: combine
+            # as host records (correct, slower)
+            out = HostStore()
+            for s in stores:
+                out.extend(self._decode_store(s))
+            return out
         flags = [getattr(s, "partitioned", True) for s in stores]
         if not all(flags) and any(flags):
             # mixing hashed and unrouted partition-0 runs would corrupt
@@ -945,6 +979,9 @@ class GpuRunner(RunnerBase):
 
     def run_reduce(self, stage, ins):
         spec = stage.options.get("device_reduce")
+        if len(ins) > 1 and any(
+                getattr(s, "str_table", None) is not None for s in ins):
+            return self._host_reduce(stage, ins)
         ins = [self._ensure_partitioned(s) for s in ins]
         if len(ins) == 1 and isinstance(ins[0], TokenStore) \
                 and spec == ("sum",):
@@ -955,7 +992,9 @@ class GpuRunner(RunnerBase):
             return self._host_reduce(stage, ins)
         kind = spec[0]
         in_fkeys = any(getattr(s, "fkeys", False) for s in ins)
-        out = PartStore(keyed=True, fkeys=in_fkeys)
+        out = PartStore(keyed=True, fkeys=in_fkeys,
+                        str_table=getattr(ins[0], "str_table", None)
+                        if len(ins) == 1 else None)
         if kind in ("sum", "min", "max"):
             for p in self._parts(ins):
                 uk, agg = self._reduce_partition(ins, p, kind)
@@ -1138,6 +1177,7 @@ class GpuRunner(RunnerBase):
     def _decode_store(self, store):
         if isinstance(store, HostStore):
             return list(store)
+        tbl = getattr(store, "str_table", None)
         if isinstance(store, TextSource):
             records = []
             pos = 0
@@ -1163,6 +1203,8 @@ class GpuRunner(RunnerBase):
             if fkeys:
                 keys = _decode_f64_sortable(keys)
             kl = keys.cpu().tolist()
+            if tbl is not None:
+                kl = [tbl[k] for k in kl]
             vl = vals.cpu().tolist()
             if keyed:
                 records.extend((k, (k, v)) for k, v in zip(kl, vl))

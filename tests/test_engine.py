@@ -455,3 +455,40 @@ def test_skewed_partition_batched_reduce(tmp_path):
         assert got == {int(v): int(v) for v in set(vals.tolist())}
     finally:
         st.gpu_batch_records = old
+
+
+def test_string_keys_device_dictionary():
+    """String keys dictionary-encode to rank ids: device count/sort with
+    host-ordered decode."""
+    from dampr_amd.gpu.engine import GpuRunner
+    words = ["pear", "apple", "pear", "fig", "apple", "pear"]
+    got = list(Dampr.memory(words).count().run(runner=GpuRunner).read())
+    assert got == [("apple", 2), ("fig", 1), ("pear", 3)]  # lex order
+
+    got = dict(Dampr.memory(words)
+               .fold_by(lambda w: w, lambda a, b: a + b,
+                        value=lambda _w: 1).run(runner=GpuRunner).read())
+    assert got == {"apple": 2, "fig": 1, "pear": 3}
+
+
+def test_string_keys_join_falls_back_correct():
+    """Rank ids from different encodes are incompatible; joins over
+    string-keyed stores combine as host records with exact results."""
+    from dampr_amd.gpu.engine import GpuRunner
+    left = Dampr.memory([("a", 1), ("b", 2)]).group_by(lambda x: x[0],
+                                                       lambda x: x[1])
+    right = Dampr.memory([("b", 10), ("c", 20)]).group_by(lambda x: x[0],
+                                                          lambda x: x[1])
+    out = left.join(right).reduce(lambda l, r: sum(l) + sum(r)) \
+        .run(runner=GpuRunner)
+    assert sorted(out.read()) == [("b", 12)]
+
+
+def test_string_keys_host_engine_parity():
+    from dampr_amd.gpu.engine import GpuRunner
+    from dampr_amd.runner import MTRunner
+    rng = np.random.default_rng(55)
+    words = ["w{}".format(int(i)) for i in rng.integers(0, 40, size=2000)]
+    dev = sorted(Dampr.memory(words).count().run(runner=GpuRunner).read())
+    host = sorted(Dampr.memory(words).count().run(runner=MTRunner).read())
+    assert dev == host
