@@ -196,3 +196,14 @@ def test_cross_engine_agreement_on_hardware():
     for name in dev:
         assert dev[name] == host[name], (name, dev[name][:4],
                                          host[name][:4])
+
+
+def test_string_keys_device_gpu():
+    from dampr_amd.gpu.engine import GpuRunner
+    rng = np.random.default_rng(60)
+    words = ["w{}".format(int(i))
+             for i in rng.integers(0, 5000, size=200_000)]
+    got = dict(Dampr.memory(words).count()
+               .run(runner=GpuRunner, device="cuda:0").read())
+    want = collections.Counter(words)
+    assert got == dict(want)
