@@ -699,7 +699,7 @@ class GpuRunner(RunnerBase):
         partition ids, stable sort by id, slice contiguous segments."""
         P = self.n_partitions
         store = PartStore(keyed=keyed, fkeys=fkeys)
-        if keys.numel() == 0:
+        if keys.numel() == 0 and self.world == 1:
             return store
         if P == 1 and self.world == 1:
             # no routing needed: one resident partition
@@ -836,21 +836,42 @@ class GpuRunner(RunnerBase):
                     out.setdefault(q, []).extend(runs)
                 return out
 
-            for store in ins:
-                if isinstance(store, PartStore) and not store.partitioned:
-                    # record-wise op: stream run by run, no merge needed
-                    for run in store.get(0, []):
-                        self.pool.touch(run, self.device)
-                        k, v = run.keys, run.vals
-                        self.pool.release(run)
-                        out = fold(out, kv_batch(k, v))
-                else:
-                    for p in self._parts([store]):
-                        keys, vals = self._merged_partition(
-                            [store], p, need_sorted=False)
-                        if keys is None:
-                            continue
-                        out = fold(out, kv_batch(keys, vals))
+            def batches():
+                for store in ins:
+                    if isinstance(store, PartStore) \
+                            and not store.partitioned:
+                        # record-wise op: stream run by run
+                        for run in store.get(0, []):
+                            self.pool.touch(run, self.device)
+                            k, v = run.keys, run.vals
+                            self.pool.release(run)
+                            yield k, v
+                    else:
+                        for p in self._parts([store]):
+                            keys, vals = self._merged_partition(
+                                [store], p, need_sorted=False)
+                            if keys is None:
+                                continue
+                            yield keys, vals
+
+            if self.world > 1:
+                # collectives must line up across ranks: apply the column
+                # funcs locally, then ONE exchange-partition for the
+                # whole stage (ranks may own different partition counts)
+                acc_k, acc_v = [], []
+                for k, v in batches():
+                    acc_k.append(self._apply_colfunc(keyf, k, v))
+                    acc_v.append(self._apply_colfunc(valf, k, v))
+                nk = torch.cat(acc_k) if acc_k else torch.zeros(
+                    0, dtype=torch.int64, device=self.device)
+                nv = torch.cat(acc_v) if acc_v else torch.zeros(
+                    0, dtype=torch.int64, device=self.device)
+                fkeys = nk.dtype == torch.float64
+                if fkeys:
+                    nk = _encode_f64_sortable(nk)
+                return self._partition(nk, nv, fkeys=fkeys)
+            for k, v in batches():
+                out = fold(out, kv_batch(k, v))
             return out if out is not None else PartStore()
         if kind == "identity":
             return self._merge_stores(ins)

@@ -193,11 +193,11 @@ def _engine_rank(rank, world, port, q):
         q.put((rank, traceback.format_exc()))
 
 
-def test_engine_gloo_world2():
+@pytest.mark.parametrize("world,port", [(2, 29517), (3, 29527)])
+def test_engine_gloo_multi_world(world, port):
     ctx = multiprocessing.get_context("spawn")
     q = ctx.Queue()
-    world = 2
-    procs = [ctx.Process(target=_engine_rank, args=(r, world, 29517, q))
+    procs = [ctx.Process(target=_engine_rank, args=(r, world, port, q))
              for r in range(world)]
     for p in procs:
         p.start()
@@ -396,11 +396,11 @@ def _engine_rank2(rank, world, port, q):
         q.put((rank, traceback.format_exc()))
 
 
-def test_engine_gloo_join_topk_world2():
+@pytest.mark.parametrize("world,port", [(2, 29519), (3, 29529)])
+def test_engine_gloo_join_topk(world, port):
     ctx = multiprocessing.get_context("spawn")
     q = ctx.Queue()
-    world = 2
-    procs = [ctx.Process(target=_engine_rank2, args=(r, world, 29519, q))
+    procs = [ctx.Process(target=_engine_rank2, args=(r, world, port, q))
              for r in range(world)]
     for p in procs:
         p.start()
