@@ -44,6 +44,12 @@ def main():
     top = Dampr.columns(rng.standard_normal(1_000_000)).topk(5).run("topk")
     print("top-5:", top.read())
 
+    # string keys dictionary-encode onto the device kernels
+    words = ["w%d" % i for i in rng.integers(0, 500, size=200_000)]
+    from dampr_amd.gpu.engine import GpuRunner
+    wc = Dampr.memory(words).count().run("wc", runner=GpuRunner)
+    print("distinct words:", len(wc.read()))
+
 
 if __name__ == "__main__":
     main()

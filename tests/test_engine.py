@@ -492,3 +492,14 @@ def test_string_keys_host_engine_parity():
     dev = sorted(Dampr.memory(words).count().run(runner=GpuRunner).read())
     host = sorted(Dampr.memory(words).count().run(runner=MTRunner).read())
     assert dev == host
+
+
+def test_mixed_sources_one_run():
+    """Dampr.run over a graph mixing columnar and host-memory inputs:
+    the device engine carries both (HostStore for object records)."""
+    from dampr_amd.gpu.engine import GpuRunner
+    a = Dampr.columns(np.array([1, 1, 2], dtype=np.int64)).count()
+    b = Dampr.memory(["x", "y", "x"]).count()
+    ea, eb = Dampr.run(a, b, runner=GpuRunner)
+    assert sorted(ea.read()) == [(1, 2), (2, 1)]
+    assert sorted(eb.read()) == [("x", 2), ("y", 1)]
