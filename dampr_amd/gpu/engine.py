@@ -1027,14 +1027,15 @@ class GpuRunner(RunnerBase):
             return out
         if kind == "mean":
             for p in self._parts(ins):
-                keys, vals = self._merged_partition(ins, p)
-                if keys is None:
+                uk, sm = self._reduce_partition(
+                    ins, p, "sum", vt=lambda v: v.to(torch.float64))
+                if uk is None:
                     continue
-                uk, s = self.ops.seg_reduce_sorted(
-                    keys, vals.to(torch.float64), "sum")
-                _uk, c = self.ops.seg_reduce_sorted(
-                    keys, torch.ones_like(keys), "sum")
-                run = DeviceRun(uk, s / c.to(torch.float64), sorted=True)
+                _uk, c = self._reduce_partition(
+                    ins, p, "sum",
+                    vt=lambda v: torch.ones_like(v,
+                                                 dtype=torch.float64))
+                run = DeviceRun(uk, sm / c, sorted=True)
                 out.setdefault(p, []).append(run)
                 self.pool.admit(run)
             return out
@@ -1042,14 +1043,10 @@ class GpuRunner(RunnerBase):
             # first value per key: stable sorts keep insertion order
             # within equal keys, so the segment head is the first seen
             for p in self._parts(ins):
-                keys, vals = self._merged_partition(ins, p)
-                if keys is None:
+                uk, fv = self._reduce_partition(ins, p, "first")
+                if uk is None:
                     continue
-                _seg, uk = self.ops.segment_ids(keys)
-                flags = torch.ones_like(keys)
-                flags[1:] = (keys[1:] != keys[:-1]).to(torch.int64)
-                first_idx = torch.nonzero(flags.bool()).flatten()
-                run = DeviceRun(uk, vals[first_idx], sorted=True)
+                run = DeviceRun(uk, fv, sorted=True)
                 out.setdefault(p, []).append(run)
                 self.pool.admit(run)
             return out
@@ -1081,36 +1078,50 @@ class GpuRunner(RunnerBase):
             return self._device_join(ins[0], ins[1], how, stage)
         raise ValueError("unknown device_reduce spec {!r}".format(spec))
 
-    def _reduce_partition(self, ins, p, kind):
-        """Segmented reduce of one partition.  A skewed partition whose
+    def _first_sorted(self, keys, vals):
+        """(unique_keys, first value per key) over a key-sorted column;
+        stable sorts preserve insertion order within equal keys."""
+        _seg, uk = self.ops.segment_ids(keys)
+        flags = torch.ones_like(keys)
+        flags[1:] = (keys[1:] != keys[:-1]).to(torch.int64)
+        first_idx = torch.nonzero(flags.bool()).flatten()
+        return uk, vals[first_idx]
+
+    def _reduce_partition(self, ins, p, kind, vt=None):
+        """One partition's reduce (kind: sum/min/max/first; ``vt``
+        transforms the value column first).  A skewed partition whose
         runs exceed half the pool is reduced run-by-run and the partial
-        aggregates re-reduced (associativity) — bounded memory instead of
-        a giant merge (SURVEY §7 "skewed keys")."""
+        aggregates re-reduced (associativity; "first" re-reduces in run
+        order under stable sorts) — bounded memory instead of a giant
+        merge (SURVEY §7 "skewed keys")."""
+        def one(k_sorted, v):
+            if kind == "first":
+                return self._first_sorted(k_sorted, v)
+            return self.ops.seg_reduce_sorted(k_sorted, v, kind)
+
         runs = [r for store in ins for r in store.get(p, [])]
         if not runs:
             return None, None
+        fkeys = any(getattr(st, "fkeys", False) for st in ins)
         total = sum(r.nbytes for r in runs)
         if total <= self.pool.capacity // 2 or len(runs) == 1:
             keys, vals = self._merged_partition(ins, p)
-            return self.ops.seg_reduce_sorted(keys, vals, kind)
+            return one(keys, vt(vals) if vt else vals)
         pk, pv = [], []
         for run in runs:
             self.pool.touch(run, self.device)
             k, v = run.keys, run.vals
             self.pool.release(run)
             if not run.sorted:
-                fkeys = any(getattr(st, "fkeys", False) for st in ins)
                 k, sp = self._sort(k, fkeys=fkeys)
                 v = v[sp.to(torch.int64)]
-            uk, agg = self.ops.seg_reduce_sorted(k, v, kind)
+            uk, agg = one(k, vt(v) if vt else v)
             pk.append(uk)
             pv.append(agg)
         keys = torch.cat(pk)
         vals = torch.cat(pv)
-        fkeys = any(getattr(st, "fkeys", False) for st in ins)
         sk, sp = self._sort(keys, fkeys=fkeys)
-        return self.ops.seg_reduce_sorted(sk, vals[sp.to(torch.int64)],
-                                          kind)
+        return one(sk, vals[sp.to(torch.int64)])
 
     def _device_join(self, left, right, how, stage):
         """Per-partition device hash join (K8); emits matched value pairs.

@@ -503,3 +503,26 @@ def test_mixed_sources_one_run():
     ea, eb = Dampr.run(a, b, runner=GpuRunner)
     assert sorted(ea.read()) == [(1, 2), (2, 1)]
     assert sorted(eb.read()) == [("x", 2), ("y", 1)]
+
+
+def test_skewed_mean_and_first(tmp_path):
+    import dampr_amd.settings as st
+    old = st.gpu_batch_records
+    st.gpu_batch_records = 1500
+    try:
+        rng = np.random.default_rng(17)
+        vals = rng.integers(0, 40, size=20000)
+        kw = dict(hbm_bytes=16384, host_bytes=16384,
+                  spill_dir=str(tmp_path))
+        got = dict(Dampr.columns(vals).mean(funcs.identity)
+                   .run(**kw).read())
+        groups = collections.defaultdict(list)
+        for v in vals:
+            groups[int(v)].append(int(v))
+        want = {k: sum(g) / float(len(g)) for k, g in groups.items()}
+        assert got == want
+        got = dict(Dampr.columns(vals).a_group_by().first()
+                   .run(**kw).read())
+        assert got == {int(v): int(v) for v in set(vals.tolist())}
+    finally:
+        st.gpu_batch_records = old
