@@ -336,6 +336,9 @@ class HbmPool(object):
         self._run_tag = "dampr_amd_{}".format(uuid.uuid4().hex[:10])
         self._file_ctr = 0
         self._disk_paths = []
+        self.spilled_host = 0
+        self.spilled_disk = 0
+        self.reloaded = 0
         self._os = os
 
     def _next_path(self):
@@ -357,6 +360,12 @@ class HbmPool(object):
                 pass
         self._disk_paths = []
 
+    def stats(self):
+        return {"hbm_used": self.used, "host_used": self.host_used,
+                "spilled_to_host_bytes": self.spilled_host,
+                "spilled_to_disk_bytes": self.spilled_disk,
+                "reloads_bytes": self.reloaded}
+
     def admit(self, run):
         self.used += run.nbytes
         self._lru.append(run)
@@ -367,6 +376,7 @@ class HbmPool(object):
             if run in self._host_lru:
                 self._host_lru.remove(run)
                 self.host_used -= run.nbytes
+            self.reloaded += run.nbytes
             run.load(device)
             self.used += run.nbytes
             self._lru.append(run)
@@ -391,11 +401,13 @@ class HbmPool(object):
             self._lru.remove(victim)
             self.used -= victim.nbytes
             victim.spill()
+            self.spilled_host += victim.nbytes
             self.host_used += victim.nbytes
             self._host_lru.append(victim)
         while self.host_used > self.host_capacity and self._host_lru:
             v = self._host_lru.pop(0)
             self.host_used -= v.nbytes
+            self.spilled_disk += v.nbytes
             v.spill_to_disk(self._next_path())
 
 
@@ -478,6 +490,7 @@ class GpuRunner(RunnerBase):
             self.rank = torch.distributed.get_rank()
         else:
             self.world, self.rank = 1, 0
+        self.exchanged_rows = 0
         if n_partitions:
             self.n_partitions = n_partitions
         elif self.world == 1 and self._inputs_fit(cap):
@@ -528,6 +541,11 @@ class GpuRunner(RunnerBase):
         for source in outputs:
             store = data[source]
             rets.append(self._collect(store))
+        st = self.pool.stats()
+        st["exchanged_rows"] = self.exchanged_rows
+        self.stats = st
+        if st["spilled_to_host_bytes"] or st["exchanged_rows"]:
+            log.info("[device] run stats: %s", st)
         self.pool.cleanup()
         return rets
 
@@ -772,6 +790,7 @@ class GpuRunner(RunnerBase):
         """RCCL all-to-all: route rows to the partition's owning rank
         (p % world); returns this rank's rows."""
         from ..parallel.shuffle import exchange_columns
+        self.exchanged_rows += keys.numel()
         return exchange_columns(keys, vals, pid, self.world)
 
     def _merged_partition(self, stores, p, need_sorted=True):

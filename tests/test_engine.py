@@ -526,3 +526,17 @@ def test_skewed_mean_and_first(tmp_path):
         assert got == {int(v): int(v) for v in set(vals.tolist())}
     finally:
         st.gpu_batch_records = old
+
+
+def test_pool_stats_exposed(tmp_path):
+    from dampr_amd.gpu.engine import GpuRunner
+    rng = np.random.default_rng(19)
+    vals = rng.integers(0, 100, size=20000)
+    pm = Dampr.columns(vals).count().checkpoint()
+    runner = GpuRunner("statrun", pm.pmer.graph, hbm_bytes=8192,
+                       host_bytes=8192, spill_dir=str(tmp_path))
+    runner.run([pm.source])
+    st = runner.stats
+    assert st["spilled_to_host_bytes"] > 0
+    assert st["reloads_bytes"] > 0
+    assert "spilled_to_disk_bytes" in st
