@@ -348,9 +348,19 @@ class PMap(PBase):
             if seen:
                 yield 1, count
 
-        return self.partition_map(_map_count) \
-                   .partition_reduce(_reduce_count) \
-                   .map(lambda x: x[1])
+        if self.agg:
+            return self.partition_map(_map_count) \
+                       .partition_reduce(_reduce_count) \
+                       .map(lambda x: x[1])
+        # no pending opaque maps: the device engine answers from run
+        # lengths without reading any data
+        me = self._add_mapper(StreamMapper(_map_count)) \
+            .checkpoint(options={"device_map": ("len_local",)})
+        source, pmer = me.pmer._add_reducer(
+            [me.source], StreamReducer(_reduce_count),
+            options={"device_reduce": ("sum",)})
+        out = PMap(source, pmer).map(lambda x: x[1])
+        return out.checkpoint(options={"device_map": ("unkey",)})
 
     def topk(self, k, value=None):
         """Top-k values ordered by value(x) (K11).

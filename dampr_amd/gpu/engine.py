@@ -241,6 +241,15 @@ class DeviceRun(object):
         self._meta = None
 
     @property
+    def n(self):
+        """Row count, available without paging the run in."""
+        if self.keys is not None:
+            return self.keys.numel()
+        if self._meta is not None:
+            return self._meta[0]
+        return 0
+
+    @property
     def nbytes(self):
         if self.keys is not None:
             return self.keys.numel() * 8 + self.vals.element_size() * \
@@ -894,6 +903,48 @@ class GpuRunner(RunnerBase):
             return out if out is not None else PartStore()
         if kind == "identity":
             return self._merge_stores(ins)
+        if kind == "unkey":
+            # strip the keyed-reducer value convention: values become the
+            # bare aggregates (the host stage extracts x[1])
+            merged = self._merge_stores(ins)
+            if isinstance(merged, (HostStore, TokenStore)) \
+                    or not isinstance(merged, PartStore):
+                return self._host_map(stage, ins)
+            out = PartStore(keyed=False, fkeys=merged.fkeys,
+                            partitioned=merged.partitioned,
+                            str_table=merged.str_table)
+            for q, runs in merged.items():
+                out[q] = runs
+            return out
+        if kind == "len_local":
+            # row count from run metadata: no data is read at all
+            total = 0
+            ok = True
+            for store in ins:
+                if isinstance(store, PartStore):
+                    for runs in store.values():
+                        for r in runs:
+                            total += r.n
+                else:
+                    ok = False
+            if not ok:
+                return self._host_map(stage, ins)
+            if self.world > 1:
+                keys = torch.ones(1, dtype=torch.int64,
+                                  device=self.device)
+                vals = torch.tensor([total], dtype=torch.int64,
+                                    device=self.device)
+                return self._partition(keys, vals)
+            if total == 0:
+                return PartStore()
+            store = PartStore()
+            run = DeviceRun(
+                torch.ones(1, dtype=torch.int64, device=self.device),
+                torch.tensor([total], dtype=torch.int64,
+                             device=self.device), sorted=True)
+            store[0] = [run]
+            self.pool.admit(run)
+            return store
         if kind == "text_df":
             src = ins[0]
             if not isinstance(src, TextSource) or \

@@ -540,3 +540,14 @@ def test_pool_stats_exposed(tmp_path):
     assert st["spilled_to_host_bytes"] > 0
     assert st["reloads_bytes"] > 0
     assert "spilled_to_disk_bytes" in st
+
+
+def test_len_device_metadata():
+    rng = np.random.default_rng(23)
+    vals = rng.integers(0, 10, size=12345)
+    got = Dampr.columns(vals).len().run().read()
+    assert got == [12345]
+    # host parity
+    from dampr_amd.runner import MTRunner
+    host = Dampr.memory(vals.tolist()).len().run(runner=MTRunner).read()
+    assert host == [12345]
