@@ -381,6 +381,13 @@ def _engine_rank2(rank, world, port, q):
                     want.append((int(lk[i]), int(lv[i] + rv[j])))
         assert merged == sorted(want), (len(merged), len(want))
 
+        # device len across ranks: local counts summed on the owner rank
+        nv = np.arange(4000)
+        got_len = Dampr.columns(nv).len().run().read()
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got_len)
+        assert sorted(x for lst in gathered for x in lst) == [4000]
+
         # device topk across ranks: candidates meet on rank 0
         vals = np.arange(5000)
         got = Dampr.columns(vals).topk(7).run().read()
