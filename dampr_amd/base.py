@@ -2,8 +2,9 @@
 
 Role parity with the reference's ``dampr/base.py``.  Notable deltas:
 
-* ``Splitter`` uses the stable ``keyhash.partition_of`` (device-reproducible)
-  instead of Python's salted ``hash(key) % n`` (reference: base.py:6-8).
+* ``Splitter`` uses the builtin salted hash like the reference (C speed;
+  fork-stable within a run) — ``StableSplitter`` is the opt-in
+  device-reproducible form over ``keyhash``.
 * ``OuterJoin`` works — the reference's is dead code with an undefined
   variable and a wrong drain loop (reference: base.py:337-366, SURVEY.md
   §2.5).
