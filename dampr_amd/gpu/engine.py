@@ -380,6 +380,33 @@ class HbmPool(object):
         self._lru.append(run)
         self.balance()
 
+    def prefetch(self, runs, device, stream):
+        """Begin paging spilled runs back on a side HIP stream (overlaps
+        the H2D copies with the current partition's compute).  The caller
+        must make its stream wait on ``stream`` before consuming."""
+        if stream is None:
+            return
+        main = torch.cuda.current_stream(device)
+        with torch.cuda.stream(stream):
+            for run in runs:
+                if run.resident:
+                    continue
+                if run in self._host_lru:
+                    self._host_lru.remove(run)
+                    self.host_used -= run.nbytes
+                self.reloaded += run.nbytes
+                run.load(device)
+                # tensors are allocated on the side stream but consumed
+                # on the main stream: mark the cross-stream use so the
+                # caching allocator does not reuse them early
+                run.keys.record_stream(main)
+                run.vals.record_stream(main)
+                self.used += run.nbytes
+                self._lru.append(run)
+        # NOTE: no balance() here — eviction during an in-flight copy
+        # could spill the very runs being loaded; the next touch() call
+        # rebalances on the main stream.
+
     def touch(self, run, device):
         if not run.resident:
             if run in self._host_lru:
@@ -500,6 +527,8 @@ class GpuRunner(RunnerBase):
         else:
             self.world, self.rank = 1, 0
         self.exchanged_rows = 0
+        self._side_stream = (torch.cuda.Stream(device=self.device)
+                             if self.device.type == "cuda" else None)
         if n_partitions:
             self.n_partitions = n_partitions
         elif self.world == 1 and self._inputs_fit(cap):
@@ -1087,7 +1116,10 @@ class GpuRunner(RunnerBase):
                         str_table=getattr(ins[0], "str_table", None)
                         if len(ins) == 1 else None)
         if kind in ("sum", "min", "max"):
-            for p in self._parts(ins):
+            parts = self._parts(ins)
+            for i, p in enumerate(parts):
+                self._prefetch_partition(ins, parts, i + 1)
+                self._wait_prefetch()
                 uk, agg = self._reduce_partition(ins, p, kind)
                 if uk is None:
                     continue
@@ -1147,6 +1179,22 @@ class GpuRunner(RunnerBase):
             assert len(ins) == 2, "join takes two inputs"
             return self._device_join(ins[0], ins[1], how, stage)
         raise ValueError("unknown device_reduce spec {!r}".format(spec))
+
+    def _prefetch_partition(self, ins, parts, i):
+        """Kick off async page-in of partition parts[i] on the side
+        stream (no-op without a CUDA device)."""
+        if self._side_stream is None or i >= len(parts):
+            return
+        nxt = parts[i]
+        runs = [r for store in ins
+                if isinstance(store, PartStore)
+                for r in store.get(nxt, [])]
+        self.pool.prefetch(runs, self.device, self._side_stream)
+
+    def _wait_prefetch(self):
+        if self._side_stream is not None:
+            torch.cuda.current_stream(self.device).wait_stream(
+                self._side_stream)
 
     def _first_sorted(self, keys, vals):
         """(unique_keys, first value per key) over a key-sorted column;
