@@ -558,6 +558,14 @@ class GpuRunner(RunnerBase):
         from ..utils.trace import get_trace, trace_stage
         get_trace().clear()
         data = {}
+        # consumer refcounts: a store is freed after its LAST consuming
+        # stage (keeps true >HBM jobs from pinning dead intermediates)
+        consumers = {}
+        for stage in self.graph.stages:
+            for src in stage.inputs:
+                consumers[src] = consumers.get(src, 0) + 1
+        for src in outputs:
+            consumers[src] = consumers.get(src, 0) + 1
         for src, inp in self.graph.inputs.items():
             with trace_stage("ingest {}".format(src), self.device):
                 data[src] = self._ingest(inp)
@@ -575,6 +583,21 @@ class GpuRunner(RunnerBase):
                 else:
                     raise TypeError(stage)
             data[stage.output] = out
+            if cleanup:
+                for src in set(stage.inputs):
+                    consumers[src] -= stage.inputs.count(src)
+                    if consumers.get(src, 0) <= 0:
+                        victim = data.get(src)
+                        # pass-through stages (identity/unkey/merges)
+                        # share RUN objects between stores: free only
+                        # runs not referenced by any other live source
+                        live = {id(r)
+                                for k, v in data.items()
+                                if k != src and isinstance(v, PartStore)
+                                for runs in v.values()
+                                for r in runs}
+                        self._free_store(victim, live)
+                        data[src] = PartStore()
         rets = []
         for source in outputs:
             store = data[source]
@@ -586,6 +609,36 @@ class GpuRunner(RunnerBase):
             log.info("[device] run stats: %s", st)
         self.pool.cleanup()
         return rets
+
+    def _free_store(self, store, live_ids=frozenset()):
+        """Release a fully-consumed store's memory across all tiers
+        (skipping runs still shared with live stores)."""
+        if not isinstance(store, PartStore):
+            return
+        for part in list(store):
+            kept = []
+            for run in store[part]:
+                if id(run) in live_ids:
+                    kept.append(run)
+                    continue
+                self.pool.release(run)
+                if run in self.pool._host_lru:
+                    self.pool._host_lru.remove(run)
+                    self.pool.host_used -= run.nbytes
+                if run.on_disk:
+                    try:
+                        self.pool._os.unlink(run._disk)
+                    except OSError:
+                        pass
+                    run._disk = None
+                run.keys = None
+                run.vals = None
+                run._host = None
+                run._meta = None
+            if kept:
+                store[part] = kept
+            else:
+                del store[part]
 
     # -- ingest / collect --------------------------------------------------
 

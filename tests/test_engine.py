@@ -558,3 +558,23 @@ def test_len_device_metadata():
     from dampr_amd.runner import MTRunner
     host = Dampr.memory(vals.tolist()).len().run(runner=MTRunner).read()
     assert host == [12345]
+
+
+def test_eager_free_releases_consumed_runs(tmp_path):
+    """Intermediate stores free after their last consumer; shared/output
+    runs survive (identity/unkey aliasing)."""
+    from dampr_amd.gpu.engine import GpuRunner
+    rng = np.random.default_rng(29)
+    vals = rng.integers(0, 50, size=5000)
+    pm = Dampr.columns(vals).count()
+    runner = GpuRunner("freerun", pm.pmer.graph, spill_dir=str(tmp_path))
+    before = runner.pool.used
+    [ds] = runner.run([pm.source])
+    # output intact
+    assert sorted(r for r in ds.read()) == sorted(
+        (int(k), (int(k), int(c)))
+        for k, c in __import__("collections").Counter(
+            int(v) for v in vals).items())
+    # only the output's bytes remain accounted in the pool
+    out_bytes = ds.keys_t.numel() * 8 + ds.vals_t.numel() * 8
+    assert runner.pool.used <= out_bytes * 2 + 4096
