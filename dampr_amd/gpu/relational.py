@@ -39,13 +39,17 @@ def radix_sort_pairs(keys, payload=None):
     # two owned buffers (using the input as scratch would clobber it).
     src_k, src_p = keys.contiguous(), payload.contiguous()
     out_k, out_p = torch.empty_like(src_k), torch.empty_like(src_p)
+    # one read decides which byte passes are constant (skippable): a
+    # single host sync instead of one per pass, and no per-pass
+    # exploratory histograms
+    hist8 = ext.rs_hist_global(src_k).cpu()
+    active = [int((hist8[b] != 0).sum().item()) > 1 for b in range(8)]
     n_done = 0
     for byte in range(8):
         shift = byte * 8
-        hist = ext.rs_hist(src_k, shift, nblocks).to(torch.int64)
-        per_bin = hist.view(256, nblocks).sum(1)
-        if int((per_bin != 0).sum().item()) <= 1:
+        if not active[byte]:
             continue                      # constant digit: skip pass
+        hist = ext.rs_hist(src_k, shift, nblocks).to(torch.int64)
         scanned = torch.cumsum(hist, 0) - hist
         ext.rs_scatter(src_k, src_p, scanned, shift, nblocks, out_k, out_p)
         n_done += 1

@@ -110,6 +110,41 @@ __global__ void rs_scatter_kernel(const u64* __restrict__ keys,
     }
 }
 
+// All 8 byte-histograms in one read (skip-pass detection): 8x256 LDS
+// counters per block, atomically folded into a [8][256] global table.
+__global__ void rs_hist_global_kernel(const u64* __restrict__ keys, long n,
+                                      u32* __restrict__ hist8) {
+    __shared__ u32 h[8][256];
+    for (int i = threadIdx.x; i < 8 * 256; i += blockDim.x)
+        (&h[0][0])[i] = 0;
+    __syncthreads();
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        const u64 k = keys[i];
+        #pragma unroll
+        for (int b = 0; b < 8; ++b)
+            atomicAdd(&h[b][(u32)((k >> (8 * b)) & 255)], 1u);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < 8 * 256; i += blockDim.x)
+        if ((&h[0][0])[i])
+            atomicAdd(&hist8[i], (&h[0][0])[i]);
+}
+
+torch::Tensor rs_hist_global(torch::Tensor keys) {
+    auto hist8 = torch::zeros({8, 256},
+        torch::TensorOptions().dtype(torch::kUInt32)
+            .device(keys.device()));
+    long n = keys.numel();
+    if (n)
+        hipLaunchKernelGGL(rs_hist_global_kernel,
+            dim3(grid_for(n, RS_BLOCK, 2048)), dim3(RS_BLOCK), 0,
+            cur_stream(), (const u64*)keys.data_ptr(), n,
+            (u32*)hist8.data_ptr());
+    return hist8;
+}
+
 torch::Tensor rs_hist(torch::Tensor keys, long shift, long nblocks) {
     auto hist = torch::empty({256L * nblocks},
         torch::TensorOptions().dtype(torch::kUInt32)
