@@ -133,8 +133,10 @@ __global__ void rs_hist_global_kernel(const u64* __restrict__ keys, long n,
 }
 
 torch::Tensor rs_hist_global(torch::Tensor keys) {
+    // int32 storage (barebones-uint32 fill kernels are not reliable on
+    // every backend); the kernel's u32 atomics are bit-identical
     auto hist8 = torch::zeros({8, 256},
-        torch::TensorOptions().dtype(torch::kUInt32)
+        torch::TensorOptions().dtype(torch::kInt)
             .device(keys.device()));
     long n = keys.numel();
     if (n)
