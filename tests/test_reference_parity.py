@@ -76,3 +76,18 @@ def test_reference_parity(name, engines):
     got = sorted(map(repr, build(OursD).run().read()))
     assert want, "vacuous comparison"
     assert got == want, (name, got[:4], want[:4])
+
+
+@pytest.mark.parametrize("name", sorted(PIPELINES))
+def test_reference_parity_device_engine(name, engines):
+    """Same pipelines through the device engine (TorchOps here; HipOps
+    on GPU boxes) — fallback and columnar paths must both match the
+    reference."""
+    from dampr_amd.gpu.engine import GpuRunner
+    RefD, OursD = engines
+    build = PIPELINES[name]
+    want = sorted(map(repr, build(RefD).run().read()))
+    got = sorted(map(repr,
+                     build(OursD).run(runner=GpuRunner).read()))
+    assert want, "vacuous comparison"
+    assert got == want, (name, got[:4], want[:4])
