@@ -8,8 +8,8 @@ projected distance to the watermark, clamped to the settings bounds.  The
 reference's broken ``ExponentialMemoryChecker`` has no
 analog here.
 
-The GPU engine uses the same control law over HBM pool occupancy instead of
-RSS (see gpu/pool.py).
+The GPU engine applies the same watermark idea over HBM pool occupancy
+instead of RSS (gpu/engine.py ``HbmPool``).
 """
 import os
 
