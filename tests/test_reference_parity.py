@@ -91,3 +91,86 @@ def test_reference_parity_device_engine(name, engines):
                      build(OursD).run(runner=GpuRunner).read()))
     assert want, "vacuous comparison"
     assert got == want, (name, got[:4], want[:4])
+
+
+def test_parity_multi_output_shared_subgraph(engines, tmp_path):
+    """Dampr.run with a shared checkpointed root (word-stats idiom)."""
+    RefD, OursD = engines
+
+    def build(D):
+        words = D.memory(["a b", "b c b", "a"]) \
+            .flat_map(lambda s: s.split())
+        counts = words.count()
+        total = counts.fold_by(lambda _wc: 1, lambda x, y: x + y,
+                               value=lambda wc: wc[1])
+        return D.run(counts, total)
+
+    ref_counts, ref_total = build(RefD)
+    our_counts, our_total = build(OursD)
+    assert sorted(our_counts.read()) == sorted(ref_counts.read())
+    assert sorted(our_total.read()) == sorted(ref_total.read())
+
+
+def test_parity_text_input(engines, tmp_path):
+    RefD, OursD = engines
+    f = tmp_path / "corpus.txt"
+    f.write_text("x y z\nz z y\n" * 50)
+    want = sorted(RefD.text(str(f))
+                  .flat_map(lambda l: l.split()).count().run().read())
+    got = sorted(OursD.text(str(f))
+                 .flat_map(lambda l: l.split()).count().run().read())
+    assert got == want and want
+
+
+def test_parity_sink(engines, tmp_path):
+    RefD, OursD = engines
+    rd = tmp_path / "ref"
+    od = tmp_path / "ours"
+    RefD.memory(list(range(20))).map(str).sink(str(rd)).run()
+    OursD.memory(list(range(20))).map(str).sink(str(od)).run()
+
+    def lines(d):
+        out = []
+        for fn in sorted(os.listdir(d)):
+            with open(os.path.join(d, fn)) as fh:
+                out.extend(ln.strip() for ln in fh if ln.strip())
+        return sorted(out)
+
+    assert lines(od) == lines(rd) and lines(od)
+
+
+def test_parity_partition_map_reduce(engines):
+    RefD, OursD = engines
+
+    def build(D):
+        def pm(values):
+            m = 0
+            for v in values:
+                m = max(m, v)
+            yield 1, m
+
+        def pr(groups):
+            best = 0
+            for _k, vs in groups:
+                for v in vs:
+                    best = max(best, v)
+            yield 1, best
+
+        return D.memory(list(range(100))) \
+            .partition_map(pm).partition_reduce(pr)
+
+    want = sorted(map(repr, build(RefD).run().read()))
+    got = sorted(map(repr, build(OursD).run().read()))
+    assert got == want and want
+
+
+def test_parity_cached(engines):
+    RefD, OursD = engines
+
+    def build(D):
+        base = D.memory(list(range(30))).map(lambda v: v * 2).cached()
+        return base.count(lambda v: v % 3)
+
+    want = sorted(build(RefD).run().read())
+    got = sorted(build(OursD).run().read())
+    assert got == want and want
