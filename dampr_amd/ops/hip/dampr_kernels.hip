@@ -281,283 +281,6 @@ __device__ __forceinline__ int lds_set_insert(u64* set, u64 h) {
     return -1;
 }
 
-#ifdef DOC_PAIR
-// Interleaved-pair variant: each wave processes TWO adjacent doc groups,
-// alternating 64-byte windows so the two independent cross-lane/LDS
-// dependency chains overlap (the kernel is wait-bound: SQ_WAIT_ANY ~15x
-// SQ_BUSY on the single-group version).  Same algorithm per group as the
-// default kernel below; selected with -DDOC_PAIR=1 (sweep variant).
-
-struct EmitCtx {
-    u64* cck; u32* ccv;
-    u64* cnt_keys; u64* cnt_vals; u64 cnt_mask;
-    u64* dict_keys; u64* dict_vals; u64 dict_mask;
-    u64* fb_seen; u64 fb_mask;
-    u32* err_flag; u64 pos_base; u32 ablate;
-};
-
-__device__ __forceinline__ void emit_token_p(const EmitCtx& C, u64* set,
-                                             u64 gh, u32 tl, long tstart,
-                                             long doc_id) {
-    if (C.ablate & 1) {
-        if (gh == 0xdeadbeefdeadbeefULL) C.err_flag[1] = 1;
-        return;
-    }
-    const u64 hh = tokmix_final(gh, tl);
-    const u64 key = hh ? hh : 1ULL;
-    u64 dk = hh ^ splitmix64((u64)doc_id + 0x5bd1e995ULL);
-    if (!dk) dk = 1;
-    int fresh = lds_set_insert(set, dk);
-    if (fresh < 0) {
-        u64 sk = splitmix64(hh ^ ((u64)doc_id * 0x9E3779B97F4A7C15ULL));
-        if (!sk) sk = 1;
-        u64 slot = sk & C.fb_mask;
-        fresh = 0;
-        int probe = 0;
-        while (true) {
-            u64 prev = atomicCAS(&C.fb_seen[slot], 0ULL, sk);
-            if (prev == 0ULL) { fresh = 1; break; }
-            if (prev == sk) break;
-            slot = (slot + 1) & C.fb_mask;
-            if (++probe > FB_PROBE_CAP) {
-                atomicOr(C.err_flag, 1u);
-                break;
-            }
-        }
-    }
-    if (fresh == 1 && !(C.ablate & 2)) {
-        block_count_add(C.cck, C.ccv, key, C.cnt_keys, C.cnt_vals,
-                        C.cnt_mask);
-        u64 slot;
-        if (!(C.ablate & 4)
-            && table_insert_u64(C.dict_keys, C.dict_mask, key, &slot))
-            C.dict_vals[slot] = ((C.pos_base + (u64)tstart) << 8)
-                                | (u64)min(tl, 255u);
-    }
-}
-
-struct GState {
-    long cursor;          // next byte (absolute)
-    long gend;            // group end (exclusive trailing newline)
-    long gd;              // first doc index of the group
-    u32 carry_word, carry_len, carry_lines;
-    u64 carry_g;
-    long carry_start;
-    u64* set;
-};
-
-// One 64-byte window of one group, read from the staged LDS window
-// [aseg, win_hi).  Advances S.cursor.
-__device__ __forceinline__ void step_window_p(
-        const EmitCtx& C, GState& S, const u8* st, const u64* toktab,
-        long aseg, long win_hi, int lane) {
-    const long seg_end = min(S.gend, win_hi);
-    const long p_abs = S.cursor + lane;
-    const bool valid = p_abs < seg_end;
-    const u8 c = valid ? st[p_abs - aseg] : (u8)0;
-    const bool w = valid && is_word(c);
-    const u64 wm = __ballot(w);
-    const u64 nlm = __ballot(valid && c == '\n');
-    if (S.carry_word && !(wm & 1ULL)) {
-        if (lane == 0)
-            emit_token_p(C, S.set, S.carry_g, S.carry_len, S.carry_start,
-                         S.gd + S.carry_lines);
-        S.carry_word = 0;
-        S.carry_len = 0;
-        S.carry_g = 0;
-    }
-    const u64 sm = wm & ~((wm << 1) | (u64)S.carry_word);
-    const u64 below_inc = (lane == 63)
-        ? ~0ULL : ((1ULL << (lane + 1)) - 1ULL);
-    const u64 sm_le = sm & below_inc;
-    int st_off = WAVE;
-    int pos = 0;
-    u64 g = 0;
-    if (w) {
-        st_off = sm_le ? (63 - __clzll(sm_le)) : -(int)S.carry_len;
-        pos = lane - st_off;
-        g = rotl64(toktab[lower_ascii(c)], (u32)pos);
-    }
-    #pragma unroll
-    for (int dsh = 1; dsh <= 4; dsh <<= 1) {
-        const u64 g2 = __shfl_up(g, dsh, WAVE);
-        if (w && lane >= dsh && (lane - dsh) >= st_off) g ^= g2;
-    }
-    if (__ballot(w && pos >= 8)) {
-        #pragma unroll
-        for (int dsh = 8; dsh < WAVE; dsh <<= 1) {
-            const u64 g2 = __shfl_up(g, dsh, WAVE);
-            if (w && lane >= dsh && (lane - dsh) >= st_off) g ^= g2;
-        }
-    }
-    if (w && st_off < 0) g ^= S.carry_g;
-
-    const int last_valid = (int)min(seg_end - S.cursor, (long)WAVE) - 1;
-    const bool group_continues = seg_end < S.gend;
-    bool at_end = w && (lane == 63 ? true : !((wm >> (lane + 1)) & 1));
-    if (lane == last_valid && w
-        && (last_valid == 63 || group_continues))
-        at_end = false;
-
-    if (at_end) {
-        const u64 below = (1ULL << lane) - 1ULL;
-        const long doc_id = S.gd + S.carry_lines + __popcll(nlm & below);
-        emit_token_p(C, S.set, g, (u32)(pos + 1),
-                     (st_off >= 0) ? (S.cursor + st_off) : S.carry_start,
-                     doc_id);
-    }
-
-    const int t_w = (int)((wm >> last_valid) & 1);
-    const int t_end = __shfl((int)at_end, last_valid, WAVE);
-    if (t_w && !t_end) {
-        const int t_pos = __shfl(pos, last_valid, WAVE);
-        const int t_s = __shfl(st_off, last_valid, WAVE);
-        const u64 t_g = __shfl(g, last_valid, WAVE);
-        S.carry_word = 1;
-        S.carry_len = (u32)(t_pos + 1);
-        S.carry_g = t_g;
-        if (t_s >= 0) S.carry_start = S.cursor + t_s;
-    } else {
-        S.carry_word = 0;
-        S.carry_len = 0;
-        S.carry_g = 0;
-    }
-    S.carry_lines += (u32)__popcll(nlm);
-    S.cursor += last_valid + 1;
-}
-
-__global__ void __launch_bounds__(DOC_WAVES * WAVE)
-tfidf_docs_kernel(const u8* __restrict__ text, long n,
-                  const u32* __restrict__ nl_pos, long n_nl, long n_docs,
-                  u64* __restrict__ cnt_keys, u64* __restrict__ cnt_vals,
-                  u64 cnt_mask,
-                  u64* __restrict__ dict_keys, u64* __restrict__ dict_vals,
-                  u64 dict_mask, u64 pos_base,
-                  u64* __restrict__ fb_seen, u64 fb_mask,
-                  u32* __restrict__ err_flag, u32 ablate) {
-    __shared__ __align__(16) u8 stage[DOC_WAVES][STAGE_B + 16];
-    __shared__ u64 dset[DOC_WAVES][2][DOC_SET];
-    __shared__ u64 cck[CCACHE];
-    __shared__ u32 ccv[CCACHE];
-    __shared__ u64 toktab[256];
-    const int wid = threadIdx.x / WAVE;
-    const int lane = threadIdx.x % WAVE;
-    const long gwave = (long)blockIdx.x * DOC_WAVES + wid;
-    const long nwaves = (long)gridDim.x * DOC_WAVES;
-    u8* st = stage[wid];
-
-    for (int i = threadIdx.x; i < CCACHE; i += blockDim.x) {
-        cck[i] = 0;
-        ccv[i] = 0;
-    }
-    for (int i = threadIdx.x; i < 256; i += blockDim.x)
-        toktab[i] = toktab_entry((u8)i);
-    __syncthreads();
-
-    EmitCtx C{cck, ccv, cnt_keys, cnt_vals, cnt_mask, dict_keys,
-              dict_vals, dict_mask, fb_seen, fb_mask, err_flag, pos_base,
-              ablate};
-
-    for (long dbase = gwave * DOC_BLK; dbase < n_docs;
-         dbase += nwaves * DOC_BLK) {
-        const long dlim = min(dbase + (long)DOC_BLK, n_docs);
-        const long di = dbase + lane;
-        const u32 nl_lane = (di < n_nl) ? nl_pos[di] : (u32)n;
-        long win_lo = -1, win_hi = -1, aseg = 0;
-        long gd = dbase;
-        long gs = dbase ? (long)nl_pos[dbase - 1] + 1 : 0;
-
-        while (gd < dlim) {
-            // group A = docs [gd, geA); group B = docs [geA, geB)
-            auto group_of = [&](long g0, long s0, long& ge, long& gend) {
-                const u64 fit = __ballot(
-                    di < dlim && di >= g0
-                    && (long)nl_lane <= s0 + (long)GROUP_BYTES);
-                ge = fit ? (dbase + 63 - __clzll(fit) + 1) : (g0 + 1);
-                gend = (ge - 1 < n_nl)
-                    ? (long)(u32)__shfl((int)nl_lane,
-                                        (int)(ge - 1 - dbase), WAVE)
-                    : n;
-            };
-            long geA, gendA;
-            group_of(gd, gs, geA, gendA);
-            GState A{gs, gendA, gd, 0, 0, 0, 0, gs, dset[wid][0]};
-            bool hasB = geA < dlim;
-            long geB = geA, gendB = gendA;
-            GState B{0, 0, 0, 0, 0, 0, 0, 0, dset[wid][1]};
-            if (hasB) {
-                group_of(geA, gendA + 1, geB, gendB);
-                B = GState{gendA + 1, gendB, geA, 0, 0, 0, 0, gendA + 1,
-                           dset[wid][1]};
-            }
-            for (int sl = lane; sl < DOC_SET; sl += WAVE) {
-                dset[wid][0][sl] = 0;
-                dset[wid][1][sl] = 0;
-            }
-
-            while (A.cursor < A.gend || (hasB && B.cursor < B.gend)) {
-                // window must cover the lowest unfinished cursor
-                const long need = (A.cursor < A.gend)
-                    ? A.cursor : B.cursor;
-                if (need < win_lo || need >= win_hi) {
-                    aseg = need & ~15L;
-                    const int stage_bytes =
-                        (int)min((long)(STAGE_B + 16), n - aseg);
-                    for (int i = lane * 16; i < stage_bytes;
-                         i += WAVE * 16) {
-                        if (aseg + i + 16 <= n) {
-                            *reinterpret_cast<uint4*>(st + i) =
-                                *reinterpret_cast<const uint4*>(
-                                    text + aseg + i);
-                        } else {
-                            for (int j = i; j < stage_bytes; ++j)
-                                st[j] = (aseg + j < n) ? text[aseg + j]
-                                                       : (u8)0;
-                        }
-                    }
-                    asm volatile(
-                        "s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
-                    __builtin_amdgcn_wave_barrier();
-                    win_lo = need;
-                    win_hi = aseg + stage_bytes;
-                }
-                // interleave windows of A and B while both are inside
-                // the staged range
-                const bool aIn = A.cursor < A.gend && A.cursor >= win_lo
-                                 && A.cursor < win_hi;
-                const bool bIn = hasB && B.cursor < B.gend
-                                 && B.cursor >= win_lo
-                                 && B.cursor < win_hi;
-                if (aIn && bIn) {
-                    step_window_p(C, A, st, toktab, aseg, win_hi, lane);
-                    step_window_p(C, B, st, toktab, aseg, win_hi, lane);
-                } else if (aIn) {
-                    step_window_p(C, A, st, toktab, aseg, win_hi, lane);
-                } else if (bIn) {
-                    step_window_p(C, B, st, toktab, aseg, win_hi, lane);
-                }
-                __builtin_amdgcn_wave_barrier();
-            }
-            if (A.carry_word && lane == 0)
-                emit_token_p(C, A.set, A.carry_g, A.carry_len,
-                             A.carry_start, A.gd + A.carry_lines);
-            if (hasB && B.carry_word && lane == 0)
-                emit_token_p(C, B.set, B.carry_g, B.carry_len,
-                             B.carry_start, B.gd + B.carry_lines);
-            gd = hasB ? geB : geA;
-            gs = (hasB ? gendB : gendA) + 1;
-        }
-    }
-
-    __syncthreads();
-    for (int i = threadIdx.x; i < CCACHE; i += blockDim.x)
-        if (cck[i])
-            table_add_u64(cnt_keys, cnt_vals, cnt_mask, cck[i],
-                          (u64)ccv[i]);
-}
-
-#else  // !DOC_PAIR
-
 __global__ void __launch_bounds__(DOC_WAVES * WAVE)
 tfidf_docs_kernel(const u8* __restrict__ text, long n,
                   const u32* __restrict__ nl_pos, long n_nl, long n_docs,
@@ -812,8 +535,6 @@ tfidf_docs_kernel(const u8* __restrict__ text, long n,
             table_add_u64(cnt_keys, cnt_vals, cnt_mask, cck[i],
                           (u64)ccv[i]);
 }
-
-#endif  // DOC_PAIR
 
 // ---------------------------------------------------------------- table ops
 __global__ void table_merge_kernel(const u64* __restrict__ in_keys,
