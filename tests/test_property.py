@@ -69,3 +69,25 @@ def test_join_engines_agree(lk, rk):
             if k == k2:
                 want.append((k, int(lva[i] + rva[j])))
     assert got == sorted(want)
+
+
+@hyp_settings(max_examples=20, deadline=None)
+@given(
+    vals=st.lists(st.integers(min_value=-500, max_value=500),
+                  min_size=1, max_size=200),
+    mod=st.integers(min_value=1, max_value=9),
+    thresh=st.integers(min_value=-500, max_value=500),
+)
+def test_composite_pipelines_agree(vals, mod, thresh):
+    """filter -> map -> count composites: device engine (fallback + device
+    stages) vs host engine."""
+    arr = np.array(vals, dtype=np.int64)
+
+    def build(pm):
+        return pm.filter(lambda v: v > thresh) \
+            .map(lambda v: v % mod).count()
+
+    dev = sorted(build(Dampr.columns(arr)).run().read())
+    host = sorted(build(Dampr.memory(arr.tolist()))
+                  .run(runner=MTRunner, n_maps=2, n_reducers=2).read())
+    assert dev == host

@@ -174,3 +174,38 @@ def test_parity_cached(engines):
     want = sorted(build(RefD).run().read())
     got = sorted(build(OursD).run().read())
     assert got == want and want
+
+
+def test_parity_filter_by_count(engines):
+    RefD, OursD = engines
+    sys.path.insert(0, REF)
+    from dampr.utils import filter_by_count as ref_fbc
+    sys.path.pop(0)
+    from dampr_amd.utils import filter_by_count as our_fbc
+    data = ["a", "b", "a", "c", "a", "b"]
+    want = sorted(ref_fbc(RefD.memory(data), lambda x: x,
+                          lambda c: c >= 2).read())
+    got = sorted(our_fbc(OursD.memory(data), lambda x: x,
+                         lambda c: c >= 2).read())
+    assert got == want and want
+
+
+def test_parity_indexer(engines, tmp_path):
+    RefD, OursD = engines
+    f1 = tmp_path / "r" / "data.txt"
+    f1.parent.mkdir()
+    f1.write_text("apple red\nbanana yellow\ncherry red\n")
+    f2 = tmp_path / "o" / "data.txt"
+    f2.parent.mkdir()
+    f2.write_text("apple red\nbanana yellow\ncherry red\n")
+    sys.path.insert(0, REF)
+    from dampr.utils import Indexer as RefIndexer
+    sys.path.pop(0)
+    from dampr_amd.utils.indexer import Indexer as OurIndexer
+    ri = RefIndexer(str(f1))
+    ri.build(lambda line: line.split())
+    oi = OurIndexer(str(f2))
+    oi.build(lambda line: line.split())
+    want = sorted(ri.union(["red"]).read())
+    got = sorted(oi.union(["red"]).read())
+    assert got == want and want
