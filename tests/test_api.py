@@ -523,3 +523,33 @@ def test_worker_hard_death_detected():
 
     with pytest.raises(WorkerCrash):
         Dampr.memory(list(range(100))).map(die).count().run()
+
+
+def test_json_input(tmp_path):
+    import json as json_mod
+    f = tmp_path / "data.json"
+    rows = [{"a": 1}, {"a": 2}, {"a": 5}]
+    f.write_text("\n".join(json_mod.dumps(r) for r in rows) + "\n")
+    got = sorted(Dampr.json(str(f)).map(lambda d: d["a"]).run().read())
+    assert got == [1, 2, 5]
+
+
+def test_gzip_text_input(tmp_path):
+    import gzip
+    f = tmp_path / "data.txt.gz"
+    with gzip.open(str(f), "wt") as fh:
+        fh.write("x\ny\nx\n")
+    got = sorted(Dampr.text(str(f)).count().run().read())
+    assert got == [("x", 2), ("y", 1)]
+
+
+def test_urls_input_file_scheme(tmp_path):
+    """UrlsInput through file:// URLs — the reference's live-network test
+    (test_dampr.py:369-378) made offline-safe."""
+    from dampr_amd.inputs import UrlsInput
+    f = tmp_path / "remote.txt"
+    f.write_text("hello\nworld\nhello\n")
+    got = sorted(Dampr.read_input(
+        UrlsInput(["file://" + str(f)])).count(lambda x: x.strip())
+        .run().read())
+    assert got == [("hello", 2), ("world", 1)]
