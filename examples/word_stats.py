@@ -51,7 +51,7 @@ def main(fname):
     print("\nLength histogram")
     for length, count in wl.read(20):
         print(" ", length, count)
-    print("\nAverage word length:", awl.read(1)[0])
+    print("\nAverage word length:", awl.read(1)[0][1])
 
 
 if __name__ == "__main__":
