@@ -1419,8 +1419,15 @@ class GpuRunner(RunnerBase):
         """Decode -> run the stage's (opaque) mapper on host -> re-encode.
         The shuffle core stays on device either side of this boundary.
         Input 0 is the primary dataset; the rest ride as supplemental
-        dataset-lists (cross joins — reference: stagerunner.py:66-74)."""
+        dataset-lists (cross joins — reference: stagerunner.py:66-74).
+        Concat stages treat every input as a primary."""
         from ..dataset import MemoryDataset
+        if stage.options.get("concat"):
+            out_records = []
+            for store in ins:
+                out_records.extend(stage.mapper.map(
+                    MemoryDataset(self._decode_store(store))))
+            return self._encode_or_host(out_records)
         primary = MemoryDataset(self._decode_store(ins[0]))
         supplemental = [[MemoryDataset(self._decode_store(s))]
                         for s in ins[1:]]

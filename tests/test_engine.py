@@ -609,3 +609,15 @@ def test_sample_device_engine():
         .run(runner=GpuRunner).read()
     assert len(got) == 1
     assert 300 < got[0][1] < 700
+
+
+def test_concat_device_engine():
+    from dampr_amd.gpu.engine import GpuRunner
+    a = Dampr.columns(np.array([1, 2], dtype=np.int64))
+    b = Dampr.columns(np.array([3, 4], dtype=np.int64))
+    got = sorted(a.concat(b).run(runner=GpuRunner).read())
+    assert got == [1, 2, 3, 4]
+    # object records too
+    c = Dampr.memory(["x"]).concat(Dampr.memory(["y", "x"]))
+    assert sorted(c.count().run(runner=GpuRunner).read()) == \
+        [("x", 2), ("y", 1)]
