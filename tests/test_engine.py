@@ -621,3 +621,13 @@ def test_concat_device_engine():
     c = Dampr.memory(["x"]).concat(Dampr.memory(["y", "x"]))
     assert sorted(c.count().run(runner=GpuRunner).read()) == \
         [("x", 2), ("y", 1)]
+
+
+def test_pjoin_default_run_device_engine():
+    """PJoin.run() (no reduce) materializes (left_list, right_list)
+    values through the device engine's HostStore path."""
+    from dampr_amd.gpu.engine import GpuRunner
+    l = Dampr.columns(np.array([10, 20]), keys=np.array([1, 2]))
+    r = Dampr.columns(np.array([5]), keys=np.array([2]))
+    got = sorted(l.join(r).run(runner=GpuRunner).read())
+    assert got == [(2, ([20], [5]))]
