@@ -516,23 +516,25 @@ class ARReduce(object):
         return PReduce(pm.source, pm.pmer).reduce(_reduce,
                                                   options=red_opts)
 
-    def reduce(self, binop, reduce_buffer=1000, **options):
+    def reduce(self, binop, reduce_buffer=None, **options):
         """Reduce each group with an associative binop.  ``reduce_buffer``
         caps the map-side combine dictionary (keys held in memory before a
-        spill) — unlike the reference, it is honored.
-        Recognized binops (operator.add, min, max — dampr_amd.funcs) lower
-        to the device segmented-reduce kernel on the columnar engine."""
+        spill) — unlike the reference, it is honored when set (SURVEY.md
+        §2.5); the default defers to the RSS watermark like the
+        reference's actual behavior.  Recognized binops (operator.add,
+        min, max — dampr_amd.funcs) lower to the device segmented-reduce
+        kernel on the columnar engine."""
         name = funcs.binop_name(binop)
         return self._run(binop, (name,) if name else None, reduce_buffer,
                          options)
 
     def first(self, **options):
         """First value seen per key."""
-        return self._run(lambda x, _y: x, ("first",), 1000, options)
+        return self._run(lambda x, _y: x, ("first",), None, options)
 
     def sum(self, **options):
         """Sum of values per key."""
-        return self._run(operator.add, ("sum",), 1000, options)
+        return self._run(operator.add, ("sum",), None, options)
 
 
 class PReduce(PBase):
