@@ -216,7 +216,12 @@ class _CombineMapWorker(StageWorker):
             stream = EmptyDataset()
         shuffler = DefaultShuffler(self.n_partitions, Splitter(),
                                    memory=self.memory)
-        return shuffler.shuffle(self.fs, [stream])
+        out = shuffler.shuffle(self.fs, [stream])
+        # the shuffle consumed the pre-shuffle combine runs fully:
+        # remove their spill files (they would otherwise leak in /tmp)
+        for r in runs:
+            r.delete()
+        return out
 
 
 class _ReduceWorker(StageWorker):

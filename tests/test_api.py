@@ -553,3 +553,16 @@ def test_urls_input_file_scheme(tmp_path):
         UrlsInput(["file://" + str(f)])).count(lambda x: x.strip())
         .run().read())
     assert got == [("hello", 2), ("world", 1)]
+
+
+def test_no_spill_file_leak(tmp_path):
+    """After run + delete, no spill files remain (the reference leaks
+    its pre-shuffle combine runs)."""
+    res = Dampr.memory(list(range(1000))).count(lambda x: x % 7) \
+        .run("leakcheck", working_dir=str(tmp_path))
+    assert len(res.read()) == 7
+    res.delete()
+    leftover = []
+    for root, _dirs, files in os.walk(str(tmp_path)):
+        leftover.extend(os.path.join(root, f) for f in files)
+    assert leftover == []
