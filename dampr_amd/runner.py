@@ -352,6 +352,15 @@ class RunnerBase(object):
                     for ds_list in dm.values():
                         for ds in ds_list:
                             ds.delete()
+            # prune now-empty stage/worker directories (kept outputs and
+            # their parents survive; removal is best-effort bottom-up)
+            root = self.file_system.path
+            for dirpath, _dirs, _files in sorted(
+                    os.walk(root), key=lambda w: -len(w[0])):
+                try:
+                    os.rmdir(dirpath)
+                except OSError:
+                    pass
         log.info("Finished")
         return rets
 
