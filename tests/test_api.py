@@ -566,3 +566,15 @@ def test_no_spill_file_leak(tmp_path):
     for root, _dirs, files in os.walk(str(tmp_path)):
         leftover.extend(os.path.join(root, f) for f in files)
     assert leftover == []
+
+
+def test_reducer_exception_fails_fast():
+    from dampr_amd.executor import WorkerCrash
+
+    def bad_reduce(_k, vs):
+        raise RuntimeError("reduce-boom")
+
+    with pytest.raises(WorkerCrash) as ei:
+        Dampr.memory(list(range(50))).group_by(lambda x: x % 3) \
+            .reduce(bad_reduce).run()
+    assert "reduce-boom" in str(ei.value)
