@@ -61,7 +61,7 @@ def _run_rank(rank, world, port, q):
 def test_exchange_gloo_world2():
     ctx = multiprocessing.get_context("spawn")
     q = ctx.Queue()
-    port = 29511
+    port = 29000 + (os.getpid() + 40) % 900
     world = 2
     procs = [ctx.Process(target=_run_rank, args=(r, world, port, q))
              for r in range(world)]

@@ -193,8 +193,9 @@ def _engine_rank(rank, world, port, q):
         q.put((rank, traceback.format_exc()))
 
 
-@pytest.mark.parametrize("world,port", [(2, 29517), (3, 29527)])
-def test_engine_gloo_multi_world(world, port):
+@pytest.mark.parametrize("world,port_off", [(2, 0), (3, 10)])
+def test_engine_gloo_multi_world(world, port_off):
+    port = 29000 + (os.getpid() + port_off) % 900
     ctx = multiprocessing.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_engine_rank, args=(r, world, port, q))
@@ -403,8 +404,9 @@ def _engine_rank2(rank, world, port, q):
         q.put((rank, traceback.format_exc()))
 
 
-@pytest.mark.parametrize("world,port", [(2, 29519), (3, 29529)])
-def test_engine_gloo_join_topk(world, port):
+@pytest.mark.parametrize("world,port_off", [(2, 20), (3, 30)])
+def test_engine_gloo_join_topk(world, port_off):
+    port = 29000 + (os.getpid() + port_off) % 900
     ctx = multiprocessing.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_engine_rank2, args=(r, world, port, q))
