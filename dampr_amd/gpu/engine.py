@@ -710,16 +710,24 @@ class GpuRunner(RunnerBase):
                 torch.zeros(0, dtype=torch.int64))
         ks = [k for k, _ in records]
         vs = [v for _, v in records]
+        def _num(x):
+            # bools stay Python objects: columnar i64 would decode True
+            # as 1 (repr-visible divergence from the host engine)
+            return isinstance(x, (int, float)) and not isinstance(x, bool)
+
+        def _int(x):
+            return isinstance(x, int) and not isinstance(x, bool)
+
         # host keyed-reducer convention: value = (key, scalar)
         keyed = all(
             isinstance(v, tuple) and len(v) == 2 and v[0] == k
-            and isinstance(v[1], (int, float, bool))
+            and _num(v[1])
             for (k, _), v in zip(records, vs))
         if keyed:
             vs = [v[1] for v in vs]
-        if all(isinstance(v, (int, bool)) for v in vs):
+        if all(_int(v) for v in vs):
             vt = torch.from_numpy(np.asarray(vs, dtype=np.int64))
-        elif all(isinstance(v, (int, float, bool)) for v in vs):
+        elif all(_num(v) for v in vs):
             vt = torch.from_numpy(np.asarray(vs, dtype=np.float64))
         else:
             raise TypeError(
@@ -727,9 +735,9 @@ class GpuRunner(RunnerBase):
                 "engine for object records")
         fkeys = False
         str_table = None
-        if all(isinstance(k, (int, bool)) for k in ks):
+        if all(_int(k) for k in ks):
             kt = torch.from_numpy(np.asarray(ks, dtype=np.int64))
-        elif all(isinstance(k, (int, float, bool)) for k in ks):
+        elif all(_num(k) for k in ks):
             kt = _encode_f64_sortable(
                 torch.from_numpy(np.asarray(ks, dtype=np.float64)))
             fkeys = True

@@ -65,6 +65,18 @@ PIPELINES = {
         .reduce(lambda l, r: sum(l) * sum(r))),
     "cross_right": lambda D: D.memory([1, 2, 3, 4])
         .cross_right(D.memory([10]), lambda v, t: v * t),
+    "map_values": lambda D: D.memory([("a", ("x", 1)), ("b", ("y", 2))])
+        .map_values(lambda v: v * 10),
+    "map_keys": lambda D: D.memory([("a", ("x", 1)), ("b", ("y", 2))])
+        .map_keys(lambda k: k.upper()),
+    "prefix": lambda D: D.memory([3, 1, 2]).prefix(lambda v: v % 2),
+    "suffix": lambda D: D.memory([3, 1, 2]).suffix(lambda v: v + 1),
+    "inspect": lambda D: D.memory([1, 2]).inspect("t").count(),
+    "cross_left": lambda D: D.memory([1, 2])
+        .cross_left(D.memory([5, 7]), lambda a, b: (a, b)),
+    "cross_set": lambda D: D.memory([1, 2, 3])
+        .cross_set(D.memory([2, 3]), lambda v, right: v in right,
+                   agg=set),
 }
 
 
