@@ -77,6 +77,18 @@ PIPELINES = {
     "cross_set": lambda D: D.memory([1, 2, 3])
         .cross_set(D.memory([2, 3]), lambda v, right: v in right,
                    agg=set),
+    # each key's 3 values are identical, so "first value seen" is
+    # deterministic (arrival order within a key is scheduling-dependent
+    # in both engines)
+    "first": lambda D: D.memory(NUMS).a_group_by(lambda v: v).first(),
+    "left_reduce": lambda D: (
+        D.memory(ITEMS).group_by(lambda kv: kv[0], lambda kv: kv[1])
+        .join(D.memory([("apple", 1), ("mango", 2)])
+              .group_by(lambda kv: kv[0], lambda kv: kv[1]))
+        .left_reduce(lambda l, r: (sorted(l), sorted(r)))),
+    "checkpoint": lambda D: D.memory(NUMS)
+        .map(lambda v: v + 1).checkpoint().count(lambda v: v % 5),
+    "sample_all": lambda D: D.memory(NUMS).sample(1.0).len(),
 }
 
 
@@ -210,6 +222,47 @@ def test_parity_filter_by_count(engines):
     got = sorted(our_fbc(OursD.memory(data), lambda x: x,
                          lambda c: c >= 2).read())
     assert got == want and want
+
+
+def test_parity_custom_mapper_reducer(engines):
+    """custom_mapper/custom_reducer take each library's own Map/Reduce
+    operator objects — built per-engine, compared on output."""
+    RefD, OursD = engines
+    sys.path.insert(0, REF)
+    from dampr.base import Map as RefMap, Reduce as RefReduce
+    sys.path.pop(0)
+    from dampr_amd.base import Map as OurMap, Reduce as OurReduce
+
+    def build(D, Map, Reduce):
+        return D.memory(list(range(20))) \
+            .custom_mapper(Map(lambda _k, v: [(v % 3, v)])) \
+            .custom_reducer(Reduce(lambda k, vs: sum(vs)))
+
+    want = sorted(map(repr, build(RefD, RefMap, RefReduce).run().read()))
+    got = sorted(map(repr, build(OursD, OurMap, OurReduce).run().read()))
+    assert got == want and want
+
+
+def test_parity_sink_tsv_json(engines, tmp_path):
+    RefD, OursD = engines
+    rows = [("hank", 755, 2.5), ("babe", 714, 3.5)]
+    objs = [{"name": "hank", "hr": 755}, {"name": "babe", "hr": 714}]
+
+    def lines(d):
+        out = []
+        for fn in sorted(os.listdir(d)):
+            with open(os.path.join(d, fn)) as fh:
+                out.extend(ln.rstrip("\n") for ln in fh if ln.strip())
+        return sorted(out)
+
+    RefD.memory(rows).sink_tsv(str(tmp_path / "rt")).run()
+    OursD.memory(rows).sink_tsv(str(tmp_path / "ot")).run()
+    assert lines(tmp_path / "ot") == lines(tmp_path / "rt") \
+        and lines(tmp_path / "ot")
+    RefD.memory(objs).sink_json(str(tmp_path / "rj")).run()
+    OursD.memory(objs).sink_json(str(tmp_path / "oj")).run()
+    assert sorted(lines(tmp_path / "oj")) == sorted(lines(tmp_path / "rj")) \
+        and lines(tmp_path / "oj")
 
 
 def test_parity_indexer(engines, tmp_path):
