@@ -15,6 +15,7 @@ Additions over the reference:
 """
 import itertools
 import json as json_mod
+import os
 import logging
 import operator
 import random
@@ -36,8 +37,10 @@ log = logging.getLogger("dampr_amd")
 class ValueEmitter(object):
     """Reads values out of a finished computation."""
 
-    def __init__(self, dataset):
+    def __init__(self, dataset, run_root=None):
         self.dataset = dataset
+        # run-scratch root, for pruning after a late delete()
+        self.run_root = run_root
 
     def stream(self):
         for _k, v in self.dataset.read():
@@ -53,6 +56,15 @@ class ValueEmitter(object):
 
     def delete(self):
         self.dataset.delete()
+        # the run's tree is only dirs + this output now: prune bottom-up
+        # (best-effort; shared parents with live files survive the rmdir)
+        if self.run_root and os.path.isdir(self.run_root):
+            for dirpath, _dirs, _files in sorted(
+                    os.walk(self.run_root), key=lambda w: -len(w[0])):
+                try:
+                    os.rmdir(dirpath)
+                except OSError:
+                    pass
 
 
 class PBase(object):
@@ -72,9 +84,11 @@ class PBase(object):
             name = "dampr/{}".format(random.random())
         runner_cls = kwargs.pop("runner", None) or \
             _pick_runner(self.pmer.graph, self.pmer.runner)
-        ds = runner_cls(name, self.pmer.graph, **kwargs) \
-            .run([self.source])
-        return ValueEmitter(ds[0])
+        runner = runner_cls(name, self.pmer.graph, **kwargs)
+        ds = runner.run([self.source])
+        return ValueEmitter(
+            ds[0], run_root=getattr(
+                getattr(runner, "file_system", None), "path", None))
 
     def read(self, k=None, **kwargs):
         """run() followed by read()."""
@@ -717,8 +731,10 @@ class Dampr(object):
         name = kwargs.pop("name", "dampr/{}".format(random.random()))
         runner_cls = kwargs.pop("runner", None) or \
             _pick_runner(graph, last.pmer.runner)
-        ds = runner_cls(name, graph, **kwargs).run(sources)
-        return [ValueEmitter(d) for d in ds]
+        runner = runner_cls(name, graph, **kwargs)
+        ds = runner.run(sources)
+        root = getattr(getattr(runner, "file_system", None), "path", None)
+        return [ValueEmitter(d, run_root=root) for d in ds]
 
     def _add_mapper(self, *args, **kwargs):
         output, ng = self.graph.add_mapper(*args, **kwargs)

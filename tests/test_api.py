@@ -578,3 +578,27 @@ def test_reducer_exception_fails_fast():
         Dampr.memory(list(range(50))).group_by(lambda x: x % 3) \
             .reduce(bad_reduce).run()
     assert "reduce-boom" in str(ei.value)
+
+
+def test_delete_prunes_empty_run_dirs(tmp_path):
+    """ValueEmitter.delete removes the run's now-empty directory tree,
+    not just the data files (ROADMAP 8: host-tier niceties)."""
+    res = Dampr.memory(list(range(200))).count(lambda x: x % 5) \
+        .run("prunecheck", working_dir=str(tmp_path))
+    assert len(res.read()) == 5
+    res.delete()
+    assert not os.path.exists(os.path.join(str(tmp_path), "prunecheck"))
+
+
+def test_multi_output_delete_keeps_sibling(tmp_path):
+    """Deleting one of two retained outputs must not disturb the other."""
+    words = Dampr.memory(["a b", "b c b"]).flat_map(lambda s: s.split())
+    counts = words.count()
+    total = counts.fold_by(lambda _wc: 1, lambda x, y: x + y,
+                           value=lambda wc: wc[1])
+    rc, rt = Dampr.run(counts, total, name="sibcheck",
+                       working_dir=str(tmp_path))
+    rc.delete()
+    assert sorted(rt.read()) == [(1, 5)]
+    rt.delete()
+    assert not os.path.exists(os.path.join(str(tmp_path), "sibcheck"))
