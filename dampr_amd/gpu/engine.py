@@ -31,6 +31,7 @@ pools):
   is owned by rank ``p % world`` (parallel/shuffle.py).
 """
 import logging
+import os
 
 import torch
 
@@ -1314,6 +1315,8 @@ class GpuRunner(RunnerBase):
             keyed=True,
             fkeys=getattr(left, "fkeys", False)
             or getattr(right, "fkeys", False))
+        cap = int(os.environ.get("DAMPR_JOIN_PROBE_ROWS",
+                                 settings.gpu_join_probe_rows))
         for p in self._parts([left, right]):
             # sorted probe keys touch table slots near-sequentially
             # (slot = key & mask), turning random HBM probes into
@@ -1328,21 +1331,47 @@ class GpuRunner(RunnerBase):
             if rk is None:
                 rk = torch.zeros(0, dtype=torch.int64, device=self.device)
                 rv = torch.zeros(0, dtype=torch.int64, device=self.device)
-            li, ri = self.ops.hash_join(lk, rk, how)
-            keys = torch.where(li >= 0, lk[torch.clamp(li, min=0)],
-                               rk[torch.clamp(ri, min=0)])
-            lvm = torch.where(li >= 0, lv[torch.clamp(li, min=0)],
-                              torch.zeros_like(li))
-            rvm = torch.where(ri >= 0, rv[torch.clamp(ri, min=0)],
-                              torch.zeros_like(ri))
-            valid_l = li >= 0
-            valid_r = ri >= 0
-            merged = self._apply_pair_op(pair_op, lvm, rvm, valid_l,
-                                         valid_r)
-            sk, sp = self._sort(keys, fkeys=out.fkeys)
-            run = DeviceRun(sk, merged[sp.to(torch.int64)], sorted=True)
-            out.setdefault(p, []).append(run)
-            self.pool.admit(run)
+            # skewed-join guard (ROADMAP 7): the table is built on the
+            # RIGHT side, probes stream the LEFT.  Inner joins are
+            # symmetric, so if the build side is the oversized one, swap
+            # so the big side is probed in batches instead of blowing
+            # out the table build.
+            swap = (how == "inner" and rk.numel() > cap
+                    and rk.numel() > lk.numel())
+            if swap:
+                lk, lv, rk, rv = rk, rv, lk, lv
+            # inner/left matches of a probe row are independent of every
+            # other probe row, so probing distributes over contiguous
+            # batches; full-outer tracks unmatched RIGHT rows across the
+            # whole probe and must run in one piece.
+            n = lk.numel()
+            step = n if (how == "outer" or n <= cap) else cap
+            for a in range(0, max(n, 1), max(step, 1)):
+                b = min(n, a + max(step, 1))
+                li, ri = self.ops.hash_join(lk[a:b], rk, how)
+                li = torch.where(li >= 0, li + a, li)
+                if swap:
+                    li, ri = ri, li
+                    klk, krk, klv, krv = rk, lk, rv, lv
+                else:
+                    klk, krk, klv, krv = lk, rk, lv, rv
+                keys = torch.where(li >= 0, klk[torch.clamp(li, min=0)],
+                                   krk[torch.clamp(ri, min=0)])
+                lvm = torch.where(li >= 0, klv[torch.clamp(li, min=0)],
+                                  torch.zeros_like(li))
+                rvm = torch.where(ri >= 0, krv[torch.clamp(ri, min=0)],
+                                  torch.zeros_like(ri))
+                valid_l = li >= 0
+                valid_r = ri >= 0
+                merged = self._apply_pair_op(pair_op, lvm, rvm, valid_l,
+                                             valid_r)
+                if keys.numel() == 0 and n > 0:
+                    continue
+                sk, sp = self._sort(keys, fkeys=out.fkeys)
+                run = DeviceRun(sk, merged[sp.to(torch.int64)],
+                                sorted=True)
+                out.setdefault(p, []).append(run)
+                self.pool.admit(run)
         return out
 
     @staticmethod

@@ -57,6 +57,11 @@ gpu_partitions_per_rank = 8
 # one partition of a 288 GB-per-GPU job fits comfortably in the pool.
 gpu_partitions = 64
 
+# Skewed-join guard: probe-side rows per hash-join batch.  A join
+# partition larger than this is probed in chunks against the built
+# table (inner/left; ROADMAP 7) so one hot key cannot blow out HBM.
+gpu_join_probe_rows = 1 << 26
+
 # Columnar engine HBM pool capacity (MB) before runs spill to pinned host.
 # Default stays small enough for CPU test runs; bench/production set it to
 # ~0.9 * free HBM.

@@ -644,3 +644,47 @@ def test_text_input_device_engine(tmp_path):
     got = dict(Dampr.text(str(f)).map(int).count()
                .run(runner=GpuRunner).read())
     assert got == {k: 20 for k in range(5)}
+
+
+def test_device_join_probe_batching(monkeypatch):
+    """Skewed-join guard (ROADMAP 7): an oversized probe side is joined
+    in batches against the once-built table; results must be identical
+    to the unbatched join."""
+    monkeypatch.setenv("DAMPR_JOIN_PROBE_ROWS", "1000")
+    rng = np.random.RandomState(7)
+    # hot key 5 dominates the left side -> many probe batches
+    lk = np.concatenate([np.full(5000, 5), rng.randint(0, 50, 500)]) \
+        .astype(np.int64)
+    lv = rng.randint(1, 100, lk.size).astype(np.int64)
+    rk = np.array([5, 5, 7, 9], dtype=np.int64)
+    rv = np.array([2, 3, 4, 5], dtype=np.int64)
+    out = Dampr.columns(lv, keys=lk).join(Dampr.columns(rv, keys=rk)) \
+        .reduce(funcs.pair_product, many=True).run()
+    got = sorted(out.read())
+    rmap = {}
+    for j, k in enumerate(rk):
+        rmap.setdefault(int(k), []).append(int(rv[j]))
+    want = sorted((int(k), int(v) * w)
+                  for k, v in zip(lk, lv) for w in rmap.get(int(k), []))
+    assert got == want and len(want) > 10000
+
+
+def test_device_join_build_side_swap(monkeypatch):
+    """Inner join with the oversized side on the RIGHT (the build side):
+    the engine swaps sides so the big side is probed in batches."""
+    monkeypatch.setenv("DAMPR_JOIN_PROBE_ROWS", "1000")
+    rng = np.random.RandomState(11)
+    lk = np.array([3, 4, 4, 8], dtype=np.int64)
+    lv = np.array([7, 1, 2, 9], dtype=np.int64)
+    rk = np.concatenate([np.full(4000, 4), rng.randint(0, 20, 400)]) \
+        .astype(np.int64)
+    rv = rng.randint(1, 50, rk.size).astype(np.int64)
+    out = Dampr.columns(lv, keys=lk).join(Dampr.columns(rv, keys=rk)) \
+        .reduce(funcs.pair_sum, many=True).run()
+    got = sorted(out.read())
+    rmap = {}
+    for j, k in enumerate(rk):
+        rmap.setdefault(int(k), []).append(int(rv[j]))
+    want = sorted((int(k), int(v) + w)
+                  for k, v in zip(lk, lv) for w in rmap.get(int(k), []))
+    assert got == want and len(want) > 4000
