@@ -648,10 +648,23 @@ class GpuRunner(RunnerBase):
         host Datasets/Chunkers decode through the numeric encoder."""
         if isinstance(inp, TextSource):
             if self.world > 1:
-                raise NotImplementedError(
-                    "device_text on multi-rank runs: use the explicit "
-                    "tfidf pipeline (bench.py) which exchanges partials "
-                    "over RCCL")
+                # per-rank newline-aligned slice of the corpus; every
+                # rank computes the identical boundaries (same input),
+                # so slices partition the text exactly
+                t = inp.text
+                n = t.shape[0]
+
+                def _align(b):
+                    # first byte AFTER the next newline at/past b
+                    import numpy as np
+                    if b <= 0 or b >= n:
+                        return min(max(b, 0), n)
+                    nl = np.flatnonzero(t[b - 1:] == ord("\n"))
+                    return (b - 1 + int(nl[0]) + 1) if nl.size else n
+
+                lo = _align(n * self.rank // self.world)
+                hi = _align(n * (self.rank + 1) // self.world)
+                return TextSource(t[lo:hi].copy())
             return inp
         if isinstance(inp, ColumnSource):
             keys = inp.keys.to(self.device)
@@ -1062,6 +1075,21 @@ class GpuRunner(RunnerBase):
             for s0, e0 in zip(bounds, bounds[1:]):
                 eng.count_chunk(text[s0:e0].contiguous(), pos_base=s0)
             keys, df = eng.extract()
+            if self.world > 1:
+                # ONE exchange of (key, df, token-bytes) partials routed
+                # by key % world; each rank rebuilds its owned shard
+                # (same pattern as bench.py's explicit pipeline)
+                from ..parallel.shuffle import (all_reduce_scalar,
+                                                exchange_keyed_payload)
+                blob, lens = eng.token_strings_dev(keys, text)
+                rk, rdf, rblob, rlens = exchange_keyed_payload(
+                    keys, df, blob, lens)
+                eng.merge_exchanged(rk, rdf, rblob, rlens)
+                keys, df = eng.extract()
+                eng.n_docs = all_reduce_scalar(eng.n_docs,
+                                               device=self.device)
+                # received token strings live in the exchanged blob
+                text = rblob
             return TokenStore(eng, keys, df, text)
         if kind == "topk_local":
             # per-partition top-k candidates by value (K11); all

@@ -688,3 +688,44 @@ def test_device_join_build_side_swap(monkeypatch):
     want = sorted((int(k), int(v) + w)
                   for k, v in zip(lk, lv) for w in rmap.get(int(k), []))
     assert got == want and len(want) > 4000
+
+
+def _engine_rank_text(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        text = b"apple pear\npear fig apple\nfig fig kiwi\napple kiwi\n" * 40
+        got = sorted(Dampr.device_text(np.frombuffer(text, dtype=np.uint8))
+                     .flat_map(funcs.tokenize_set).count().run().read())
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got)
+        merged = sorted(p for lst in gathered for p in lst)
+        # per-line token SET counts: each word's df over 160 lines
+        want = sorted([("apple", 120), ("fig", 80),
+                       ("kiwi", 80), ("pear", 80)])
+        assert merged == want, merged
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [2])
+def test_engine_gloo_device_text(world):
+    """Multi-rank device_text (ROADMAP 3): each rank counts its
+    newline-aligned slice; partial dfs merge across ranks."""
+    port = 29000 + (os.getpid() + 20) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_text, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
