@@ -192,8 +192,9 @@ class PartStore(dict):
         self.fkeys = fkeys
         # dictionary-encoded string keys: key column holds ranks into
         # this sorted tuple (rank order == lexicographic order, so sorted
-        # device output decodes to host-ordered strings).  Stores with
-        # tables only combine with themselves; cross-store combinations
+        # device output decodes to host-ordered strings).  Cross-store
+        # combinations remap ids into the union table on device
+        # (_unify_str_stores); only dictionary + dictionary-less mixes
         # fall back to host records (_decode_store).
         self.str_table = str_table
         # False = runs in pseudo-partition 0, not yet routed by key hash
@@ -772,10 +773,88 @@ class GpuRunner(RunnerBase):
         return store
 
     def _encode_or_host(self, records):
+        if self.world > 1:
+            return self._encode_records_world(records)
         try:
             return self._encode_records(records)
         except TypeError:
             return HostStore(records)
+
+    def _encode_records_world(self, records):
+        """world>1 encode: all ranks agree on ONE layout via a single
+        all_gather of local metadata (key/value kinds, keyed flag,
+        string table) before building any tensor.  A per-rank choice
+        would (a) desync the exchange collectives when branches differ
+        and (b) assign per-rank-incompatible string dictionary ids —
+        the same id on two ranks decoding to different strings after
+        the exchange."""
+        import numpy as np
+        import torch.distributed as dist
+
+        def _num(x):
+            return isinstance(x, (int, float)) and not isinstance(x, bool)
+
+        def _int(x):
+            return isinstance(x, int) and not isinstance(x, bool)
+
+        ks = [k for k, _ in records]
+        vs = [v for _, v in records]
+        keyed = None
+        if records:
+            keyed = all(
+                isinstance(v, tuple) and len(v) == 2 and v[0] == k
+                and _num(v[1]) for k, v in zip(ks, vs))
+        pv = [v[1] for v in vs] if keyed else vs
+
+        def kind_of(xs, allow_str):
+            if not xs:
+                return "empty", None
+            if all(_int(x) for x in xs):
+                return "int", None
+            if all(_num(x) for x in xs):
+                return "float", None
+            if allow_str and all(isinstance(x, str) for x in xs):
+                return "str", tuple(sorted(set(xs)))
+            return "obj", None
+
+        kk, table = kind_of(ks, True)
+        vk, _ = kind_of(pv, False)
+        gathered = [None] * self.world
+        dist.all_gather_object(gathered, (kk, vk, keyed, table))
+        kks = {m[0] for m in gathered} - {"empty"}
+        vks = {m[1] for m in gathered} - {"empty"}
+        keyeds = {m[2] for m in gathered} - {None}
+        if ("obj" in kks or "obj" in vks or len(keyeds) > 1
+                or ("str" in kks and len(kks) > 1)):
+            # no common columnar layout: every rank falls back to host
+            # records (symmetric — no further collectives here; host
+            # reduces exchange via _host_exchange)
+            return HostStore(records)
+        g_keyed = keyeds.pop() if keyeds else False
+        vals = pv if g_keyed else vs
+        v_np = np.int64 if vks in ({"int"}, set()) else np.float64
+        vt = torch.from_numpy(np.asarray(vals, dtype=v_np))
+        fkeys = False
+        str_table = None
+        if kks == {"str"}:
+            str_table = tuple(sorted(
+                set().union(*[set(m[3]) for m in gathered if m[3]])))
+            ranks = {t: i for i, t in enumerate(str_table)}
+            kt = torch.from_numpy(np.fromiter(
+                (ranks[k] for k in ks), dtype=np.int64, count=len(ks)))
+        elif kks in ({"int"}, set()):
+            kt = torch.from_numpy(np.asarray(ks, dtype=np.int64))
+        else:
+            # float keys, or int+float mixed (ints beyond 2^53 would
+            # lose precision here; such pipelines belong on the host
+            # engine)
+            kt = _encode_f64_sortable(
+                torch.from_numpy(np.asarray(ks, dtype=np.float64)))
+            fkeys = True
+        store = self._partition(kt.to(self.device), vt.to(self.device),
+                                keyed=bool(g_keyed), fkeys=fkeys)
+        store.str_table = str_table
+        return store
 
     def _collect(self, store):
         """Partition store -> one key-sorted ColumnDataset (the engine's
@@ -1154,6 +1233,42 @@ class GpuRunner(RunnerBase):
             return keys
         raise ValueError("unknown column func {!r}".format(name))
 
+    def _unify_str_stores(self, stores):
+        """Cross-encode string keys (ROADMAP 2's remap): rank ids from
+        different dictionaries are incompatible, so remap every run's
+        key column through a device lut into the union table.  The lut
+        is monotone (sorted table -> sorted union), so sorted runs stay
+        sorted; hash ROUTING changes, so remapped stores come back
+        unpartitioned and re-route in _ensure_partitioned.  Returns the
+        unified store list, or None when any input has no dictionary
+        (numeric/host/token stores: caller falls back to host records)."""
+        tables = [getattr(s, "str_table", None) for s in stores]
+        if not all(isinstance(s, PartStore) and t is not None
+                   for s, t in zip(stores, tables)):
+            return None
+        merged = tuple(sorted(set().union(*map(set, tables))))
+        index = {t: i for i, t in enumerate(merged)}
+        out = []
+        for s, t in zip(stores, tables):
+            if t == merged:
+                out.append(s)
+                continue
+            lut = torch.tensor([index[x] for x in t], dtype=torch.int64,
+                               device=self.device)
+            ns = PartStore(keyed=getattr(s, "keyed", False),
+                           partitioned=False, str_table=merged)
+            ns[0] = []
+            for p in sorted(s):
+                for run in s[p]:
+                    self.pool.touch(run, self.device)
+                    nr = DeviceRun(lut[run.keys], run.vals.clone(),
+                                   sorted=run.sorted)
+                    self.pool.release(run)
+                    ns[0].append(nr)
+                    self.pool.admit(nr)
+            out.append(ns)
+        return out
+
     def _merge_stores(self, stores):
         if len(stores) == 1:
             return stores[0]
@@ -1165,21 +1280,29 @@ class GpuRunner(RunnerBase):
         if len(stores) > 1 and any(
                 getattr(s, "str_table", None) is not None
                 for s in stores):
-            # rank ids from different encodes are incompatible: combine
-            # as host records (correct, slower)
-            out = HostStore()
-            for s in stores:
-                out.extend(self._decode_store(s))
-            return out
+            uni = self._unify_str_stores(stores)
+            if uni is None:
+                # dictionary + dictionary-less mix: combine as host
+                # records (correct, slower)
+                out = HostStore()
+                for s in stores:
+                    out.extend(self._decode_store(s))
+                return out
+            stores = uni
         flags = [getattr(s, "partitioned", True) for s in stores]
         if not all(flags) and any(flags):
             # mixing hashed and unrouted partition-0 runs would corrupt
             # co-location: route everything first
             stores = [self._ensure_partitioned(s) for s in stores]
+            flags = [getattr(s, "partitioned", True) for s in stores]
         out = PartStore(
             keyed=any(getattr(s, "keyed", False) for s in stores),
             fkeys=any(getattr(s, "fkeys", False) for s in stores),
-            partitioned=all(flags))
+            partitioned=all(flags),
+            str_table=next((getattr(s, "str_table", None)
+                            for s in stores
+                            if getattr(s, "str_table", None) is not None),
+                           None))
         for s in stores:
             for p, runs in s.items():
                 out.setdefault(p, []).extend(runs)
@@ -1189,9 +1312,14 @@ class GpuRunner(RunnerBase):
 
     def run_reduce(self, stage, ins):
         spec = stage.options.get("device_reduce")
+        uni_table = None
         if len(ins) > 1 and any(
                 getattr(s, "str_table", None) is not None for s in ins):
-            return self._host_reduce(stage, ins)
+            uni = self._unify_str_stores(ins)
+            if uni is None:
+                return self._host_reduce(stage, ins)
+            ins = uni
+            uni_table = ins[0].str_table
         ins = [self._ensure_partitioned(s) for s in ins]
         if len(ins) == 1 and isinstance(ins[0], TokenStore) \
                 and spec == ("sum",):
@@ -1204,7 +1332,7 @@ class GpuRunner(RunnerBase):
         in_fkeys = any(getattr(s, "fkeys", False) for s in ins)
         out = PartStore(keyed=True, fkeys=in_fkeys,
                         str_table=getattr(ins[0], "str_table", None)
-                        if len(ins) == 1 else None)
+                        if len(ins) == 1 else uni_table)
         if kind in ("sum", "min", "max"):
             parts = self._parts(ins)
             for i, p in enumerate(parts):
@@ -1267,7 +1395,9 @@ class GpuRunner(RunnerBase):
         if kind == "join":
             how = spec[1]
             assert len(ins) == 2, "join takes two inputs"
-            return self._device_join(ins[0], ins[1], how, stage)
+            res = self._device_join(ins[0], ins[1], how, stage)
+            res.str_table = uni_table
+            return res
         raise ValueError("unknown device_reduce spec {!r}".format(spec))
 
     def _prefetch_partition(self, ins, parts, i):
@@ -1504,7 +1634,22 @@ class GpuRunner(RunnerBase):
         datasets = []
         for store in ins:
             recs = self._decode_store(store)
+            if self.world > 1:
+                recs = self._host_exchange(recs)
             recs.sort(key=lambda r: r[0])     # reducers need sorted streams
             datasets.append([MemoryDataset(recs)])
         out_records = list(stage.reducer.reduce(*datasets))
         return self._encode_or_host(out_records)
+
+    def _host_exchange(self, recs):
+        """Host-record analog of the column exchange (one collective per
+        input store, deterministic across ranks): gather every rank's
+        records, keep the keys this rank owns by the process-stable
+        hash.  Without this, a multi-rank host-fallback reduce would
+        silently fold only the local slice of each key's values."""
+        import torch.distributed as dist
+        from ..keyhash import stable_hash64
+        gathered = [None] * self.world
+        dist.all_gather_object(gathered, recs)
+        return [r for lst in gathered for r in lst
+                if stable_hash64(r[0]) % self.world == self.rank]

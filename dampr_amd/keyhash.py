@@ -67,3 +67,14 @@ def key_hash64(key) -> int:
 def partition_of(key, n_partitions: int) -> int:
     """Partition assignment; device kernels use hash % n identically."""
     return key_hash64(key) % n_partitions
+
+
+def stable_hash64(key) -> int:
+    """key_hash64, but process-independent for EVERY key type: arbitrary
+    objects hash via their repr instead of the per-process-salted
+    interpreter hash.  Used for cross-rank ownership decisions, where
+    all ranks must compute identical values (spawned ranks do not share
+    a hash salt the way forked workers do)."""
+    if isinstance(key, (bool, int, str, bytes)):
+        return key_hash64(key)
+    return fnv1a64(repr(key).encode("utf-8"))

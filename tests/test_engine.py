@@ -729,3 +729,92 @@ def test_engine_gloo_device_text(world):
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def test_unify_str_stores_remap():
+    """Cross-encode dictionary remap (ROADMAP 2): two stores whose rank
+    ids came from different string tables combine on device through the
+    union table; decoded records must keep their original strings."""
+    import torch
+    from dampr_amd.gpu.engine import DeviceRun, GpuRunner, PartStore
+    from dampr_amd.runner import Graph
+    r = GpuRunner("unify-test", Graph())
+    a = PartStore(str_table=("apple", "fig"))
+    a[0] = [DeviceRun(torch.tensor([0, 1, 0]), torch.tensor([1, 2, 3]),
+                      sorted=True)]
+    b = PartStore(str_table=("apple", "kiwi"))
+    b[0] = [DeviceRun(torch.tensor([0, 1]), torch.tensor([4, 5]),
+                      sorted=True)]
+    for run in a[0] + b[0]:
+        r.pool.admit(run)
+    ua, ub = r._unify_str_stores([a, b])
+    assert ua.str_table == ub.str_table == ("apple", "fig", "kiwi")
+    got_a = sorted(r._decode_store(ua))
+    got_b = sorted(r._decode_store(ub))
+    assert got_a == [("apple", 1), ("apple", 3), ("fig", 2)]
+    assert got_b == [("apple", 4), ("kiwi", 5)]
+    # dictionary + dictionary-less mix: no unification
+    c = PartStore()
+    c[0] = [DeviceRun(torch.tensor([7]), torch.tensor([8]), sorted=True)]
+    assert r._unify_str_stores([a, c]) is None
+    r.pool.cleanup()
+
+
+def _engine_rank_strkeys(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from dampr_amd.gpu.engine import GpuRunner
+
+        # string keys whose per-rank slices have DIFFERENT vocabularies:
+        # a per-rank dictionary would give 'fig' and 'kiwi' the same id
+        items = ["apple", "fig"] * 50 + ["apple", "kiwi"] * 50
+        got = sorted(Dampr.memory(items).count()
+                     .run(runner=GpuRunner).read())
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got)
+        merged = sorted(p for lst in gathered for p in lst)
+        assert merged == [("apple", 100), ("fig", 50), ("kiwi", 50)], \
+            merged
+
+        # object (string) VALUES force the host-record path: the reduce
+        # must still see every rank's values for a key
+        pairs = [("a", "x"), ("b", "y"), ("a", "z"), ("b", "w")] * 10
+        got2 = sorted(Dampr.memory(pairs)
+                      .group_by(lambda kv: kv[0], lambda kv: kv[1])
+                      .reduce(lambda _k, vs: "".join(sorted(vs)))
+                      .run(runner=GpuRunner).read())
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got2)
+        merged2 = sorted(p for lst in gathered for p in lst)
+        assert merged2 == [("a", "x" * 10 + "z" * 10),
+                           ("b", "w" * 10 + "y" * 10)], merged2
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_engine_gloo_string_keys(world):
+    """Multi-rank string keys: the encode agrees on ONE dictionary via a
+    single metadata all_gather, and host-fallback reduces exchange
+    records by a process-stable hash (regression: per-rank dictionaries
+    conflated different strings; host reduces folded only local
+    slices)."""
+    port = 29000 + (os.getpid() + 40 + world) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_strkeys,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
