@@ -207,3 +207,54 @@ def test_string_keys_device_gpu():
                .run(runner=GpuRunner, device="cuda:0").read())
     want = collections.Counter(words)
     assert got == dict(want)
+
+
+def test_join_probe_batching_device(monkeypatch):
+    """Skewed-join guard on HipOps: batched probes + build-side swap
+    must match the exact oracle (same pipelines as the CPU-forced
+    tests in test_engine.py)."""
+    monkeypatch.setenv("DAMPR_JOIN_PROBE_ROWS", "1000")
+    rng = np.random.RandomState(7)
+    lk = np.concatenate([np.full(5000, 5), rng.randint(0, 50, 500)]) \
+        .astype(np.int64)
+    lv = rng.randint(1, 100, lk.size).astype(np.int64)
+    rk = np.array([5, 5, 7, 9], dtype=np.int64)
+    rv = np.array([2, 3, 4, 5], dtype=np.int64)
+    got = sorted(_run_dev(
+        Dampr.columns(lv, keys=lk).join(Dampr.columns(rv, keys=rk))
+        .reduce(funcs.pair_product, many=True)).read())
+    rmap = {}
+    for j, k in enumerate(rk):
+        rmap.setdefault(int(k), []).append(int(rv[j]))
+    want = sorted((int(k), int(v) * w)
+                  for k, v in zip(lk, lv) for w in rmap.get(int(k), []))
+    assert got == want
+    # swapped: oversized build side
+    got2 = sorted(_run_dev(
+        Dampr.columns(rv, keys=rk).join(Dampr.columns(lv, keys=lk))
+        .reduce(funcs.pair_product, many=True)).read())
+    want2 = sorted((int(k), int(v) * w) for k, v in zip(lk, lv)
+                   for w in rmap.get(int(k), []))
+    assert got2 == want2
+
+
+def test_unify_str_stores_device():
+    """Cross-encode dictionary remap runs on the HIP backend."""
+    from dampr_amd.gpu.engine import DeviceRun, GpuRunner, PartStore
+    from dampr_amd.runner import Graph
+    dev = torch.device("cuda:0")
+    r = GpuRunner("unify-gpu", Graph(), device=dev)
+    a = PartStore(str_table=("apple", "fig"))
+    a[0] = [DeviceRun(torch.tensor([0, 1, 0], device=dev),
+                      torch.tensor([1, 2, 3], device=dev), sorted=True)]
+    b = PartStore(str_table=("apple", "kiwi"))
+    b[0] = [DeviceRun(torch.tensor([0, 1], device=dev),
+                      torch.tensor([4, 5], device=dev), sorted=True)]
+    for run in a[0] + b[0]:
+        r.pool.admit(run)
+    ua, ub = r._unify_str_stores([a, b])
+    assert ua.str_table == ("apple", "fig", "kiwi")
+    assert sorted(r._decode_store(ua)) == [("apple", 1), ("apple", 3),
+                                           ("fig", 2)]
+    assert sorted(r._decode_store(ub)) == [("apple", 4), ("kiwi", 5)]
+    r.pool.cleanup()
