@@ -1624,10 +1624,26 @@ class GpuRunner(RunnerBase):
                     MemoryDataset(self._decode_store(store))))
             return self._encode_or_host(out_records)
         primary = MemoryDataset(self._decode_store(ins[0]))
-        supplemental = [[MemoryDataset(self._decode_store(s))]
-                        for s in ins[1:]]
+        supplemental = []
+        for s in ins[1:]:
+            recs = self._decode_store(s)
+            if self.world > 1:
+                # supplemental sides (cross/broadcast joins) must be
+                # COMPLETE on every rank: computed stores are rank-owned
+                # shards and raw inputs are rank slices, so gather the
+                # union (the RCCL-bcast analog of K9's broadcast side)
+                recs = self._host_gather(recs)
+            supplemental.append([MemoryDataset(recs)])
         out_records = list(stage.mapper.map(primary, *supplemental))
         return self._encode_or_host(out_records)
+
+    def _host_gather(self, recs):
+        """Union of every rank's records, identical on all ranks (one
+        collective; deterministic rank order)."""
+        import torch.distributed as dist
+        gathered = [None] * self.world
+        dist.all_gather_object(gathered, recs)
+        return [r for lst in gathered for r in lst]
 
     def _host_reduce(self, stage, ins):
         from ..dataset import MemoryDataset

@@ -818,3 +818,61 @@ def test_engine_gloo_string_keys(world):
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def _engine_rank_cross(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from dampr_amd.gpu.engine import GpuRunner
+
+        # word-stats idiom: every rank must see the COMPLETE computed
+        # total on the supplemental side (regression: rank slices /
+        # rank-owned shards crossed partially)
+        vals = list(range(1, 9)) * 5          # sum = 180
+        got = Dampr.memory(vals).cross_right(
+            Dampr.memory(vals).fold_by(lambda _v: 1,
+                                       lambda a, b: a + b),
+            lambda v, kv: v * kv[1]).run(runner=GpuRunner).read()
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got)
+        merged = sorted(p for lst in gathered for p in lst)
+        assert merged == sorted(v * 180 for v in vals), merged[:6]
+
+        # cross_set: aggregate side must be the full multi-rank union.
+        # NB the reference streams `other` and aggregates `self`
+        # (its docstring example contradicts its own behavior); we
+        # match the behavior, so the output is one record per `other`
+        # element
+        got2 = Dampr.memory([1, 2, 3, 4, 5, 6]).cross_set(
+            Dampr.memory([2, 9]), lambda v, right: v in right,
+            agg=set).run(runner=GpuRunner).read()
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got2)
+        merged2 = sorted(p for lst in gathered for p in lst)
+        assert merged2 == [False, True], merged2
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [2])
+def test_engine_gloo_cross_joins(world):
+    """Multi-rank cross/broadcast joins gather the supplemental side."""
+    port = 29000 + (os.getpid() + 60) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_cross,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
