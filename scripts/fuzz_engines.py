@@ -63,6 +63,63 @@ def main(trials=500, seed=123):
     return fails
 
 
+
+
+
+def main_str(trials=300, seed=321):
+    """String-keyed variant: records (word, int) through group/count/
+    first/join-free pipelines; exercises dictionary encode + remap."""
+    rng = random.Random(seed)
+    words = ["apple", "fig", "kiwi", "pear", "plum", "yam", "oat",
+             "rye", "teff", "corn", "bean", "lime"]
+    fails = 0
+    for trial in range(trials):
+        n = rng.randint(1, 300)
+        vocab = rng.sample(words, rng.randint(1, len(words)))
+        recs = [(rng.choice(vocab), rng.randint(-50, 50))
+                for _ in range(n)]
+        op = rng.choice(["count", "sum", "min", "max", "first", "mean",
+                         "concat_count"])
+
+        def build(D, recs=recs, op=op):
+            pm = D.memory(recs)
+            if op == "concat_count":
+                half = len(recs) // 2
+                return D.memory(recs[:half]).concat(
+                    D.memory(recs[half:])).count(lambda kv: kv[0])
+            g = pm.a_group_by(lambda kv: kv[0], lambda kv: kv[1])
+            if op == "count":
+                return pm.count(lambda kv: kv[0])
+            if op == "sum":
+                return g.sum()
+            if op == "min":
+                return g.reduce(min)
+            if op == "max":
+                return g.reduce(max)
+            if op == "first":
+                return g.reduce(lambda a, _b: a)
+            return pm.mean(lambda kv: kv[0], lambda kv: kv[1])
+
+        from dampr_amd.gpu.engine import GpuRunner
+        dev = sorted(map(repr, build(Dampr).run(runner=GpuRunner)
+                         .read()))
+        host = sorted(map(repr, build(Dampr).run(
+            runner=MTRunner, n_maps=2, n_reducers=2).read()))
+        if op == "first":
+            # first-seen is scheduling-dependent; compare keys only
+            dev = sorted(r.split(",")[0] for r in dev)
+            host = sorted(r.split(",")[0] for r in host)
+        if dev != host:
+            fails += 1
+            print("STR MISMATCH", trial, op, n, dev[:3], host[:3])
+            if fails > 5:
+                break
+        if trial and trial % 100 == 0:
+            print("... str", trial, "trials, fails:", fails)
+    print("done str:", trials, "trials, fails:", fails)
+    return fails
+
+
 if __name__ == "__main__":
-    sys.exit(1 if main(int(sys.argv[1])
-                       if len(sys.argv) > 1 else 500) else 0)
+    t = int(sys.argv[1]) if len(sys.argv) > 1 else 500
+    sys.exit(1 if (main(t) + main_str(max(t // 2, 100))) else 0)
