@@ -876,3 +876,49 @@ def test_engine_gloo_cross_joins(world):
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def _engine_rank_mixed_kinds(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from dampr_amd.gpu.engine import GpuRunner
+
+        # rank 0's slice has int keys only, rank 1's has floats: a
+        # per-rank encode would exchange i64 against f64-encoded
+        # columns; the world encode must agree on one layout
+        items = [(1, 10), (2, 20)] * 25 + [(1.5, 5), (2.5, 7)] * 25
+        got = sorted(Dampr.memory(items)
+                     .a_group_by(lambda kv: kv[0], lambda kv: kv[1])
+                     .sum().run(runner=GpuRunner).read())
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got)
+        merged = sorted(p for lst in gathered for p in lst)
+        assert merged == [(1, 250), (1.5, 125), (2, 500), (2.5, 175)], \
+            merged
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [2])
+def test_engine_gloo_mixed_key_kinds(world):
+    """Ranks whose record slices disagree on key dtype agree on one
+    layout through the world encode (f64 order-preserving keys)."""
+    port = 29000 + (os.getpid() + 70) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_mixed_kinds,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
