@@ -1272,7 +1272,9 @@ class GpuRunner(RunnerBase):
     def _merge_stores(self, stores):
         if len(stores) == 1:
             return stores[0]
-        if any(isinstance(s, HostStore) for s in stores):
+        if any(not isinstance(s, PartStore) for s in stores):
+            # HostStore / TokenStore / TextSource in the mix: combine as
+            # host records (only PartStores share the run layout)
             out = HostStore()
             for s in stores:
                 out.extend(self._decode_store(s))
