@@ -713,11 +713,11 @@ def _engine_rank_text(rank, world, port, q):
         q.put((rank, traceback.format_exc()))
 
 
-@pytest.mark.parametrize("world", [2])
+@pytest.mark.parametrize("world", [2, 3])
 def test_engine_gloo_device_text(world):
     """Multi-rank device_text (ROADMAP 3): each rank counts its
     newline-aligned slice; partial dfs merge across ranks."""
-    port = 29000 + (os.getpid() + 20) % 900
+    port = 29000 + (os.getpid() + 20 + world) % 900
     ctx = multiprocessing.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_engine_rank_text, args=(r, world, port, q))
@@ -860,10 +860,10 @@ def _engine_rank_cross(rank, world, port, q):
         q.put((rank, traceback.format_exc()))
 
 
-@pytest.mark.parametrize("world", [2])
+@pytest.mark.parametrize("world", [2, 3])
 def test_engine_gloo_cross_joins(world):
     """Multi-rank cross/broadcast joins gather the supplemental side."""
-    port = 29000 + (os.getpid() + 60) % 900
+    port = 29000 + (os.getpid() + 60 + world) % 900
     ctx = multiprocessing.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_engine_rank_cross,
@@ -905,11 +905,11 @@ def _engine_rank_mixed_kinds(rank, world, port, q):
         q.put((rank, traceback.format_exc()))
 
 
-@pytest.mark.parametrize("world", [2])
+@pytest.mark.parametrize("world", [2, 3])
 def test_engine_gloo_mixed_key_kinds(world):
     """Ranks whose record slices disagree on key dtype agree on one
     layout through the world encode (f64 order-preserving keys)."""
-    port = 29000 + (os.getpid() + 70) % 900
+    port = 29000 + (os.getpid() + 70 + world) % 900
     ctx = multiprocessing.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_engine_rank_mixed_kinds,
