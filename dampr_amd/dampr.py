@@ -678,7 +678,12 @@ class Dampr(object):
         int64 or float64): records are (key_i, val_i), keys default to the
         row index.  Runs on the device engine — built-in ops (count, sum,
         fold_by with recognized binops, joins, first, ...) execute as
-        gfx950 kernels over HBM-resident columns."""
+        gfx950 kernels over HBM-resident columns.
+
+        ``keys`` may also be a string array/list: keys dictionary-encode
+        at ingest (sorted table, rank ids) and run on the same kernels;
+        joins across different vocabularies remap through the union
+        dictionary on device."""
         from .gpu.engine import ColumnSource
         src_obj = ColumnSource.from_data(vals, keys)
         source, ng = Graph().add_input(src_obj)

@@ -65,27 +65,57 @@ def _as_column(x, device=None, dtype=None):
 
 
 class ColumnSource(object):
-    """A device-columnar input: (keys i64, vals i64|f64)."""
+    """A device-columnar input: (keys i64, vals i64|f64).  String key
+    arrays dictionary-encode at ingest (np.unique: sorted uniques +
+    inverse ids, so rank order == lexicographic order); the table is
+    built from the FULL input before any per-rank slicing, so every
+    rank shares one dictionary."""
 
     dampr_columnar = True          # engine-selection sentinel (dampr.py)
 
-    def __init__(self, keys, vals):
+    def __init__(self, keys, vals, str_table=None):
         assert keys.dtype == torch.int64
         assert vals.dtype in _VAL_DTYPES
         assert keys.numel() == vals.numel()
         self.keys = keys
         self.vals = vals
+        self.str_table = str_table
+
+    @staticmethod
+    def _is_str_array(x):
+        import numpy as np
+        if isinstance(x, np.ndarray):
+            return x.dtype.kind in ("U", "S") or (
+                x.dtype.kind == "O" and x.size
+                and all(isinstance(e, str) for e in x.flat))
+        return (isinstance(x, (list, tuple)) and len(x)
+                and all(isinstance(e, str) for e in x))
 
     @classmethod
     def from_data(cls, vals, keys=None, device=None):
+        import numpy as np
+        if cls._is_str_array(vals):
+            raise TypeError(
+                "string VALUE columns are not supported on the device "
+                "engine (string keys are); use the host engine")
         v = _as_column(vals, device)
+        str_table = None
         if keys is None:
             k = torch.arange(v.numel(), dtype=torch.int64, device=v.device)
+        elif cls._is_str_array(keys):
+            arr = np.asarray(keys)
+            if arr.dtype.kind == "S":
+                arr = arr.astype("U")
+            uniq, inv = np.unique(arr, return_inverse=True)
+            str_table = tuple(str(u) for u in uniq)
+            k = torch.from_numpy(inv.astype(np.int64))
+            if device is not None or k.device != v.device:
+                k = k.to(v.device)
         else:
             k = _as_column(keys, device, torch.int64)
             if k.device != v.device:
                 k = k.to(v.device)
-        return cls(k, v)
+        return cls(k, v, str_table=str_table)
 
 
 class TextSource(object):
@@ -676,13 +706,18 @@ class GpuRunner(RunnerBase):
                 lo = n * self.rank // self.world
                 hi = n * (self.rank + 1) // self.world
                 keys, vals = keys[lo:hi], vals[lo:hi]
-                return self._partition(keys, vals)
+                st = self._partition(keys, vals)
+                st.str_table = inp.str_table
+                return st
             if self.n_partitions == 1:
-                return self._partition(keys, vals)
+                st = self._partition(keys, vals)
+                st.str_table = inp.str_table
+                return st
             # lazy ingest: batched unpartitioned runs — partitioning by
             # the input keys is wasted work (and a wasted spill round
             # trip) when the first stage re-keys anyway
-            store = PartStore(partitioned=False)
+            store = PartStore(partitioned=False,
+                              str_table=inp.str_table)
             store[0] = []
             n = keys.numel()
             step = max(1, settings.gpu_batch_records)
@@ -958,9 +993,26 @@ class GpuRunner(RunnerBase):
     def _ensure_partitioned(self, store):
         """Route an unpartitioned store's runs to hash partitions, one
         run at a time (each run is bounded, so this streams through the
-        pool)."""
+        pool).  At world > 1 the runs concatenate FIRST: run counts
+        differ across ranks, and _partition exchanges — one collective
+        per stage per rank is the invariant."""
         if not isinstance(store, PartStore) or store.partitioned:
             return store
+        if self.world > 1:
+            ks, vs = [], []
+            for run in store.get(0, []):
+                self.pool.touch(run, self.device)
+                ks.append(run.keys)
+                vs.append(run.vals)
+                self.pool.release(run)
+            k = torch.cat(ks) if ks else torch.zeros(
+                0, dtype=torch.int64, device=self.device)
+            v = torch.cat(vs) if vs else torch.zeros(
+                0, dtype=torch.int64, device=self.device)
+            out = self._partition(k, v, keyed=store.keyed,
+                                  fkeys=store.fkeys)
+            out.str_table = getattr(store, "str_table", None)
+            return out
         out = None
         for run in store.get(0, []):
             self.pool.touch(run, self.device)

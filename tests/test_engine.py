@@ -922,3 +922,90 @@ def test_engine_gloo_mixed_key_kinds(world):
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+# ------------------------------------------------- string-keyed columns
+
+def test_columns_string_keys_collect():
+    # non-keyed collect yields values (ValueEmitter contract); the
+    # string keys surface through keyed ops -- see the join test
+    vals = np.array([3, 1, 2, 9], dtype=np.int64)
+    keys = np.array(["pear", "apple", "pear", "fig"])
+    got = sorted(Dampr.columns(vals, keys=keys).run().read())
+    assert got == [1, 2, 3, 9]
+
+
+def test_columns_string_keys_device_join():
+    """String-keyed columns join ON DEVICE: different vocabularies remap
+    through the union dictionary (the hash-join kernel path, not the
+    host fallback)."""
+    lk = np.array(["apple", "apple", "fig", "yam"])
+    lv = np.array([1, 2, 3, 4], dtype=np.int64)
+    rk = np.array(["apple", "kiwi", "fig"])
+    rv = np.array([10, 20, 30], dtype=np.int64)
+    out = Dampr.columns(lv, keys=lk).join(Dampr.columns(rv, keys=rk)) \
+        .reduce(funcs.pair_sum, many=True).run()
+    got = sorted(out.read())
+    assert got == [("apple", 11), ("apple", 12), ("fig", 33)]
+
+
+def test_columns_string_keys_world_pre_slice_table():
+    """The dictionary is built from the FULL input before rank slicing
+    (single-rank here; the world path shares the same table object)."""
+    from dampr_amd.gpu.engine import ColumnSource
+    src = ColumnSource.from_data(
+        np.arange(4), keys=["b", "a", "b", "c"])
+    assert src.str_table == ("a", "b", "c")
+    assert src.keys.tolist() == [1, 0, 1, 2]
+
+
+def test_columns_string_values_rejected():
+    with pytest.raises(TypeError):
+        Dampr.columns(np.array(["x", "y"]))
+
+
+def _engine_rank_strcols(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        lk = np.array(["apple", "apple", "fig", "yam"] * 20)
+        lv = np.arange(80, dtype=np.int64)
+        rk = np.array(["apple", "kiwi", "fig"] * 10)
+        rv = np.arange(30, dtype=np.int64)
+        out = Dampr.columns(lv, keys=lk).join(
+            Dampr.columns(rv, keys=rk)) \
+            .reduce(funcs.pair_sum, many=True).run()
+        got = sorted(out.read())
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got)
+        merged = sorted(p for lst in gathered for p in lst)
+        want = sorted((str(k), int(a) + int(b))
+                      for k, a in zip(lk, lv)
+                      for k2, b in zip(rk, rv) if k == k2)
+        assert merged == want, (len(merged), len(want))
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [2])
+def test_engine_gloo_string_columns_join(world):
+    """Multi-rank string-keyed column join: the ingest dictionary is
+    built pre-slice, so every rank shares it."""
+    port = 29000 + (os.getpid() + 80 + world) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_strcols,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)

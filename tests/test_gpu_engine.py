@@ -258,3 +258,17 @@ def test_unify_str_stores_device():
                                            ("fig", 2)]
     assert sorted(r._decode_store(ub)) == [("apple", 4), ("kiwi", 5)]
     r.pool.cleanup()
+
+
+def test_columns_string_keys_join_device():
+    """String-keyed columns: dictionary encode at ingest + union remap
+    + hash-join kernel on device."""
+    lk = np.array(["apple", "apple", "fig", "yam"])
+    lv = np.array([1, 2, 3, 4], dtype=np.int64)
+    rk = np.array(["apple", "kiwi", "fig"])
+    rv = np.array([10, 20, 30], dtype=np.int64)
+    out = _run_dev(
+        Dampr.columns(lv, keys=lk).join(Dampr.columns(rv, keys=rk))
+        .reduce(funcs.pair_sum, many=True))
+    assert sorted(out.read()) == [("apple", 11), ("apple", 12),
+                                  ("fig", 33)]
