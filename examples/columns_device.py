@@ -50,6 +50,18 @@ def main():
     wc = Dampr.memory(words).count().run("wc", runner=GpuRunner)
     print("distinct words:", len(wc.read()))
 
+    # string KEY columns: dictionary-encoded at ingest, joined on device
+    # across different vocabularies (union-dictionary remap)
+    accounts = np.array(["acct%03d" % i
+                         for i in rng.integers(0, 800, size=300_000)])
+    balances = rng.integers(1, 1000, size=300_000)
+    flagged = np.array(["acct%03d" % i for i in range(0, 1000, 7)])
+    ones = np.ones(flagged.size, dtype=np.int64)
+    hits = Dampr.columns(balances, keys=accounts) \
+        .join(Dampr.columns(ones, keys=flagged)) \
+        .reduce(funcs.pair_left, many=True).run("flagged")
+    print("flagged-account records:", len(hits.read()))
+
 
 if __name__ == "__main__":
     main()
