@@ -1009,3 +1009,12 @@ def test_engine_gloo_string_columns_join(world):
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def test_device_sort_by_strings():
+    """sort_by over string records through the device engine: the
+    dictionary's rank order is lexicographic, so sorted ids decode to
+    host-ordered strings."""
+    words = ["pear", "apple", "fig", "apple", "kiwi"]
+    got = Dampr.memory(words).sort_by().run(runner=GpuRunner).read()
+    assert got == sorted(words)
