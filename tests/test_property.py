@@ -91,3 +91,57 @@ def test_composite_pipelines_agree(vals, mod, thresh):
     host = sorted(build(Dampr.memory(arr.tolist()))
                   .run(runner=MTRunner, n_maps=2, n_reducers=2).read())
     assert dev == host
+
+
+_WORDS = ["apple", "fig", "kiwi", "pear", "plum", "yam", "oat", "rye"]
+
+
+@hyp_settings(max_examples=25, deadline=None, derandomize=True)
+@given(
+    recs=st.lists(
+        st.tuples(st.sampled_from(_WORDS),
+                  st.integers(min_value=-100, max_value=100)),
+        min_size=1, max_size=120),
+    op=st.sampled_from(["count", "sum", "min", "max"]),
+)
+def test_string_key_engines_agree(recs, op):
+    """String-keyed group pipelines: device engine (dictionary encode)
+    vs host engine."""
+    def build(D):
+        pm = D.memory(recs)
+        if op == "count":
+            return pm.count(lambda kv: kv[0])
+        g = pm.a_group_by(lambda kv: kv[0], lambda kv: kv[1])
+        if op == "sum":
+            return g.sum()
+        return g.reduce(min if op == "min" else max)
+
+    from dampr_amd.gpu.engine import GpuRunner
+    dev = sorted(map(repr, build(Dampr).run(runner=GpuRunner).read()))
+    host = sorted(map(repr, build(Dampr).run(
+        runner=MTRunner, n_maps=2, n_reducers=2).read()))
+    assert dev == host
+
+
+@hyp_settings(max_examples=20, deadline=None, derandomize=True)
+@given(
+    lk=st.lists(st.sampled_from(_WORDS), min_size=1, max_size=40),
+    rk=st.lists(st.sampled_from(_WORDS), min_size=1, max_size=40),
+)
+def test_string_key_join_engines_agree(lk, rk):
+    """String-keyed device joins (union-dictionary remap) vs the exact
+    oracle."""
+    lka = np.array(lk)
+    rka = np.array(rk)
+    lva = np.arange(len(lk), dtype=np.int64)
+    rva = np.arange(len(rk), dtype=np.int64) * 3
+    dev = Dampr.columns(lva, keys=lka) \
+        .join(Dampr.columns(rva, keys=rka)) \
+        .reduce(funcs.pair_sum, many=True).run()
+    got = sorted(dev.read())
+    want = []
+    for i, k in enumerate(lk):
+        for j, k2 in enumerate(rk):
+            if k == k2:
+                want.append((k, int(lva[i] + rva[j])))
+    assert got == sorted(want)
