@@ -63,20 +63,30 @@ def tokenize_set(line):
 # PJoin.reduce(aggregate, many=True) aggregates the cartesian product of
 # each key's (left values, right values).  These named per-pair forms let
 # the device engine emit the product straight from the hash-join kernel.
+#
+# The right side is materialized first: the engines hand one-pass group
+# iterators to aggregates (as the reference does), so a bare nested
+# comprehension would exhaust `right` after the first left value and
+# silently drop pairs.  These named funcs are defined as the FULL cross
+# product — identical on the host path and the hash-join kernel.
 
 def pair_sum(left, right):
+    right = list(right)
     return [lv + rv for lv in left for rv in right]
 
 
 def pair_product(left, right):
+    right = list(right)
     return [lv * rv for lv in left for rv in right]
 
 
 def pair_left(left, right):
+    right = list(right)
     return [lv for lv in left for _rv in right]
 
 
 def pair_right(left, right):
+    right = list(right)
     return [rv for _lv in left for rv in right]
 
 

@@ -36,6 +36,7 @@ def engines():
 
 ITEMS = [("apple", 3), ("pear", 1), ("apple", 2), ("fig", 9),
          ("pear", 4), ("apple", 1), ("kiwi", 7)]
+
 NUMS = list(range(40)) * 3
 
 
@@ -89,6 +90,25 @@ PIPELINES = {
     "checkpoint": lambda D: D.memory(NUMS)
         .map(lambda v: v + 1).checkpoint().count(lambda v: v % 5),
     "sample_all": lambda D: D.memory(NUMS).sample(1.0).len(),
+    # joining two a_group_by().sum() outputs: the join values are the
+    # keyed-reduce (k, v) tuples in BOTH engines (reference behavior --
+    # tuple concatenation, not scalar addition)
+    "agg_join": lambda D: (
+        D.memory(ITEMS).a_group_by(lambda kv: kv[0], lambda kv: kv[1])
+        .sum()
+        .join(D.memory([("apple", 10), ("fig", 5)])
+              .a_group_by(lambda kv: kv[0], lambda kv: kv[1]).sum())
+        .reduce(lambda l, r: [a + b for a in l for b in r])),
+    # join aggregates receive ONE-PASS group iterators (reference
+    # semantics): a bare nested comprehension exhausts the right side
+    # after the first left value -- only the first pair survives.  Both
+    # our engines reproduce this for opaque lambdas (funcs.pair_* are
+    # defined as the full product instead; see test_engine.py)
+    "groupby_join_lazy": lambda D: (
+        D.memory(ITEMS).group_by(lambda kv: kv[0], lambda kv: kv[1])
+        .join(D.memory([("apple", 10), ("fig", 5)])
+              .group_by(lambda kv: kv[0], lambda kv: kv[1]))
+        .reduce(lambda l, r: [a + b for a in l for b in r], many=True)),
 }
 
 
