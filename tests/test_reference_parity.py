@@ -176,6 +176,17 @@ def test_parity_text_input(engines, tmp_path):
     assert got == want and want
 
 
+def test_parity_json_input(engines, tmp_path):
+    RefD, OursD = engines
+    f = tmp_path / "data.json"
+    f.write_text('{"a": 1}\n{"a": 2}\n{"a": 5}\n')
+    want = sorted(RefD.json(str(f)).map(lambda d: d["a"] * 2)
+                  .run().read())
+    got = sorted(OursD.json(str(f)).map(lambda d: d["a"] * 2)
+                 .run().read())
+    assert got == want and want
+
+
 def test_parity_sink(engines, tmp_path):
     RefD, OursD = engines
     rd = tmp_path / "ref"
@@ -304,3 +315,6 @@ def test_parity_indexer(engines, tmp_path):
     want = sorted(ri.union(["red"]).read())
     got = sorted(oi.union(["red"]).read())
     assert got == want and want
+    wanti = sorted(ri.intersect(["red", "apple"]).read())
+    goti = sorted(oi.intersect(["red", "apple"]).read())
+    assert goti == wanti and wanti
