@@ -58,11 +58,11 @@ def _run_rank(rank, world, port, q):
         q.put((rank, traceback.format_exc()))
 
 
-def test_exchange_gloo_world2():
+@pytest.mark.parametrize("world", [2, 3])
+def test_exchange_gloo(world):
     ctx = multiprocessing.get_context("spawn")
     q = ctx.Queue()
-    port = 29000 + (os.getpid() + 40) % 900
-    world = 2
+    port = 29000 + (os.getpid() + 90 + world) % 900
     procs = [ctx.Process(target=_run_rank, args=(r, world, port, q))
              for r in range(world)]
     for p in procs:
