@@ -50,13 +50,19 @@ COLUMN_FUNCS = {
 }
 
 
+_TOKEN_RE = None
+
+
 def tokenize_set(line):
     """Distinct ASCII word-tokens of a line, lowercased — the host mirror
     of the device tokenizer (ops/hip: is_word/lower_ascii).  Built-in so
     ``device_text(...).flat_map(funcs.tokenize_set).count()`` lowers to
     the fused single-pass document-frequency kernel."""
-    import re
-    return set(re.findall(r"[a-z0-9_]+", line.lower()))
+    global _TOKEN_RE
+    if _TOKEN_RE is None:
+        import re
+        _TOKEN_RE = re.compile(r"[a-z0-9_]+")
+    return set(_TOKEN_RE.findall(line.lower()))
 
 
 # --- join pair aggregates ---------------------------------------------------
