@@ -44,9 +44,19 @@ def main(trials=60, seed=99):
         k = rng.randint(1, 8)
         op = rng.choice(
             ["count", "sort", "topk", "len", "fold", "group_sum",
-             "filter_count", "prefix_count", "chain"])
+             "filter_count", "prefix_count", "chain",
+             "join_sum", "left_join", "cross_right", "agg_join"])
+        if op in ("join_sum", "left_join", "agg_join") \
+                and kind == "ints":
+            # int keys joined against string rdata keys would crash the
+            # reference worker (its parent then deadlocks -- no fault
+            # tolerance); keep join key types aligned
+            op = "count"
+        rdata = [(rng.choice(words), rng.randint(-9, 9))
+                 for _ in range(rng.randint(1, 20))]
 
-        def build(D, data=data, kind=kind, mod=mod, k=k, op=op):
+        def build(D, data=data, kind=kind, mod=mod, k=k, op=op,
+                  rdata=rdata):
             pm = D.memory(data)
             keyf = ((lambda kv: kv[0]) if kind == "pairs"
                     else (lambda v: v))
@@ -66,6 +76,39 @@ def main(trials=60, seed=99):
             if op == "group_sum":
                 return pm.group_by(keyf, valf) \
                     .reduce(lambda _k, vs: sum(vs))
+            if op == "join_sum":
+                kf = ((lambda kv: kv[0]) if kind == "pairs"
+                      else (lambda v: v))
+                vf = ((lambda kv: kv[1]) if kind == "pairs"
+                      else (lambda v: 1))
+                return pm.group_by(kf, vf).join(
+                    D.memory(rdata).group_by(lambda kv: kv[0],
+                                             lambda kv: kv[1])) \
+                    .reduce(lambda l, r: (sorted(l), sorted(r)))
+            if op == "left_join":
+                kf = ((lambda kv: kv[0]) if kind == "pairs"
+                      else (lambda v: v))
+                return pm.group_by(kf).join(
+                    D.memory(rdata).group_by(lambda kv: kv[0])) \
+                    .left_reduce(lambda l, r: (sorted(map(repr, l)),
+                                               sorted(map(repr, r))))
+            if op == "cross_right":
+                return pm.cross_right(
+                    D.memory(rdata).fold_by(lambda _x: 1,
+                                            lambda a, b: a + b,
+                                            value=lambda kv: kv[1]),
+                    lambda v, kv: (repr(v), kv[1]))
+            if op == "agg_join":
+                kf = ((lambda kv: kv[0]) if kind == "pairs"
+                      else (lambda v: v))
+                vf = ((lambda kv: kv[1]) if kind == "pairs"
+                      else (lambda v: 1))
+                return pm.a_group_by(kf, vf).sum().join(
+                    D.memory(rdata)
+                    .a_group_by(lambda kv: kv[0],
+                                lambda kv: kv[1]).sum()) \
+                    .reduce(lambda l, r: (sorted(map(repr, l)),
+                                          sorted(map(repr, r))))
             if op == "filter_count":
                 return pm.filter(lambda v: hash(repr(v)) % 2 == 0) \
                     .count(keyf)
