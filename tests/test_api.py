@@ -602,3 +602,20 @@ def test_multi_output_delete_keeps_sibling(tmp_path):
     assert sorted(rt.read()) == [(1, 5)]
     rt.delete()
     assert not os.path.exists(os.path.join(str(tmp_path), "sibcheck"))
+
+
+def test_mixed_type_join_never_hangs():
+    """Joining int keys against string keys: cross-type keys are never
+    equal, so the join is empty -- unless both types collide in one
+    partition, where the comparison raises and the engine FAILS FAST
+    (the reference deadlocks forever on the same pipeline: its crashed
+    worker never posts to the result queue)."""
+    from dampr_amd.executor import WorkerCrash
+    try:
+        got = Dampr.memory([1, 2, 3]).group_by(lambda v: v).join(
+            Dampr.memory([("a", 1), ("b", 2)])
+            .group_by(lambda kv: kv[0])) \
+            .reduce(lambda l, r: (list(l), list(r))).run().read()
+        assert got == []
+    except WorkerCrash:
+        pass                       # collision partition: fail-fast
