@@ -251,7 +251,9 @@ def _decode_f64_sortable(enc):
 
 
 def _encode_f64_sortable(x):
-    b = x.view(torch.int64)
+    # + 0.0 canonicalizes -0.0 to +0.0 (Python == merges them as one
+    # group key; distinct bit patterns would split it)
+    b = (x + 0.0).view(torch.int64)
     sign_bit = -(1 << 63)
     return torch.where(b < 0, ~b, b ^ sign_bit)
 

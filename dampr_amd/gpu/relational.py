@@ -146,8 +146,10 @@ def topk_by(values_u64, k, largest=True):
 
 def encode_f64_sortable(x):
     """Map float64 to u64 preserving order (IEEE trick): sortable keys for
-    sort_by/topk on floats."""
-    b = x.view(torch.int64)
+    sort_by/topk on floats.  ``+ 0.0`` canonicalizes -0.0 to +0.0 first
+    (Python == merges them; distinct bit patterns would split the
+    group), and leaves every other value including NaN unchanged."""
+    b = (x + 0.0).view(torch.int64)
     neg = b < 0
     sign_bit = -(1 << 63)                 # 0x8000000000000000 as int64
     out = torch.where(neg, ~b, b ^ sign_bit)

@@ -1043,3 +1043,14 @@ def test_pair_funcs_full_product_both_engines():
     want = sorted((k, lv + rv) for k, lv in items
                   for k2, rv in right if k == k2)
     assert dev == want and host == want
+
+
+def test_negative_zero_float_keys_merge():
+    """-0.0 and 0.0 are one group (Python ==): the f64 sortable encode
+    canonicalizes the sign bit of zero before grouping."""
+    from dampr_amd.runner import MTRunner
+    vals = np.array([0.0, -0.0, 1.5, -0.0], dtype=np.float64)
+    dev = sorted(map(repr, Dampr.columns(vals).count().run().read()))
+    host = sorted(map(repr, Dampr.memory(vals.tolist()).count()
+                      .run(runner=MTRunner).read()))
+    assert dev == host == ["(0.0, 3)", "(1.5, 1)"]
