@@ -683,7 +683,12 @@ class Dampr(object):
         ``keys`` may also be a string array/list: keys dictionary-encode
         at ingest (sorted table, rank ids) and run on the same kernels;
         joins across different vocabularies remap through the union
-        dictionary on device."""
+        dictionary on device.
+
+        Numeric domain: typed columns use fixed-width i64/f64
+        arithmetic — integer aggregates wrap at 64 bits, where the host
+        engine's Python ints are arbitrary precision.  Pipelines whose
+        sums exceed +/-2**63 belong on the host engine."""
         from .gpu.engine import ColumnSource
         src_obj = ColumnSource.from_data(vals, keys)
         source, ng = Graph().add_input(src_obj)
