@@ -992,7 +992,7 @@ def _engine_rank_strcols(rank, world, port, q):
         q.put((rank, traceback.format_exc()))
 
 
-@pytest.mark.parametrize("world", [2])
+@pytest.mark.parametrize("world", [2, 3])
 def test_engine_gloo_string_columns_join(world):
     """Multi-rank string-keyed column join: the ingest dictionary is
     built pre-slice, so every rank shares it."""
