@@ -1054,3 +1054,51 @@ def test_negative_zero_float_keys_merge():
     host = sorted(map(repr, Dampr.memory(vals.tolist()).count()
                       .run(runner=MTRunner).read()))
     assert dev == host == ["(0.0, 3)", "(1.5, 1)"]
+
+
+def _engine_rank_tiny(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from dampr_amd.gpu.engine import GpuRunner
+        # fewer rows than ranks: some slices are EMPTY, and empty ranks
+        # must still take part in every exchange
+        got = Dampr.columns(np.array([7, 7], dtype=np.int64)).count() \
+            .run().read()
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got)
+        merged = sorted(p for lst in gathered for p in lst)
+        assert merged == [(7, 2)], merged
+        # same through host records (string key)
+        got2 = Dampr.memory(["only"]).count().run(runner=GpuRunner) \
+            .read()
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got2)
+        merged2 = sorted(p for lst in gathered for p in lst)
+        assert merged2 == [("only", 1)], merged2
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [3])
+def test_engine_gloo_tiny_inputs(world):
+    """Inputs smaller than the world: empty rank slices still join every
+    collective."""
+    port = 29000 + (os.getpid() + 100 + world) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_tiny,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
