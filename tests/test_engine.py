@@ -1102,3 +1102,13 @@ def test_engine_gloo_tiny_inputs(world):
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def test_columns_unicode_string_keys():
+    keys = np.array(["éclair", "zèbre", "日本", "éclair", "ß", "𝄞note"])
+    vals = np.arange(6, dtype=np.int64)
+    got = sorted(Dampr.columns(vals, keys=keys).join(
+        Dampr.columns(np.array([100], dtype=np.int64),
+                      keys=np.array(["éclair"])))
+        .reduce(funcs.pair_sum, many=True).run().read())
+    assert got == [("éclair", 100), ("éclair", 103)]
