@@ -73,13 +73,15 @@ class ColumnSource(object):
 
     dampr_columnar = True          # engine-selection sentinel (dampr.py)
 
-    def __init__(self, keys, vals, str_table=None):
+    def __init__(self, keys, vals, str_table=None, fkeys=False):
         assert keys.dtype == torch.int64
         assert vals.dtype in _VAL_DTYPES
         assert keys.numel() == vals.numel()
         self.keys = keys
         self.vals = vals
         self.str_table = str_table
+        # keys hold the f64 order-preserving encode (decoded on read)
+        self.fkeys = fkeys
 
     @staticmethod
     def _is_str_array(x):
@@ -100,8 +102,21 @@ class ColumnSource(object):
                 "engine (string keys are); use the host engine")
         v = _as_column(vals, device)
         str_table = None
+        fkeys = False
         if keys is None:
             k = torch.arange(v.numel(), dtype=torch.int64, device=v.device)
+        elif (isinstance(keys, torch.Tensor)
+              and keys.dtype.is_floating_point) or (
+                  isinstance(keys, np.ndarray)
+                  and keys.dtype.kind == "f"):
+            # float keys: order-preserving f64 encode, NOT an int cast
+            # (which would silently truncate 1.5 and 2.5 into key 1)
+            kf = keys.to(torch.float64) if isinstance(keys, torch.Tensor) \
+                else torch.from_numpy(keys.astype(np.float64))
+            if kf.device != v.device:
+                kf = kf.to(v.device)
+            k = _encode_f64_sortable(kf)
+            fkeys = True
         elif cls._is_str_array(keys):
             arr = np.asarray(keys)
             if arr.dtype.kind == "S":
@@ -115,7 +130,7 @@ class ColumnSource(object):
             k = _as_column(keys, device, torch.int64)
             if k.device != v.device:
                 k = k.to(v.device)
-        return cls(k, v, str_table=str_table)
+        return cls(k, v, str_table=str_table, fkeys=fkeys)
 
 
 class TextSource(object):
@@ -708,18 +723,19 @@ class GpuRunner(RunnerBase):
                 lo = n * self.rank // self.world
                 hi = n * (self.rank + 1) // self.world
                 keys, vals = keys[lo:hi], vals[lo:hi]
-                st = self._partition(keys, vals)
+                st = self._partition(keys, vals, fkeys=inp.fkeys)
                 st.str_table = inp.str_table
                 return st
             if self.n_partitions == 1:
-                st = self._partition(keys, vals)
+                st = self._partition(keys, vals, fkeys=inp.fkeys)
                 st.str_table = inp.str_table
                 return st
             # lazy ingest: batched unpartitioned runs — partitioning by
             # the input keys is wasted work (and a wasted spill round
             # trip) when the first stage re-keys anyway
             store = PartStore(partitioned=False,
-                              str_table=inp.str_table)
+                              str_table=inp.str_table,
+                              fkeys=inp.fkeys)
             store[0] = []
             n = keys.numel()
             step = max(1, settings.gpu_batch_records)

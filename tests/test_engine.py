@@ -1112,3 +1112,15 @@ def test_columns_unicode_string_keys():
                       keys=np.array(["éclair"])))
         .reduce(funcs.pair_sum, many=True).run().read())
     assert got == [("éclair", 100), ("éclair", 103)]
+
+
+def test_columns_float_keys_join():
+    """Float key columns: order-preserving f64 encode at ingest
+    (regression: an int64 cast silently truncated 1.5 and 2.5 to 1)."""
+    keys = np.array([1.5, 2.5, 1.5], dtype=np.float64)
+    vals = np.array([1, 2, 3], dtype=np.int64)
+    out = sorted(Dampr.columns(vals, keys=keys).join(
+        Dampr.columns(np.array([10], dtype=np.int64),
+                      keys=np.array([1.5])))
+        .reduce(funcs.pair_sum, many=True).run().read())
+    assert out == [(1.5, 11), (1.5, 13)]
