@@ -1124,3 +1124,14 @@ def test_columns_float_keys_join():
                       keys=np.array([1.5])))
         .reduce(funcs.pair_sum, many=True).run().read())
     assert out == [(1.5, 11), (1.5, 13)]
+
+
+def test_device_sink_tsv_string_keys(tmp_path):
+    path = str(tmp_path / "tsv")
+    Dampr.memory([("b", 2), ("a", 1), ("b", 5)]).count(lambda kv: kv[0]) \
+        .sink_tsv(path).run(runner=GpuRunner)
+    lines = []
+    for f in sorted(os.listdir(path)):
+        with open(os.path.join(path, f)) as fh:
+            lines.extend(ln.rstrip("\n") for ln in fh if ln.strip())
+    assert sorted(lines) == ["a\t1", "b\t2"]
