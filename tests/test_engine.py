@@ -1135,3 +1135,15 @@ def test_device_sink_tsv_string_keys(tmp_path):
         with open(os.path.join(path, f)) as fh:
             lines.extend(ln.rstrip("\n") for ln in fh if ln.strip())
     assert sorted(lines) == ["a\t1", "b\t2"]
+
+
+def test_device_sink_json_objects(tmp_path):
+    import json
+    path = str(tmp_path / "j")
+    Dampr.memory([{"a": 1}, {"a": 2}]).sink_json(path) \
+        .run(runner=GpuRunner)
+    vals = []
+    for f in sorted(os.listdir(path)):
+        with open(os.path.join(path, f)) as fh:
+            vals.extend(json.loads(ln) for ln in fh if ln.strip())
+    assert sorted(v["a"] for v in vals) == [1, 2]
