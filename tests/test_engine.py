@@ -1147,3 +1147,28 @@ def test_device_sink_json_objects(tmp_path):
         with open(os.path.join(path, f)) as fh:
             vals.extend(json.loads(ln) for ln in fh if ln.strip())
     assert sorted(v["a"] for v in vals) == [1, 2]
+
+
+def test_device_engine_misc_inputs(tmp_path):
+    """Input taps through the device engine: gzip text files and
+    multi-dataset read_input (host decode, columnar shuffle core)."""
+    import gzip
+    from dampr_amd import Dataset
+    p = str(tmp_path / "x.gz")
+    with gzip.open(p, "wt") as fh:
+        fh.write("a b\nb b\n" * 10)
+    got = sorted(Dampr.text(p).flat_map(str.split).count()
+                 .run(runner=GpuRunner).read())
+    assert got == [("a", 10), ("b", 30)]
+
+    class RangeDataset(Dataset):
+        def __init__(self, n):
+            self.n = n
+
+        def read(self):
+            for i in range(self.n):
+                yield i, i
+
+    got2 = sorted(Dampr.read_input(RangeDataset(5), RangeDataset(3))
+                  .count().run(runner=GpuRunner).read())
+    assert got2 == [(0, 2), (1, 2), (2, 2), (3, 1), (4, 1)]
