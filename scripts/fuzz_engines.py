@@ -120,6 +120,57 @@ def main_str(trials=300, seed=321):
     return fails
 
 
+def main_float(trials=200, seed=77):
+    """Float-keyed variant over exact ops (count/min/max/sort: no
+    fp-summation-order noise); includes -0.0 and halves."""
+    rng = random.Random(seed)
+    pool = [-2.5, -1.5, -0.0, 0.0, 0.5, 1.5, 2.5, 3.25]
+    fails = 0
+    for trial in range(trials):
+        n = rng.randint(1, 250)
+        vals = np.array([rng.choice(pool) for _ in range(n)],
+                        dtype=np.float64)
+        op = rng.choice(["count", "min", "max", "sort", "topk"])
+        K = rng.randint(1, 9)
+
+        def build(pm, op=op, K=K):
+            if op == "count":
+                return pm.count()
+            if op == "min":
+                return pm.a_group_by().reduce(min)
+            if op == "max":
+                return pm.a_group_by().reduce(max)
+            if op == "topk":
+                return pm.topk(K)
+            return pm.sort_by()
+
+        def norm(rows):
+            # -0.0 == 0.0: which zero repr survives a host dict merge is
+            # arrival-order dependent (reference engine too); canonicalize
+            out = []
+            for r in rows:
+                if isinstance(r, tuple):
+                    out.append(tuple(x + 0.0 if isinstance(x, float)
+                                     else x for x in r))
+                else:
+                    out.append(r + 0.0 if isinstance(r, float) else r)
+            return sorted(map(repr, out))
+
+        dev = norm(build(Dampr.columns(vals)).run().read())
+        host = norm(build(Dampr.memory(vals.tolist())).run(
+            runner=MTRunner, n_maps=2, n_reducers=2).read())
+        if dev != host:
+            fails += 1
+            print("FLOAT MISMATCH", trial, op, n, dev[:3], host[:3])
+            if fails > 5:
+                break
+        if trial and trial % 100 == 0:
+            print("... float", trial, "trials, fails:", fails)
+    print("done float:", trials, "trials, fails:", fails)
+    return fails
+
+
 if __name__ == "__main__":
     t = int(sys.argv[1]) if len(sys.argv) > 1 else 500
-    sys.exit(1 if (main(t) + main_str(max(t // 2, 100))) else 0)
+    sys.exit(1 if (main(t) + main_str(max(t // 2, 100))
+                   + main_float(max(t // 2, 100))) else 0)
