@@ -680,9 +680,10 @@ class Dampr(object):
         fold_by with recognized binops, joins, first, ...) execute as
         gfx950 kernels over HBM-resident columns.
 
-        ``keys`` may also be a string array/list: keys dictionary-encode
-        at ingest (sorted table, rank ids) and run on the same kernels;
-        joins across different vocabularies remap through the union
+        ``keys`` may be int64, float64 (order-preserving IEEE encode at
+        ingest), or a string array/list — strings dictionary-encode
+        (sorted table, rank ids) and run on the same kernels; joins
+        across different vocabularies remap through the union
         dictionary on device.
 
         Numeric domain: typed columns use fixed-width i64/f64
