@@ -8,7 +8,11 @@ import numpy as np
 import pytest
 import torch
 
-pytestmark = pytest.mark.gpu
+pytestmark = [
+    pytest.mark.gpu,
+    pytest.mark.skipif("not __import__('torch').cuda.is_available()",
+                       reason="needs an MI355X"),
+]
 
 from dampr_amd import Dampr, funcs  # noqa: E402
 
