@@ -35,6 +35,19 @@ class Splitter(object):
     def partition(self, key, n_partitions):
         return hash(key) % n_partitions
 
+    @staticmethod
+    def check_start_method():
+        """The builtin-hash placement is only consistent across workers
+        because executor._ctx pins the *fork* start method (all workers
+        share the parent's hash salt).  Called at executor pool start so
+        a future spawn-based pool fails loudly here instead of silently
+        scattering equal keys across partitions."""
+        from .executor import _ctx
+        assert _ctx.get_start_method() == "fork", (
+            "Splitter's builtin-hash partitioning requires forked "
+            "workers (shared hash salt); use StableSplitter with a "
+            "spawn-based pool")
+
 
 class StableSplitter(object):
     def partition(self, key, n_partitions):

@@ -532,12 +532,14 @@ class ARReduce(object):
 
     def reduce(self, binop, reduce_buffer=None, **options):
         """Reduce each group with an associative binop.  ``reduce_buffer``
-        caps the map-side combine dictionary (keys held in memory before a
-        spill) — unlike the reference, it is honored when set (SURVEY.md
-        §2.5); the default defers to the RSS watermark like the
-        reference's actual behavior.  Recognized binops (operator.add,
-        min, max — dampr_amd.funcs) lower to the device segmented-reduce
-        kernel on the columnar engine."""
+        caps the map-side combine dictionary (distinct keys held in memory
+        before a spill) — unlike the reference, it is honored (SURVEY.md
+        §2.5); the default is ``settings.reduce_buffer``, a backstop
+        under the RSS watermark so high-cardinality keys cannot grow an
+        unbounded per-worker dict between amortized RSS checks.
+        Recognized binops (operator.add, min, max — dampr_amd.funcs)
+        lower to the device segmented-reduce kernel on the columnar
+        engine."""
         name = funcs.binop_name(binop)
         return self._run(binop, (name,) if name else None, reduce_buffer,
                          options)

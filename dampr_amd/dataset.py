@@ -513,8 +513,9 @@ class ContiguousWriter(DatasetWriter):
 
 
 class SinkWriter(DatasetWriter):
-    """Durable text part files; writes the VALUE only, one per line
-    (reference semantics: dataset.py:264-282)."""
+    """Durable text part files; writes the VALUE only, one per line,
+    str-formatting any value (reference semantics: dataset.py:264-282
+    ``print(value, file=f)``)."""
 
     def __init__(self, path, part_id):
         self.dir = path
@@ -526,8 +527,7 @@ class SinkWriter(DatasetWriter):
         self._fh = open(self.path, "w", buffering=1 << 20)
 
     def add_record(self, key, value):
-        self._fh.write(value)
-        self._fh.write("\n")
+        self._fh.write("{}\n".format(value))
 
     def finished(self):
         self._fh.close()

@@ -61,6 +61,8 @@ class WorkerCrash(RuntimeError):
 def run_stage(worker_factory, jobs, n_procs, stage_fs):
     """Run ``jobs`` across ``n_procs`` forked workers; returns the list of
     worker finish() results."""
+    from .base import Splitter
+    Splitter.check_start_method()   # fork-only invariant, fail loudly
     in_q = _ctx.Queue()
     out_q = _ctx.Queue()
     n_jobs = 0

@@ -258,6 +258,23 @@ class HostStore(list):
     fkeys = False
 
 
+class SinkStore(object):
+    """A sink stage's output: lazy view over the durable text part
+    files it wrote (host parity — _SinkWorker returns its
+    TextLineDatasets, so a sink's output is readable downstream without
+    re-materializing the sunk bytes in memory)."""
+
+    keyed = False
+    fkeys = False
+
+    def __init__(self, paths):
+        self.paths = paths
+
+    def datasets(self):
+        from ..dataset import TextLineDataset
+        return [TextLineDataset(p) for p in self.paths]
+
+
 def _decode_f64_sortable(enc):
     """Inverse of relational.encode_f64_sortable."""
     sign_bit = -(1 << 63)
@@ -384,12 +401,15 @@ class HbmPool(object):
         import uuid
         self.capacity = capacity_bytes
         self.used = 0
-        self._lru = []                 # device-resident, insertion order
+        # ordered sets (insertion-ordered dicts): O(1) add/remove/contains
+        # — a 2 TB job at 128 MB runs is ~16k live runs, list.remove per
+        # touch would be O(n) scans on every access
+        self._lru = {}                 # device-resident, insertion order
         self.host_capacity = (host_capacity
                               if host_capacity is not None
                               else float("inf"))
         self.host_used = 0
-        self._host_lru = []
+        self._host_lru = {}
         self.spill_dir = spill_dir or settings.spill_dir
         self._run_tag = "dampr_amd_{}".format(uuid.uuid4().hex[:10])
         self._file_ctr = 0
@@ -426,7 +446,7 @@ class HbmPool(object):
 
     def admit(self, run):
         self.used += run.nbytes
-        self._lru.append(run)
+        self._lru[run] = None
         self.balance()
 
     def prefetch(self, runs, device, stream):
@@ -441,7 +461,7 @@ class HbmPool(object):
                 if run.resident:
                     continue
                 if run in self._host_lru:
-                    self._host_lru.remove(run)
+                    del self._host_lru[run]
                     self.host_used -= run.nbytes
                 self.reloaded += run.nbytes
                 run.load(device)
@@ -451,7 +471,7 @@ class HbmPool(object):
                 run.keys.record_stream(main)
                 run.vals.record_stream(main)
                 self.used += run.nbytes
-                self._lru.append(run)
+                self._lru[run] = None
         # NOTE: no balance() here — eviction during an in-flight copy
         # could spill the very runs being loaded; the next touch() call
         # rebalances on the main stream.
@@ -459,20 +479,19 @@ class HbmPool(object):
     def touch(self, run, device):
         if not run.resident:
             if run in self._host_lru:
-                self._host_lru.remove(run)
+                del self._host_lru[run]
                 self.host_used -= run.nbytes
             self.reloaded += run.nbytes
             run.load(device)
             self.used += run.nbytes
-            self._lru.append(run)
+            self._lru[run] = None
             self.balance(exclude=run)
         return run
 
     def release(self, run):
         if run.resident:
             self.used -= run.nbytes
-        if run in self._lru:
-            self._lru.remove(run)
+        self._lru.pop(run, None)
 
     def balance(self, exclude=None):
         while self.used > self.capacity and self._lru:
@@ -483,14 +502,15 @@ class HbmPool(object):
                     break
             if victim is None:
                 return
-            self._lru.remove(victim)
+            del self._lru[victim]
             self.used -= victim.nbytes
             victim.spill()
             self.spilled_host += victim.nbytes
             self.host_used += victim.nbytes
-            self._host_lru.append(victim)
+            self._host_lru[victim] = None
         while self.host_used > self.host_capacity and self._host_lru:
-            v = self._host_lru.pop(0)
+            v = next(iter(self._host_lru))
+            del self._host_lru[v]
             self.host_used -= v.nbytes
             self.spilled_disk += v.nbytes
             v.spill_to_disk(self._next_path())
@@ -672,7 +692,7 @@ class GpuRunner(RunnerBase):
                     continue
                 self.pool.release(run)
                 if run in self.pool._host_lru:
-                    self.pool._host_lru.remove(run)
+                    del self.pool._host_lru[run]
                     self.pool.host_used -= run.nbytes
                 if run.on_disk:
                     try:
@@ -915,6 +935,9 @@ class GpuRunner(RunnerBase):
         merge)."""
         if isinstance(store, ColumnDataset):
             return store
+        if isinstance(store, SinkStore):
+            from ..dataset import cat_datasets
+            return cat_datasets(store.datasets())
         if isinstance(store, HostStore):
             from ..dataset import MemoryDataset
             return MemoryDataset(sorted(store, key=lambda r: r[0]))
@@ -1339,6 +1362,41 @@ class GpuRunner(RunnerBase):
             out.append(ns)
         return out
 
+    def _unify_fkeys_stores(self, stores):
+        """Mixed float-keyed and int-keyed inputs: the f64 side's keys are
+        order-preserving encodings (encode_f64_sortable) while the int
+        side's are raw i64 bit patterns, so comparisons AND hash routing
+        would silently miss (1 != encode(1.0)).  Re-encode the int side
+        through float64 (host semantics: 1 == 1.0).  Ints beyond 2^53
+        lose precision here — the same contract as the mixed int+float
+        record encode (_encode_records_world); such pipelines belong on
+        the host engine.  Re-encoded stores come back unpartitioned
+        (routing changed) and re-route in _ensure_partitioned."""
+        flags = [bool(getattr(s, "fkeys", False)) for s in stores]
+        if all(flags) or not any(flags):
+            return stores
+        out = []
+        for s, f in zip(stores, flags):
+            if f or not isinstance(s, PartStore):
+                out.append(s)
+                continue
+            ns = PartStore(keyed=getattr(s, "keyed", False), fkeys=True,
+                           partitioned=False,
+                           str_table=getattr(s, "str_table", None))
+            ns[0] = []
+            for p in sorted(s):
+                for run in s[p]:
+                    self.pool.touch(run, self.device)
+                    # monotone in signed i64 order, so sorted runs stay
+                    # sorted
+                    nk = _encode_f64_sortable(run.keys.to(torch.float64))
+                    nr = DeviceRun(nk, run.vals.clone(), sorted=run.sorted)
+                    self.pool.release(run)
+                    ns[0].append(nr)
+                    self.pool.admit(nr)
+            out.append(ns)
+        return out
+
     def _merge_stores(self, stores):
         if len(stores) == 1:
             return stores[0]
@@ -1361,6 +1419,8 @@ class GpuRunner(RunnerBase):
                     out.extend(self._decode_store(s))
                 return out
             stores = uni
+        if all(isinstance(s, PartStore) for s in stores):
+            stores = self._unify_fkeys_stores(stores)
         flags = [getattr(s, "partitioned", True) for s in stores]
         if not all(flags) and any(flags):
             # mixing hashed and unrouted partition-0 runs would corrupt
@@ -1392,6 +1452,8 @@ class GpuRunner(RunnerBase):
                 return self._host_reduce(stage, ins)
             ins = uni
             uni_table = ins[0].str_table
+        if len(ins) > 1 and all(isinstance(s, PartStore) for s in ins):
+            ins = self._unify_fkeys_stores(ins)
         ins = [self._ensure_partitioned(s) for s in ins]
         if len(ins) == 1 and isinstance(ins[0], TokenStore) \
                 and spec == ("sum",):
@@ -1622,9 +1684,12 @@ class GpuRunner(RunnerBase):
 
     def run_sink(self, stage, ins):
         """Sink semantics match the host SinkWriter: one text line per
-        record, value only (keys dropped — reference: dataset.py:277-278).
-        Untagged sinks (pending format maps, e.g. sink_tsv's) run the
-        mapper on host over the decoded records."""
+        record, value only, any value str-formatted (reference:
+        dataset.py:277-278 prints the value).  Untagged sinks (pending
+        format maps, e.g. sink_tsv's) run the mapper on host over the
+        decoded records.  Returns a lazy view over the written part file
+        so the sink's output is readable downstream, matching
+        _SinkWorker (runner.py:278-298)."""
         import os
         path = stage.path
         os.makedirs(path, exist_ok=True)
@@ -1640,13 +1705,15 @@ class GpuRunner(RunnerBase):
             with open(fname, "w") as fh:
                 for _kk, vv in stage.mapper.map(mem):
                     fh.write("{}\n".format(vv))
-        return PartStore()
+        return SinkStore([fname])
 
     # -- host fallback -----------------------------------------------------
 
     def _decode_store(self, store):
         if isinstance(store, HostStore):
             return list(store)
+        if isinstance(store, SinkStore):
+            return [kv for ds in store.datasets() for kv in ds.read()]
         tbl = getattr(store, "str_table", None)
         if isinstance(store, TextSource):
             records = []

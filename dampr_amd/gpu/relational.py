@@ -117,7 +117,10 @@ def hash_join(keys_l, keys_r, how="inner"):
     nr = keys_r.numel()
     cap = _pow2_at_least(max(2 * nr, 16))
     t_keys = torch.zeros(cap, dtype=torch.int64, device=dev)
-    t_head = torch.full((cap,), -1, dtype=torch.int64, device=dev)
+    # cap+1 heads: t_head[cap] is the dedicated zero-key chain (raw keys
+    # include 0 — dictionary rank ids, int columns — and the open table
+    # uses 0 as EMPTY; see hj_build_kernel)
+    t_head = torch.full((cap + 1,), -1, dtype=torch.int64, device=dev)
     nxt = torch.empty(max(nr, 1), dtype=torch.int64, device=dev)
     ext.hj_build(keys_r, t_keys, t_head, nxt)
     left_outer = 1 if how in ("left", "outer") else 0

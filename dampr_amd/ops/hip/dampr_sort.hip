@@ -276,6 +276,12 @@ void seg_reduce(torch::Tensor seg, torch::Tensor vals, torch::Tensor out,
 // Build: chain right-side rows per key (slot -> head row, next[] links).
 // Probe: count matches per left row, then emit (l, r) index pairs at
 // exclusive offsets.  Row indices allow the Python layer to gather values.
+//
+// Keys are RAW u64 (dictionary rank ids start at 0; int columns may
+// contain 0), so key 0 must be representable even though the open table
+// uses 0 as its EMPTY sentinel: slot mask+1 of t_head is a dedicated
+// zero-key chain and key 0 never enters t_keys (t_head has mask+2
+// entries; the Python layer allocates cap+1).
 
 __global__ void hj_build_kernel(const u64* __restrict__ keys_r, long nr,
                                 u64* __restrict__ t_keys,
@@ -285,9 +291,11 @@ __global__ void hj_build_kernel(const u64* __restrict__ keys_r, long nr,
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nr;
          i += stride) {
         u64 k = keys_r[i];
-        if (!k) k = 1;
         u64 slot;
-        table_insert_u64(t_keys, mask, k, &slot);
+        if (k == 0ULL)
+            slot = mask + 1;              // dedicated zero-key chain
+        else
+            table_insert_u64(t_keys, mask, k, &slot);
         long old = atomicExch((unsigned long long*)&t_head[slot],
                               (unsigned long long)i);
         next[i] = old;
@@ -297,6 +305,7 @@ __global__ void hj_build_kernel(const u64* __restrict__ keys_r, long nr,
 __device__ __forceinline__ long hj_find(const u64* t_keys,
                                         const long* t_head, u64 mask,
                                         u64 k) {
+    if (k == 0ULL) return t_head[mask + 1];
     u64 slot = k & mask;
     while (true) {
         u64 cur = t_keys[slot];
@@ -316,7 +325,6 @@ __global__ void hj_count_kernel(const u64* __restrict__ keys_l, long nl,
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nl;
          i += stride) {
         u64 k = keys_l[i];
-        if (!k) k = 1;
         long c = 0;
         for (long r = hj_find(t_keys, t_head, mask, k); r >= 0;
              r = next[r])
@@ -338,7 +346,6 @@ __global__ void hj_emit_kernel(const u64* __restrict__ keys_l, long nl,
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nl;
          i += stride) {
         u64 k = keys_l[i];
-        if (!k) k = 1;
         long o = offsets[i];
         long c = 0;
         for (long r = hj_find(t_keys, t_head, mask, k); r >= 0;

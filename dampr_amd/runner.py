@@ -426,7 +426,11 @@ class MTRunner(RunnerBase):
             factory = lambda: _MapWorker(stage.mapper, n_partitions, memory)
         else:
             binop = opts.get("binop")
+            # per-stage reduce_buffer, else the settings backstop (bounds
+            # the combine dict between amortized RSS checks)
             max_keys = opts.get("reduce_buffer")
+            if max_keys is None:
+                max_keys = settings.reduce_buffer
             factory = lambda: _CombineMapWorker(
                 stage.mapper, stage.combiner, n_partitions, memory, binop,
                 max_keys)

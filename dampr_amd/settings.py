@@ -38,6 +38,14 @@ max_memory_per_worker = 512
 memory_min_count = 10000
 memory_max_count_before_check = 100000
 
+# Default cap on the map-side combine dictionary (distinct keys held
+# before a forced spill).  A backstop under the RSS watermark: without it
+# a high-cardinality a_group_by grows an unbounded per-worker dict
+# between (amortized) RSS checks.  ARReduce.reduce's ``reduce_buffer``
+# kwarg overrides per-stage (the reference documents that kwarg but never
+# reads it — SURVEY.md §2.5).
+reduce_buffer = 1 << 21
+
 # ---------------------------------------------------------------- GPU engine
 # New knobs for the MI355X path; no reference analog.
 

@@ -7,3 +7,19 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: needs a real MI355X (run via gpurun)")
+
+
+# GPU suites run kernels-first: under ``-x`` an engine-level failure must
+# not mask the per-kernel oracle tests (round 1 the driver's -x run
+# stopped in test_gpu_engine.py and the kernel suites never executed).
+_GPU_ORDER = ["test_gpu_kernels", "test_gpu_relational", "test_gpu_engine"]
+
+
+def pytest_collection_modifyitems(config, items):
+    def rank(item):
+        mod = item.module.__name__.rsplit(".", 1)[-1]
+        try:
+            return (0, _GPU_ORDER.index(mod))
+        except ValueError:
+            return (1, 0)
+    items.sort(key=rank)

@@ -1172,3 +1172,78 @@ def test_device_engine_misc_inputs(tmp_path):
     got2 = sorted(Dampr.read_input(RangeDataset(5), RangeDataset(3))
                   .count().run(runner=GpuRunner).read())
     assert got2 == [(0, 2), (1, 2), (2, 2), (3, 1), (4, 1)]
+
+
+def test_sink_output_readable_downstream(tmp_path):
+    """A sink's output dataset is consumable downstream (host parity:
+    _SinkWorker returns its TextLineDatasets); round 1 the device sink
+    returned an empty store and silently dropped the data."""
+    path = str(tmp_path / "s")
+    em = Dampr.memory([3, 1, 2]).map(lambda v: v * 10).sink(path) \
+        .run(runner=GpuRunner)
+    got = sorted(em.read())
+    assert got == ["10", "20", "30"]          # text lines, like the host
+
+
+def test_sink_numeric_values_both_engines(tmp_path):
+    """Reference SinkWriter prints ANY value (print(value, file=f));
+    both engines must accept non-str values identically."""
+    hp = str(tmp_path / "h")
+    dp = str(tmp_path / "d")
+    Dampr.memory([3, 1, 2]).sink(hp).run()
+    Dampr.columns(np.array([3, 1, 2], dtype=np.int64)) \
+        .checkpoint(True).sink(dp).run(runner=GpuRunner)
+
+    def lines(d):
+        out = []
+        for f in sorted(os.listdir(d)):
+            with open(os.path.join(d, f)) as fh:
+                out.extend(ln.strip() for ln in fh if ln.strip())
+        return sorted(out)
+
+    assert lines(hp) == ["1", "2", "3"]
+    assert lines(dp) == ["1", "2", "3"]
+
+
+def test_join_mixed_float_int_keys_device():
+    """Float-keyed store joined with an int-keyed store: host semantics
+    say 1.0 == 1, so the int side re-encodes through the f64 sortable
+    encoding (round 1 this silently returned [])."""
+    lv = np.array([10, 20], dtype=np.int64)
+    rv = np.array([1, 2], dtype=np.int64)
+    out = Dampr.columns(lv, keys=np.array([1.0, 2.0])) \
+        .join(Dampr.columns(rv, keys=np.array([1, 2], dtype=np.int64))) \
+        .reduce(funcs.pair_sum, many=True).run(runner=GpuRunner)
+    assert sorted(out.read()) == [(1.0, 11), (2.0, 22)]
+    # and host-engine agreement
+    host = Dampr.memory([(1.0, 10), (2.0, 20)]) \
+        .group_by(lambda kv: kv[0], lambda kv: kv[1]) \
+        .join(Dampr.memory([(1, 1), (2, 2)])
+              .group_by(lambda kv: kv[0], lambda kv: kv[1])) \
+        .reduce(lambda l, r: [a + b for a in list(l) for b in list(r)],
+                many=True).run()
+    assert sorted(host.read()) == [(1.0, 11), (2.0, 22)] \
+        or sorted(host.read()) == [(1, 11), (2, 22)]
+
+
+def test_join_mixed_int_float_keys_left():
+    """Same fkeys-unification path, left join direction flipped (the
+    INT side is the left/probe side)."""
+    lv = np.array([10, 20, 30], dtype=np.int64)
+    rv = np.array([1, 2], dtype=np.int64)
+    out = Dampr.columns(lv, keys=np.array([1, 2, 7], dtype=np.int64)) \
+        .join(Dampr.columns(rv, keys=np.array([1.0, 2.0]))) \
+        .reduce(funcs.pair_sum, many=True).run(runner=GpuRunner)
+    assert sorted(out.read()) == [(1.0, 11), (2.0, 22)]
+
+
+def test_hash_join_zero_key_torchops():
+    """Engine-level zero/one key join separation (TorchOps here; the
+    HIP-kernel regression is tests/test_gpu_relational.py)."""
+    lk = np.array([0, 1, 0], dtype=np.int64)
+    lv = np.array([1, 2, 3], dtype=np.int64)
+    rk = np.array([0, 1], dtype=np.int64)
+    rv = np.array([10, 20], dtype=np.int64)
+    out = Dampr.columns(lv, keys=lk).join(Dampr.columns(rv, keys=rk)) \
+        .reduce(funcs.pair_sum, many=True).run(runner=GpuRunner)
+    assert sorted(out.read()) == [(0, 11), (0, 13), (1, 22)]

@@ -141,6 +141,48 @@ def test_hash_join(rng, how):
     assert got == _join_oracle(lk_np.tolist(), rk_np.tolist(), how)
 
 
+@pytest.mark.parametrize("how", ["inner", "left", "outer"])
+def test_hash_join_zero_and_one_keys(how):
+    """Regression (round-1 driver failure): the open table uses key 0 as
+    its EMPTY sentinel and the old remap ``if (!k) k = 1`` aliased key 0
+    with key 1 — dictionary rank ids start at 0, so "apple"(0) joined
+    "fig"(1)'s rows.  Keys 0 and 1 must join independently."""
+    from dampr_amd.gpu.relational import hash_join
+    lk_np = np.array([0, 1, 0, 2, 1, 0], dtype=np.int64)
+    rk_np = np.array([1, 0, 3, 0], dtype=np.int64)
+    li, ri = hash_join(torch.from_numpy(lk_np).to(DEV),
+                       torch.from_numpy(rk_np).to(DEV), how=how)
+    got = sorted(zip(li.cpu().numpy().tolist(),
+                     ri.cpu().numpy().tolist()))
+    assert got == _join_oracle(lk_np.tolist(), rk_np.tolist(), how)
+
+
+@pytest.mark.parametrize("how", ["inner", "left", "outer"])
+def test_hash_join_i64_extremes(how):
+    from dampr_amd.gpu.relational import hash_join
+    ext = [0, 1, -1, (1 << 63) - 1, -(1 << 63), 2, -2]
+    lk_np = np.array(ext + [0, (1 << 63) - 1], dtype=np.int64)
+    rk_np = np.array([-(1 << 63), 0, 5, (1 << 63) - 1, -1],
+                     dtype=np.int64)
+    li, ri = hash_join(torch.from_numpy(lk_np).to(DEV),
+                       torch.from_numpy(rk_np).to(DEV), how=how)
+    got = sorted(zip(li.cpu().numpy().tolist(),
+                     ri.cpu().numpy().tolist()))
+    assert got == _join_oracle(lk_np.tolist(), rk_np.tolist(), how)
+
+
+def test_hash_join_zero_key_heavy(rng):
+    """Many zero keys (dedicated chain must carry real multiplicity)."""
+    from dampr_amd.gpu.relational import hash_join
+    lk_np = rng.integers(0, 4, size=5000, dtype=np.int64)
+    rk_np = rng.integers(0, 4, size=300, dtype=np.int64)
+    li, ri = hash_join(torch.from_numpy(lk_np).to(DEV),
+                       torch.from_numpy(rk_np).to(DEV), how="inner")
+    got = sorted(zip(li.cpu().numpy().tolist(),
+                     ri.cpu().numpy().tolist()))
+    assert got == _join_oracle(lk_np.tolist(), rk_np.tolist(), "inner")
+
+
 def test_topk(rng):
     from dampr_amd.gpu.relational import topk_by, encode_f64_sortable
     vals_np = rng.standard_normal(100_000)
