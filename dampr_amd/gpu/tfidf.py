@@ -216,10 +216,15 @@ class TfidfEngine(object):
 
 def run_tfidf(text_np, device="cuda:0", sink_path=None, chunk_bytes=None,
               engine=None):
-    """One full single-GPU TF-IDF job over a host corpus; returns
-    {token: (df, idf)} (also sinks TSV when sink_path given)."""
+    """One full single-GPU TF-IDF job over a host (numpy u8) or
+    device-resident (torch u8) corpus; returns {token: (df, idf)}
+    (also sinks TSV when sink_path given)."""
     dev = torch.device(device)
-    text = torch.from_numpy(text_np).to(dev)
+    if isinstance(text_np, torch.Tensor):
+        text = text_np.to(dev)
+        text_np = text.cpu().numpy()
+    else:
+        text = torch.from_numpy(text_np).to(dev)
     eng = engine or TfidfEngine(dev)
     eng.reset()
     n = text.numel()

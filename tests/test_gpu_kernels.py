@@ -182,3 +182,24 @@ def test_tfidf_window_edge_tokens():
     arr = np.frombuffer(text, dtype=np.uint8).copy()
     got = {t: v[0] for t, v in run_tfidf(arr, device=DEV).items()}
     assert got == oracle_df(arr)
+
+
+def test_synth_corpus_device_layout_and_df():
+    """Device corpus generator: same layout contract as the CPU one
+    (uniform 96-byte lines, [a-z ]/newline bytes) and the fused df
+    pipeline agrees with the CPU oracle on its output."""
+    from dampr_amd.gpu.corpus import oracle_df, synth_corpus_device
+    from dampr_amd.gpu.tfidf import run_tfidf
+    text = synth_corpus_device(1 << 20, DEV, vocab=5000, seed=11)
+    t = text.cpu().numpy()
+    assert t.size % 96 == 0
+    lines = t.reshape(-1, 96)
+    assert (lines[:, -1] == ord("\n")).all()
+    body = lines[:, :-1].reshape(-1)
+    ok = ((body >= ord("a")) & (body <= ord("z"))) | (body == ord(" "))
+    assert ok.all()
+    got = run_tfidf(text, device=str(DEV))
+    want = oracle_df(t)
+    assert len(got) == len(want)
+    for tok, df in list(want.items())[:500]:
+        assert got[tok][0] == df, (tok, got[tok], df)
