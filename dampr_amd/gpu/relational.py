@@ -49,8 +49,9 @@ def radix_sort_pairs(keys, payload=None):
         shift = byte * 8
         if not active[byte]:
             continue                      # constant digit: skip pass
-        hist = ext.rs_hist(src_k, shift, nblocks).to(torch.int64)
-        scanned = torch.cumsum(hist, 0) - hist
+        hist = ext.rs_hist(src_k, shift, nblocks)
+        scanned = torch.cumsum(hist, 0, dtype=torch.int64)
+        scanned -= hist                # exclusive, fused i64 upcast
         ext.rs_scatter(src_k, src_p, scanned, shift, nblocks, out_k, out_p)
         n_done += 1
         if n_done == 1:
