@@ -39,11 +39,13 @@ def radix_sort_pairs(keys, payload=None):
     # two owned buffers (using the input as scratch would clobber it).
     src_k, src_p = keys.contiguous(), payload.contiguous()
     out_k, out_p = torch.empty_like(src_k), torch.empty_like(src_p)
-    # one read decides which byte passes are constant (skippable): a
-    # single host sync instead of one per pass, and no per-pass
-    # exploratory histograms
-    hist8 = ext.rs_hist_global(src_k).cpu()
-    active = [int((hist8[b] != 0).sum().item()) > 1 for b in range(8)]
+    # one streaming read decides which byte passes are constant
+    # (skippable): digit d is constant iff its byte of the AND-fold
+    # equals its byte of the OR-fold.  Single host sync for the whole
+    # sort; runs at read bandwidth (register fold + one atomic/wave).
+    a, o = ext.rs_digit_fold(src_k).cpu().tolist()
+    active = [((a >> (8 * b)) & 255) != ((o >> (8 * b)) & 255)
+              for b in range(8)]
     n_done = 0
     for byte in range(8):
         shift = byte * 8

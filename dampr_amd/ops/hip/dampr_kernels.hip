@@ -899,7 +899,7 @@ torch::Tensor tsv_format(torch::Tensor blob, torch::Tensor tok_off,
 
 // Implemented in dampr_sort.hip
 torch::Tensor rs_hist(torch::Tensor keys, long shift, long nblocks);
-torch::Tensor rs_hist_global(torch::Tensor keys);
+torch::Tensor rs_digit_fold(torch::Tensor keys);
 void rs_scatter(torch::Tensor keys, torch::Tensor payload,
                 torch::Tensor scanned, long shift, long nblocks,
                 torch::Tensor out_k, torch::Tensor out_p);
@@ -920,8 +920,8 @@ std::vector<torch::Tensor> hj_emit(torch::Tensor keys_l,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rs_hist", &rs_hist, "radix pass histogram (bin-major)");
-    m.def("rs_hist_global", &rs_hist_global,
-          "all 8 byte histograms in one pass (skip detection)");
+    m.def("rs_digit_fold", &rs_digit_fold,
+          "AND/OR fold over keys (constant-digit skip detection)");
     m.def("rs_scatter", &rs_scatter, "stable radix scatter pass");
     m.def("seg_reduce", &seg_reduce,
           "segmented reduce over sorted runs (op 0=sum,1=min,2=max)");

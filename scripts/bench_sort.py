@@ -38,6 +38,23 @@ def bench_case(name, keys, iters=5, warmup=2):
                       "gb_per_s_per_pass_if_8": n * 24 / dt / 8 / 1e9}))
 
 
+def bench_torch_sort(name, keys, iters=5, warmup=2):
+    """Calibration baseline: torch.sort = rocPRIM device radix sort."""
+    times = []
+    for i in range(warmup + iters):
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        sk, sp = torch.sort(keys, stable=True)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        if i >= warmup:
+            times.append(dt)
+        del sk, sp
+    dt = min(times)
+    print(json.dumps({"case": name + "-torchsort", "rows": keys.numel(),
+                      "ms": dt * 1e3, "rows_per_s": keys.numel() / dt}))
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--rows", type=int, default=200_000_000)
@@ -50,12 +67,14 @@ def main():
     k64 = torch.from_numpy(
         rng.integers(0, 1 << 63, size=n, dtype=np.int64)).to(dev)
     bench_case("u64-uniform", k64)
+    bench_torch_sort("u64-uniform", k64)
     del k64
 
     # 20-bit keys (dict ranks / group ids; 3 active passes)
     k20 = torch.from_numpy(
         rng.integers(0, 1 << 20, size=n, dtype=np.int64)).to(dev)
     bench_case("u20-groupids", k20)
+    bench_torch_sort("u20-groupids", k20)
     del k20
 
     # Zipf-skewed small cardinality (1M groups)
