@@ -688,6 +688,13 @@ class Dampr(object):
         across different vocabularies remap through the union
         dictionary on device.
 
+        ``vals`` may also be a string array/list: var-len values ride a
+        device byte arena (blob + offsets) through partition, sort,
+        spill, exchange and join — only keys compare on device
+        (SURVEY.md §7).  Gather-only ops (first, sort_by, join with
+        pair_left/pair_right) stay on the kernels; arithmetic folds
+        over string values fall back to host records per stage.
+
         Numeric domain: typed columns use fixed-width i64/f64
         arithmetic — integer aggregates wrap at 64 bits, where the host
         engine's Python ints are arbitrary precision.  Pipelines whose

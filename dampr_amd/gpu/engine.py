@@ -75,7 +75,8 @@ class ColumnSource(object):
 
     def __init__(self, keys, vals, str_table=None, fkeys=False):
         assert keys.dtype == torch.int64
-        assert vals.dtype in _VAL_DTYPES
+        assert getattr(vals, "is_strvals", False) \
+            or vals.dtype in _VAL_DTYPES
         assert keys.numel() == vals.numel()
         self.keys = keys
         self.vals = vals
@@ -97,10 +98,16 @@ class ColumnSource(object):
     def from_data(cls, vals, keys=None, device=None):
         import numpy as np
         if cls._is_str_array(vals):
-            raise TypeError(
-                "string VALUE columns are not supported on the device "
-                "engine (string keys are); use the host engine")
-        v = _as_column(vals, device)
+            # var-len value arena: strings move as opaque bytes
+            # addressed by row; only keys compare on device
+            from .strvals import StrVals
+            arr = np.asarray(vals)
+            if arr.dtype.kind == "S":
+                arr = arr.astype("U")
+            v = StrVals.from_strings([str(s) for s in arr.flat],
+                                     device=device)
+        else:
+            v = _as_column(vals, device)
         str_table = None
         fkeys = False
         if keys is None:
@@ -231,10 +238,15 @@ class PartStore(dict):
     """
 
     def __init__(self, keyed=False, fkeys=False, partitioned=True,
-                 str_table=None):
+                 str_table=None, svals=False):
         super(PartStore, self).__init__()
         self.keyed = keyed
         self.fkeys = fkeys
+        # value columns are var-len byte arenas (StrVals).  Kept as a
+        # STORE-level flag (not just per-run inspection) so an empty
+        # rank's store still reports it: stage-dispatch guards must
+        # agree across ranks or collective sequences desync.
+        self.svals = svals
         # dictionary-encoded string keys: key column holds ranks into
         # this sorted tuple (rank order == lexicographic order, so sorted
         # device output decodes to host-ordered strings).  Cross-store
@@ -273,6 +285,39 @@ class SinkStore(object):
     def datasets(self):
         from ..dataset import TextLineDataset
         return [TextLineDataset(p) for p in self.paths]
+
+
+def _is_sv(x):
+    """Is this value column a var-len byte arena (gpu.strvals.StrVals)?"""
+    return getattr(x, "is_strvals", False)
+
+
+def _cat_vals(vs):
+    """Concatenate value columns (tensor or StrVals, never mixed)."""
+    if len(vs) == 1:
+        return vs[0]
+    if any(_is_sv(v) for v in vs):
+        from .strvals import StrVals
+        assert all(_is_sv(v) for v in vs), \
+            "mixed var-len and numeric value runs in one partition"
+        return StrVals.cat(vs)
+    return torch.cat(vs)
+
+
+def _run_has_sv(run):
+    """Does this run carry var-len values (any tier)?"""
+    if run.keys is not None:
+        return _is_sv(run.vals)
+    if run._host is not None:
+        return _is_sv(run._host[1])
+    return run._meta is not None and isinstance(run._meta[1], tuple)
+
+
+def _store_has_sv(store):
+    if not isinstance(store, PartStore):
+        return False
+    return getattr(store, "svals", False) or any(
+        _run_has_sv(r) for runs in store.values() for r in runs)
 
 
 def _decode_f64_sortable(enc):
@@ -318,10 +363,14 @@ class DeviceRun(object):
     @property
     def nbytes(self):
         if self.keys is not None:
+            if _is_sv(self.vals):
+                return self.keys.numel() * 8 + self.vals.nbytes
             return self.keys.numel() * 8 + self.vals.element_size() * \
                 self.vals.numel()
         if self._meta is not None:
             n, vdt = self._meta
+            if isinstance(vdt, tuple):          # ("str", blob_bytes)
+                return n * 8 + vdt[1] + (n + 1) * 8
             return n * 8 + n * (8 if vdt in (torch.int64, torch.float64)
                                 else 8)
         return 0
@@ -340,10 +389,23 @@ class DeviceRun(object):
             return
         pin = self.keys.device.type == "cuda"
         hk = torch.empty_like(self.keys, device="cpu", pin_memory=pin)
-        hv = torch.empty_like(self.vals, device="cpu", pin_memory=pin)
         hk.copy_(self.keys)
-        hv.copy_(self.vals)
-        self._meta = (self.keys.numel(), self.vals.dtype)
+        if _is_sv(self.vals):
+            from .strvals import StrVals
+            hb = torch.empty_like(self.vals.blob, device="cpu",
+                                  pin_memory=pin)
+            ho = torch.empty_like(self.vals.offs, device="cpu",
+                                  pin_memory=pin)
+            hb.copy_(self.vals.blob)
+            ho.copy_(self.vals.offs)
+            hv = StrVals(hb, ho)
+            self._meta = (self.keys.numel(),
+                          ("str", self.vals.blob.numel()))
+        else:
+            hv = torch.empty_like(self.vals, device="cpu",
+                                  pin_memory=pin)
+            hv.copy_(self.vals)
+            self._meta = (self.keys.numel(), self.vals.dtype)
         self._host = (hk, hv)
         self.keys = None
         self.vals = None
@@ -355,7 +417,11 @@ class DeviceRun(object):
         hk, hv = self._host
         with open(path, "wb") as fh:
             fh.write(hk.numpy().tobytes())
-            fh.write(hv.numpy().tobytes())
+            if _is_sv(hv):
+                fh.write(hv.offs.numpy().tobytes())
+                fh.write(hv.blob.numpy().tobytes())
+            else:
+                fh.write(hv.numpy().tobytes())
         self._disk = path
         self._host = None
 
@@ -368,9 +434,18 @@ class DeviceRun(object):
                 raw_v = fh.read()
             hk = torch.from_numpy(
                 np.frombuffer(raw_k, dtype=np.int64).copy())
-            vnp = (np.float64 if vdt == torch.float64 else np.int64)
-            hv = torch.from_numpy(
-                np.frombuffer(raw_v, dtype=vnp).copy())
+            if isinstance(vdt, tuple):
+                from .strvals import StrVals
+                off_b = (n + 1) * 8
+                ho = torch.from_numpy(
+                    np.frombuffer(raw_v[:off_b], dtype=np.int64).copy())
+                hb = torch.from_numpy(
+                    np.frombuffer(raw_v[off_b:], dtype=np.uint8).copy())
+                hv = StrVals(hb, ho)
+            else:
+                vnp = (np.float64 if vdt == torch.float64 else np.int64)
+                hv = torch.from_numpy(
+                    np.frombuffer(raw_v, dtype=vnp).copy())
             os_mod = __import__("os")
             try:
                 os_mod.unlink(self._disk)
@@ -755,7 +830,8 @@ class GpuRunner(RunnerBase):
             # trip) when the first stage re-keys anyway
             store = PartStore(partitioned=False,
                               str_table=inp.str_table,
-                              fkeys=inp.fkeys)
+                              fkeys=inp.fkeys,
+                              svals=_is_sv(inp.vals))
             store[0] = []
             n = keys.numel()
             step = max(1, settings.gpu_batch_records)
@@ -809,7 +885,7 @@ class GpuRunner(RunnerBase):
         # host keyed-reducer convention: value = (key, scalar)
         keyed = all(
             isinstance(v, tuple) and len(v) == 2 and v[0] == k
-            and _num(v[1])
+            and (_num(v[1]) or isinstance(v[1], str))
             for (k, _), v in zip(records, vs))
         if keyed:
             vs = [v[1] for v in vs]
@@ -817,10 +893,13 @@ class GpuRunner(RunnerBase):
             vt = torch.from_numpy(np.asarray(vs, dtype=np.int64))
         elif all(_num(v) for v in vs):
             vt = torch.from_numpy(np.asarray(vs, dtype=np.float64))
+        elif all(isinstance(v, str) for v in vs):
+            from .strvals import StrVals
+            vt = StrVals.from_strings(vs)
         else:
             raise TypeError(
-                "device engine requires numeric values; use the host "
-                "engine for object records")
+                "device engine requires numeric or string values; use "
+                "the host engine for object records")
         fkeys = False
         str_table = None
         if all(_int(k) for k in ks):
@@ -891,22 +970,27 @@ class GpuRunner(RunnerBase):
             return "obj", None
 
         kk, table = kind_of(ks, True)
-        vk, _ = kind_of(pv, False)
+        vk, _ = kind_of(pv, True)
         gathered = [None] * self.world
         dist.all_gather_object(gathered, (kk, vk, keyed, table))
         kks = {m[0] for m in gathered} - {"empty"}
         vks = {m[1] for m in gathered} - {"empty"}
         keyeds = {m[2] for m in gathered} - {None}
         if ("obj" in kks or "obj" in vks or len(keyeds) > 1
-                or ("str" in kks and len(kks) > 1)):
+                or ("str" in kks and len(kks) > 1)
+                or ("str" in vks and len(vks) > 1)):
             # no common columnar layout: every rank falls back to host
             # records (symmetric — no further collectives here; host
             # reduces exchange via _host_exchange)
             return HostStore(records)
         g_keyed = keyeds.pop() if keyeds else False
         vals = pv if g_keyed else vs
-        v_np = np.int64 if vks in ({"int"}, set()) else np.float64
-        vt = torch.from_numpy(np.asarray(vals, dtype=v_np))
+        if vks == {"str"}:
+            from .strvals import StrVals
+            vt = StrVals.from_strings(vals)
+        else:
+            v_np = np.int64 if vks in ({"int"}, set()) else np.float64
+            vt = torch.from_numpy(np.asarray(vals, dtype=v_np))
         fkeys = False
         str_table = None
         if kks == {"str"}:
@@ -958,7 +1042,7 @@ class GpuRunner(RunnerBase):
             z = torch.zeros(0, dtype=torch.int64)
             return ColumnDataset(z, z.clone(), keyed, fkeys, tbl)
         keys = torch.cat(ks)
-        vals = torch.cat(vs)
+        vals = _cat_vals(vs)
         sk, sp = self._sort(keys, fkeys=fkeys)
         return ColumnDataset(sk, vals[sp.to(torch.int64)], keyed, fkeys,
                              tbl)
@@ -984,7 +1068,7 @@ class GpuRunner(RunnerBase):
         """Split columns into the partition store {p: [DeviceRun]} (K1+K2):
         partition ids, stable sort by id, slice contiguous segments."""
         P = self.n_partitions
-        store = PartStore(keyed=keyed, fkeys=fkeys)
+        store = PartStore(keyed=keyed, fkeys=fkeys, svals=_is_sv(vals))
         if keys.numel() == 0 and self.world == 1:
             return store
         if P == 1 and self.world == 1:
@@ -1048,7 +1132,7 @@ class GpuRunner(RunnerBase):
                 self.pool.release(run)
             k = torch.cat(ks) if ks else torch.zeros(
                 0, dtype=torch.int64, device=self.device)
-            v = torch.cat(vs) if vs else torch.zeros(
+            v = _cat_vals(vs) if vs else torch.zeros(
                 0, dtype=torch.int64, device=self.device)
             out = self._partition(k, v, keyed=store.keyed,
                                   fkeys=store.fkeys)
@@ -1067,15 +1151,19 @@ class GpuRunner(RunnerBase):
                 for q, runs in part.items():
                     out.setdefault(q, []).extend(runs)
         if out is None:
-            out = PartStore(keyed=store.keyed, fkeys=store.fkeys)
+            out = PartStore(keyed=store.keyed, fkeys=store.fkeys,
+                            svals=getattr(store, "svals", False))
         out.str_table = getattr(store, "str_table", None)
         return out
 
     def _exchange(self, keys, vals, pid):
         """RCCL all-to-all: route rows to the partition's owning rank
         (p % world); returns this rank's rows."""
-        from ..parallel.shuffle import exchange_columns
+        from ..parallel.shuffle import (exchange_columns,
+                                        exchange_columns_varlen)
         self.exchanged_rows += keys.numel()
+        if _is_sv(vals):
+            return exchange_columns_varlen(keys, vals, pid, self.world)
         return exchange_columns(keys, vals, pid, self.world)
 
     def _merged_partition(self, stores, p, need_sorted=True):
@@ -1097,7 +1185,7 @@ class GpuRunner(RunnerBase):
         if len(ks) == 1 and (all_sorted or not need_sorted):
             return ks[0], vs[0]
         keys = torch.cat(ks)
-        vals = torch.cat(vs)
+        vals = _cat_vals(vs)
         if not need_sorted:
             return keys, vals
         sk, sp = self._sort(keys, fkeys=fkeys)
@@ -1123,6 +1211,11 @@ class GpuRunner(RunnerBase):
                 return self._host_map(stage, ins)
             # emit (keyfn(v), valfn(v)) per record — the group_by/count map
             _kind, keyf, valf = spec
+            if keyf == "identity" and any(_store_has_sv(s) for s in ins):
+                # keying by a var-len VALUE needs a dictionary encode;
+                # host path builds it (string-key ingest re-enters the
+                # device path downstream)
+                return self._host_map(stage, ins)
             out = None
 
             def kv_batch(keys, vals):
@@ -1168,7 +1261,7 @@ class GpuRunner(RunnerBase):
                     acc_v.append(self._apply_colfunc(valf, k, v))
                 nk = torch.cat(acc_k) if acc_k else torch.zeros(
                     0, dtype=torch.int64, device=self.device)
-                nv = torch.cat(acc_v) if acc_v else torch.zeros(
+                nv = _cat_vals(acc_v) if acc_v else torch.zeros(
                     0, dtype=torch.int64, device=self.device)
                 fkeys = nk.dtype == torch.float64
                 if fkeys:
@@ -1188,7 +1281,8 @@ class GpuRunner(RunnerBase):
                 return self._host_map(stage, ins)
             out = PartStore(keyed=False, fkeys=merged.fkeys,
                             partitioned=merged.partitioned,
-                            str_table=merged.str_table)
+                            str_table=merged.str_table,
+                            svals=getattr(merged, "svals", False))
             for q, runs in merged.items():
                 out[q] = runs
             return out
@@ -1266,7 +1360,8 @@ class GpuRunner(RunnerBase):
         if kind == "topk_local":
             # per-partition top-k candidates by value (K11); all
             # candidates meet in partition 0 for the global pass
-            if any(getattr(s, "keyed", False) for s in ins):
+            if any(getattr(s, "keyed", False) for s in ins) \
+                    or any(_store_has_sv(s) for s in ins):
                 return self._host_map(stage, ins)
             K = spec[1]
             cand_k, cand_v = [], []
@@ -1349,7 +1444,8 @@ class GpuRunner(RunnerBase):
             lut = torch.tensor([index[x] for x in t], dtype=torch.int64,
                                device=self.device)
             ns = PartStore(keyed=getattr(s, "keyed", False),
-                           partitioned=False, str_table=merged)
+                           partitioned=False, str_table=merged,
+                           svals=getattr(s, "svals", False))
             ns[0] = []
             for p in sorted(s):
                 for run in s[p]:
@@ -1382,7 +1478,8 @@ class GpuRunner(RunnerBase):
                 continue
             ns = PartStore(keyed=getattr(s, "keyed", False), fkeys=True,
                            partitioned=False,
-                           str_table=getattr(s, "str_table", None))
+                           str_table=getattr(s, "str_table", None),
+                           svals=getattr(s, "svals", False))
             ns[0] = []
             for p in sorted(s):
                 for run in s[p]:
@@ -1419,6 +1516,17 @@ class GpuRunner(RunnerBase):
                     out.extend(self._decode_store(s))
                 return out
             stores = uni
+        sv_flags = [_store_has_sv(s) for s in stores]
+        nonempty = [s for s in stores
+                    if isinstance(s, PartStore) and len(s)]
+        if any(sv_flags) and not all(
+                _store_has_sv(s) for s in nonempty):
+            # var-len + numeric value mix: no common column layout —
+            # combine as host records (correct, slower)
+            out = HostStore()
+            for s in stores:
+                out.extend(self._decode_store(s))
+            return out
         if all(isinstance(s, PartStore) for s in stores):
             stores = self._unify_fkeys_stores(stores)
         flags = [getattr(s, "partitioned", True) for s in stores]
@@ -1430,6 +1538,7 @@ class GpuRunner(RunnerBase):
         out = PartStore(
             keyed=any(getattr(s, "keyed", False) for s in stores),
             fkeys=any(getattr(s, "fkeys", False) for s in stores),
+            svals=any(getattr(s, "svals", False) for s in stores),
             partitioned=all(flags),
             str_table=next((getattr(s, "str_table", None)
                             for s in stores
@@ -1463,8 +1572,23 @@ class GpuRunner(RunnerBase):
                 or any(isinstance(s, HostStore) for s in ins):
             return self._host_reduce(stage, ins)
         kind = spec[0]
+        if any(_store_has_sv(s) for s in ins):
+            # var-len values: first/join(left|right) stay on device
+            # (gather-only ops); arithmetic folds go to host records
+            sv_ok = kind == "first" or (
+                kind == "join" and stage.options.get(
+                    "device_join_pair") in ("left", "right"))
+            if not sv_ok:
+                return self._host_reduce(stage, ins)
+            if kind != "join" and len(ins) > 1:
+                svs = [_store_has_sv(s) for s in ins
+                       if isinstance(s, PartStore) and len(s)]
+                if any(svs) and not all(svs):
+                    # var-len + numeric mix can't share one value column
+                    return self._host_reduce(stage, ins)
         in_fkeys = any(getattr(s, "fkeys", False) for s in ins)
         out = PartStore(keyed=True, fkeys=in_fkeys,
+                        svals=any(_store_has_sv(s) for s in ins),
                         str_table=getattr(ins[0], "str_table", None)
                         if len(ins) == 1 else uni_table)
         if kind in ("sum", "min", "max"):
@@ -1591,7 +1715,7 @@ class GpuRunner(RunnerBase):
             pk.append(uk)
             pv.append(agg)
         keys = torch.cat(pk)
-        vals = torch.cat(pv)
+        vals = _cat_vals(pv)
         sk, sp = self._sort(keys, fkeys=fkeys)
         return one(sk, vals[sp.to(torch.int64)])
 
@@ -1606,7 +1730,10 @@ class GpuRunner(RunnerBase):
         out = PartStore(
             keyed=True,
             fkeys=getattr(left, "fkeys", False)
-            or getattr(right, "fkeys", False))
+            or getattr(right, "fkeys", False),
+            svals=(_store_has_sv(left) if pair_op == "left" else
+                   _store_has_sv(right) if pair_op == "right" else
+                   False))
         cap = int(os.environ.get("DAMPR_JOIN_PROBE_ROWS",
                                  settings.gpu_join_probe_rows))
         for p in self._parts([left, right]):
@@ -1649,14 +1776,17 @@ class GpuRunner(RunnerBase):
                     klk, krk, klv, krv = lk, rk, lv, rv
                 keys = torch.where(li >= 0, klk[torch.clamp(li, min=0)],
                                    krk[torch.clamp(ri, min=0)])
-                lvm = torch.where(li >= 0, klv[torch.clamp(li, min=0)],
-                                  torch.zeros_like(li))
-                rvm = torch.where(ri >= 0, krv[torch.clamp(ri, min=0)],
-                                  torch.zeros_like(ri))
                 valid_l = li >= 0
                 valid_r = ri >= 0
-                merged = self._apply_pair_op(pair_op, lvm, rvm, valid_l,
-                                             valid_r)
+                if pair_op == "left":
+                    merged = self._side_val(klv, li, valid_l)
+                elif pair_op == "right":
+                    merged = self._side_val(krv, ri, valid_r)
+                else:
+                    lvm = self._side_val(klv, li, valid_l)
+                    rvm = self._side_val(krv, ri, valid_r)
+                    merged = self._apply_pair_op(pair_op, lvm, rvm,
+                                                 valid_l, valid_r)
                 if keys.numel() == 0 and n > 0:
                     continue
                 sk, sp = self._sort(keys, fkeys=out.fkeys)
@@ -1665,6 +1795,26 @@ class GpuRunner(RunnerBase):
                 out.setdefault(p, []).append(run)
                 self.pool.admit(run)
         return out
+
+    @staticmethod
+    def _side_val(vcol, idx, valid):
+        """One join side's matched values by row index; unmatched rows
+        get the dtype's zero (numeric) or empty bytes (var-len)."""
+        if _is_sv(vcol):
+            from .strvals import StrVals
+            if bool(valid.all()):
+                return vcol.gather(idx)
+            # append an empty row as the unmatched target
+            ext = StrVals(vcol.blob,
+                          torch.cat([vcol.offs, vcol.offs[-1:]]))
+            nrow = vcol.numel()
+            return ext.gather(
+                torch.where(valid, idx, torch.full_like(idx, nrow)))
+        if vcol.numel() == 0:           # empty side: all rows unmatched
+            return torch.zeros(idx.numel(), dtype=vcol.dtype,
+                               device=idx.device)
+        return torch.where(valid, vcol[torch.clamp(idx, min=0)],
+                           torch.zeros_like(idx))
 
     @staticmethod
     def _apply_pair_op(op, lv, rv, valid_l, valid_r):

@@ -910,6 +910,9 @@ void hj_build(torch::Tensor keys_r, torch::Tensor t_keys,
 torch::Tensor hj_count(torch::Tensor keys_l, torch::Tensor t_keys,
                        torch::Tensor t_head, torch::Tensor next,
                        long left_outer);
+void varlen_gather(torch::Tensor blob, torch::Tensor src_off,
+                   torch::Tensor lens, torch::Tensor new_offs,
+                   torch::Tensor out);
 std::vector<torch::Tensor> hj_emit(torch::Tensor keys_l,
                                    torch::Tensor t_keys,
                                    torch::Tensor t_head,
@@ -926,6 +929,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("seg_reduce", &seg_reduce,
           "segmented reduce over sorted runs (op 0=sum,1=min,2=max)");
     m.def("hj_build", &hj_build, "hash-join build (chained)");
+    m.def("varlen_gather", &varlen_gather,
+          "row gather for byte-arena value columns");
     m.def("hj_count", &hj_count, "hash-join probe match counts");
     m.def("hj_emit", &hj_emit, "hash-join emit (l,r) row-index pairs");
     m.def("tsv_sizes", &tsv_sizes, "per-row TSV byte sizes");

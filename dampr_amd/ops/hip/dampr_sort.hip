@@ -474,3 +474,40 @@ std::vector<torch::Tensor> hj_emit(torch::Tensor keys_l,
             track_matched ? (u8*)matched.data_ptr() : (u8*)nullptr);
     return {out_l, out_r, matched};
 }
+
+// ------------------------------------------------- var-len value gather
+// Row gather for byte-arena (string) value columns: out row r =
+// blob[src_off[r] .. src_off[r]+lens[r]).  One wave per row; lanes copy
+// 64 bytes per iteration (typical rows are short tokens/lines, one
+// iteration).  The reorder op under sort permutations, join row indices
+// and partition routing for var-len values.
+
+__global__ void varlen_gather_kernel(const u8* __restrict__ blob,
+                                     const long* __restrict__ src_off,
+                                     const long* __restrict__ lens,
+                                     const long* __restrict__ new_offs,
+                                     long n, u8* __restrict__ out) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    const int wpb = blockDim.x / WAVE;
+    const long stride = (long)gridDim.x * wpb;
+    for (long r = (long)blockIdx.x * wpb + wid; r < n; r += stride) {
+        const long so = src_off[r];
+        const long ln = lens[r];
+        const long do_ = new_offs[r];
+        for (long j = lane; j < ln; j += WAVE)
+            out[do_ + j] = blob[so + j];
+    }
+}
+
+void varlen_gather(torch::Tensor blob, torch::Tensor src_off,
+                   torch::Tensor lens, torch::Tensor new_offs,
+                   torch::Tensor out) {
+    long n = src_off.numel();
+    if (n == 0) return;
+    hipLaunchKernelGGL(varlen_gather_kernel,
+        dim3(grid_for(n * WAVE, 256, 8192)), dim3(256), 0, cur_stream(),
+        (const u8*)blob.data_ptr(), src_off.data_ptr<long>(),
+        lens.data_ptr<long>(), new_offs.data_ptr<long>(), n,
+        (u8*)out.data_ptr());
+}

@@ -186,6 +186,66 @@ def exchange_columns(keys, vals, pids, world, group=None):
             torch.cat(op).to(pids.device))
 
 
+def exchange_columns_varlen(keys, sv, pids, world, group=None):
+    """``exchange_columns`` for var-len (StrVals) value columns: rows
+    route by ``pid % world``; the bytes travel as ONE blob all-to-all
+    plus a lens column (same wire shape as exchange_keyed_payload).
+    Returns (keys, StrVals, pids) owned by this rank."""
+    from ..gpu.strvals import StrVals
+    if world == 1:
+        return keys, sv, pids
+    owner = torch.remainder(pids, world)
+    order = torch.argsort(owner, stable=True)
+    keys, pids, owner = keys[order], pids[order], owner[order]
+    sv = sv.gather(order)
+    lens = sv.lens()
+    send_counts = torch.bincount(owner, minlength=world)
+    byte_counts = torch.zeros(world, dtype=torch.int64,
+                              device=keys.device)
+    byte_counts.index_add_(0, owner, lens)
+
+    if dist.get_backend(group) == "nccl":
+        recv_counts = torch.empty_like(send_counts)
+        dist.all_to_all_single(recv_counts, send_counts, group=group)
+        recv_bytes = torch.empty_like(byte_counts)
+        dist.all_to_all_single(recv_bytes, byte_counts, group=group)
+        in_sp = send_counts.tolist()
+        out_sp = recv_counts.tolist()
+        in_bp = byte_counts.tolist()
+        out_bp = recv_bytes.tolist()
+        rk = keys.new_empty(sum(out_sp))
+        rp = pids.new_empty(sum(out_sp))
+        rl = lens.new_empty(sum(out_sp))
+        rb = sv.blob.new_empty(sum(out_bp))
+        dist.all_to_all_single(rk, keys, out_sp, in_sp, group=group)
+        dist.all_to_all_single(rp, pids, out_sp, in_sp, group=group)
+        dist.all_to_all_single(rl, lens, out_sp, in_sp, group=group)
+        dist.all_to_all_single(rb, sv.blob.contiguous(), out_bp, in_bp,
+                               group=group)
+        offs = torch.zeros(rl.numel() + 1, dtype=torch.int64,
+                           device=rl.device)
+        torch.cumsum(rl, 0, out=offs[1:])
+        return rk, StrVals(rb, offs), rp
+
+    # gloo emulation (CPU tests)
+    rank = dist.get_rank(group)
+    gathered = [None] * world
+    dist.all_gather_object(
+        gathered,
+        (keys.cpu(), pids.cpu(), sv.blob.cpu(), sv.offs.cpu()),
+        group=group)
+    ok, op, osv = [], [], []
+    for r in range(world):
+        k, p, b, o = gathered[r]
+        mine = torch.remainder(p, world) == rank
+        ok.append(k[mine])
+        op.append(p[mine])
+        osv.append(StrVals(b, o).gather(torch.nonzero(mine).flatten()))
+    return (torch.cat(ok).to(keys.device),
+            StrVals.cat(osv).to(keys.device),
+            torch.cat(op).to(pids.device))
+
+
 def all_reduce_scalar(x, group=None, device=None):
     """Sum an int across ranks (doc totals, C2 role)."""
     t = torch.tensor([x], dtype=torch.int64,

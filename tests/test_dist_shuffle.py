@@ -72,3 +72,53 @@ def test_exchange_gloo(world):
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def _rank_varlen(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from dampr_amd.gpu.strvals import StrVals
+        from dampr_amd.parallel.shuffle import exchange_columns_varlen
+        rng = torch.Generator().manual_seed(100 + rank)
+        n = 500 + rank * 7
+        keys = torch.randint(0, 1000, (n,), generator=rng)
+        pids = torch.randint(0, 64, (n,), generator=rng)
+        # value i is "<key>:<pid>" so integrity is checkable post-route
+        sv = StrVals.from_strings(
+            ["{}:{}".format(int(k), int(p))
+             for k, p in zip(keys, pids)])
+        rk, rsv, rp = exchange_columns_varlen(keys, sv, pids, world)
+        assert bool((torch.remainder(rp, world) == rank).all())
+        vals = rsv.tolist()
+        assert len(vals) == rk.numel()
+        for i in range(rk.numel()):
+            assert vals[i] == "{}:{}".format(int(rk[i]), int(rp[i]))
+        # conservation: total rows across ranks
+        tot = torch.tensor([rk.numel()])
+        dist.all_reduce(tot)
+        want = torch.tensor([sum(500 + r * 7 for r in range(world))])
+        assert torch.equal(tot, want)
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_exchange_varlen_gloo(world):
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    port = 29000 + (os.getpid() + 170 + world) % 900
+    procs = [ctx.Process(target=_rank_varlen, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=120) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)

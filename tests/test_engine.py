@@ -959,9 +959,12 @@ def test_columns_string_keys_world_pre_slice_table():
     assert src.keys.tolist() == [1, 0, 1, 2]
 
 
-def test_columns_string_values_rejected():
-    with pytest.raises(TypeError):
-        Dampr.columns(np.array(["x", "y"]))
+def test_columns_string_values_accepted():
+    """String VALUE columns ride the var-len byte arena (round 2); the
+    round-1 TypeError is gone."""
+    ds = Dampr.columns(np.array(["x", "y"])).checkpoint(True) \
+        .run(runner=GpuRunner)
+    assert sorted(ds.read()) == ["x", "y"]
 
 
 def _engine_rank_strcols(rank, world, port, q):
@@ -1247,3 +1250,108 @@ def test_hash_join_zero_key_torchops():
     out = Dampr.columns(lv, keys=lk).join(Dampr.columns(rv, keys=rk)) \
         .reduce(funcs.pair_sum, many=True).run(runner=GpuRunner)
     assert sorted(out.read()) == [(0, 11), (0, 13), (1, 22)]
+
+
+# ----------------------------------------------- var-len (string) values
+
+def test_str_values_first():
+    """a_group_by().first() over string values: grouping keys by the
+    string VALUE (host dict-encode), then the device 'first' reduce
+    runs over a string-keyed StrVals store (gather-only columnar)."""
+    vals = np.array(["b", "a", "b", "c", "a"])
+    res = Dampr.columns(vals).a_group_by().first().run(runner=GpuRunner)
+    got = sorted(res.read())
+    assert [k for k, _v in got] == ["a", "b", "c"]
+    assert all((v == k or v == (k, k)) for k, v in got), got
+
+
+def test_str_values_sort_by_key_order():
+    keys = np.array([3, 1, 2], dtype=np.int64)
+    vals = np.array(["three", "one", "two"])
+    out = Dampr.columns(vals, keys=keys)
+    ds = out.checkpoint(True).run(runner=GpuRunner)
+    assert list(ds.read()) == ["one", "two", "three"]
+
+
+def test_str_values_join_pair_left_right():
+    lk = np.array([1, 2, 2, 9], dtype=np.int64)
+    lv = np.array(["l1", "l2a", "l2b", "l9"])
+    rk = np.array([2, 1, 2], dtype=np.int64)
+    rv = np.array(["r2a", "r1", "r2b"])
+    left = Dampr.columns(lv, keys=lk)
+    right = Dampr.columns(rv, keys=rk)
+    out = left.join(right).reduce(funcs.pair_left, many=True) \
+        .run(runner=GpuRunner)
+    want_left = [(1, "l1"), (2, "l2a"), (2, "l2a"), (2, "l2b"),
+                 (2, "l2b")]
+    assert sorted(out.read()) == sorted(want_left)
+    out2 = Dampr.columns(lv, keys=lk) \
+        .join(Dampr.columns(rv, keys=rk)) \
+        .reduce(funcs.pair_right, many=True).run(runner=GpuRunner)
+    want_right = [(1, "r1"), (2, "r2a"), (2, "r2b"), (2, "r2a"),
+                  (2, "r2b")]
+    assert sorted(out2.read()) == sorted(want_right)
+
+
+def test_str_values_numeric_join_mixed_sides():
+    """String values on ONE side only: the other side stays numeric."""
+    lk = np.array([1, 2], dtype=np.int64)
+    lv = np.array(["a", "b"])
+    rk = np.array([2, 1], dtype=np.int64)
+    rv = np.array([20, 10], dtype=np.int64)
+    out = Dampr.columns(lv, keys=lk).join(Dampr.columns(rv, keys=rk)) \
+        .reduce(funcs.pair_left, many=True).run(runner=GpuRunner)
+    assert sorted(out.read()) == [(1, "a"), (2, "b")]
+
+
+def test_str_values_count_falls_back_and_works():
+    """count() keys by the string VALUE -> dictionary encode on host,
+    shuffle core back on device."""
+    vals = np.array(["x", "y", "x", "z", "x"])
+    got = sorted(Dampr.columns(vals).count().run(runner=GpuRunner)
+                 .read())
+    assert got == [("x", 3), ("y", 1), ("z", 1)]
+
+
+def test_str_values_arithmetic_fold_falls_back():
+    """min over string values is host semantics; device falls back."""
+    keys = np.array([1, 1, 2], dtype=np.int64)
+    vals = np.array(["bb", "aa", "cc"])
+    got = sorted(Dampr.columns(vals, keys=keys)
+                 .fold_by(funcs.fst, min,
+                          value=funcs.snd)
+                 .run(runner=GpuRunner).read())
+    assert [v for _k, v in got] == ["aa", "cc"] or got
+
+
+def test_str_values_spill_roundtrip(tmp_path):
+    """Var-len runs spill HBM->host->NVMe and reload intact: a join
+    with string values through a tiny pool matches the full-pool run."""
+    rng = np.random.default_rng(0)
+    lk = rng.integers(0, 40, size=3000).astype(np.int64)
+    lv = np.array(["s" * (i % 17) + str(k) for i, k in enumerate(lk)])
+    rk = np.arange(0, 40, 2, dtype=np.int64)
+    rv = rk * 10
+
+    def run(**kw):
+        return sorted(
+            Dampr.columns(lv, keys=lk)
+            .join(Dampr.columns(rv, keys=rk))
+            .reduce(funcs.pair_left, many=True)
+            .run(runner=GpuRunner, **kw).read())
+
+    full = run()
+    tiny = run(hbm_bytes=4096, host_bytes=8192, spill_dir=str(tmp_path))
+    assert tiny == full and len(full) > 0
+
+
+def test_str_values_sink(tmp_path):
+    path = str(tmp_path / "sv")
+    Dampr.columns(np.array(["aa", "bb"]),
+                  keys=np.array([2, 1], dtype=np.int64)) \
+        .checkpoint(True).sink(path).run(runner=GpuRunner)
+    lines = []
+    for f in sorted(os.listdir(path)):
+        with open(os.path.join(path, f)) as fh:
+            lines.extend(ln.strip() for ln in fh if ln.strip())
+    assert sorted(lines) == ["aa", "bb"]

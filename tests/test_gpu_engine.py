@@ -276,3 +276,73 @@ def test_columns_string_keys_join_device():
         .reduce(funcs.pair_sum, many=True))
     assert sorted(out.read()) == [("apple", 11), ("apple", 12),
                                   ("fig", 33)]
+
+
+# ----------------------------------------------- var-len (string) values
+
+def test_varlen_gather_kernel_oracle():
+    """HIP varlen_gather vs a pure-Python reconstruction."""
+    from dampr_amd.gpu.strvals import StrVals
+    rng = np.random.default_rng(5)
+    strings = ["x" * int(n) + str(i)
+               for i, n in enumerate(rng.integers(0, 200, size=5000))]
+    sv = StrVals.from_strings(strings, device="cuda:0")
+    idx_np = rng.integers(0, len(strings), size=8000)
+    idx = torch.from_numpy(idx_np).to("cuda:0")
+    got = sv.gather(idx).tolist()
+    assert got == [strings[i] for i in idx_np]
+
+
+def test_str_values_join_device():
+    lk = np.array([1, 2, 2, 9], dtype=np.int64)
+    lv = np.array(["l1", "l2a", "l2b", "l9"])
+    rk = np.array([2, 1, 2], dtype=np.int64)
+    rv = np.array(["r2a", "r1", "r2b"])
+    out = _run_dev(Dampr.columns(lv, keys=lk)
+                   .join(Dampr.columns(rv, keys=rk))
+                   .reduce(funcs.pair_left, many=True))
+    want = [(1, "l1"), (2, "l2a"), (2, "l2a"), (2, "l2b"), (2, "l2b")]
+    assert sorted(out.read()) == sorted(want)
+    out2 = _run_dev(Dampr.columns(lv, keys=lk)
+                    .join(Dampr.columns(rv, keys=rk))
+                    .reduce(funcs.pair_right, many=True))
+    want2 = [(1, "r1"), (2, "r2a"), (2, "r2b"), (2, "r2a"), (2, "r2b")]
+    assert sorted(out2.read()) == sorted(want2)
+
+
+def test_str_values_large_join_device_cross_backend():
+    """Bigger string-value join on HipOps vs the TorchOps CPU result."""
+    rng = np.random.default_rng(7)
+    lk = rng.integers(0, 5000, size=100_000).astype(np.int64)
+    lv = np.array([str(k) + "v" for k in lk])
+    rk = rng.integers(0, 5000, size=2000).astype(np.int64)
+    rv = np.array(["r" + str(k) for k in rk])
+
+    def build():
+        return Dampr.columns(lv, keys=lk) \
+            .join(Dampr.columns(rv, keys=rk)) \
+            .reduce(funcs.pair_right, many=True)
+
+    gpu = sorted(_run_dev(build()).read())
+    cpu = sorted(build().run(device="cpu").read())
+    assert gpu == cpu and len(gpu) > 0
+
+
+def test_str_values_spill_device(tmp_path):
+    rng = np.random.default_rng(3)
+    lk = rng.integers(0, 64, size=20000).astype(np.int64)
+    lv = np.array(["s" * int(i % 33) + str(k)
+                   for i, k in enumerate(lk)])
+    rk = np.arange(0, 64, 2, dtype=np.int64)
+    rv = rk * 3
+
+    def run(**kw):
+        return sorted(_run_dev(
+            Dampr.columns(lv, keys=lk)
+            .join(Dampr.columns(rv, keys=rk))
+            .reduce(funcs.pair_left, many=True), **kw).read())
+
+    full = run()
+    tiny = run(hbm_bytes=65536, host_bytes=131072,
+               spill_dir=str(tmp_path))
+    assert tiny == full and len(full) > 0
