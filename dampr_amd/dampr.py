@@ -436,10 +436,17 @@ class PMap(PBase):
         return PMap(source, pmer)
 
     def cross_left(self, other, cross, memory=False, **options):
-        """Cross product; self is the left side."""
+        """Cross product; self is the left side.  When ``cross`` is a
+        recognized commutative binop (funcs.CROSS_BINOPS) the columnar
+        engine broadcasts the small side and fuses the apply on device
+        (K9)."""
         def _cross(k1, v1, k2, v2):
             yield k1, cross(v2, v1)
 
+        cname = funcs.cross_binop_name(cross)
+        if cname:
+            options = dict(options)
+            options.setdefault("device_map", ("cross", cname))
         me = self.checkpoint()
         other = other.checkpoint()
         pmer = Dampr(me.pmer.graph.union(other.pmer.graph))
@@ -452,7 +459,10 @@ class PMap(PBase):
         """Cross product; self is the right side.  With ``memory=True`` the
         small side is cached (GPU path: broadcast, K9)."""
         assert isinstance(other, PMap)
-        return other.cross_left(self, lambda xi, yi: cross(yi, xi), memory)
+        cname = funcs.cross_binop_name(cross)
+        kw = {"device_map": ("cross", cname)} if cname else {}
+        return other.cross_left(self, lambda xi, yi: cross(yi, xi),
+                                memory, **kw)
 
     def cross_set(self, other, cross, agg=None, **options):
         """Cross each value against the whole of ``other``, aggregated once

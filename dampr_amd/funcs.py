@@ -104,6 +104,26 @@ JOIN_PAIR_FUNCS = {
 }
 
 
+# cross-join (K9) apply ops the device can fuse over the broadcast
+# product; all COMMUTATIVE, so cross_left/cross_right share one table
+# (cross_right wraps its lambda with swapped operands).
+CROSS_BINOPS = {
+    add: "add",
+    operator.add: "add",
+    mul: "mul",
+    operator.mul: "mul",
+    min: "min",
+    max: "max",
+}
+
+
+def cross_binop_name(f):
+    try:
+        return CROSS_BINOPS.get(f)
+    except TypeError:
+        return None
+
+
 def join_pair_name(f):
     try:
         return JOIN_PAIR_FUNCS.get(f)

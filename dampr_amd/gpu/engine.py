@@ -1191,6 +1191,20 @@ class GpuRunner(RunnerBase):
         sk, sp = self._sort(keys, fkeys=fkeys)
         return sk, vals[sp.to(torch.int64)]
 
+    def _all_rows(self, store):
+        """All (keys, vals) rows of a store, concatenated on device."""
+        ks, vs = [], []
+        for p in sorted(store):
+            k, v = self._merged_partition([store], p, need_sorted=False)
+            if k is None:
+                continue
+            ks.append(k)
+            vs.append(v)
+        if not ks:
+            z = torch.zeros(0, dtype=torch.int64, device=self.device)
+            return z, z.clone()
+        return torch.cat(ks), _cat_vals(vs)
+
     def _parts(self, stores):
         ps = set()
         for s in stores:
@@ -1357,6 +1371,42 @@ class GpuRunner(RunnerBase):
                 # received token strings live in the exchanged blob
                 text = rblob
             return TokenStore(eng, keys, df, text)
+        if kind == "cross":
+            # K9 broadcast cross join: ins[0] = streamed (outer) side
+            # whose keys the output carries, ins[1] = broadcast side
+            # (gathered across ranks like the host supplemental path).
+            opn = spec[1]
+            if len(ins) != 2 \
+                    or any(not isinstance(s, PartStore) for s in ins) \
+                    or any(getattr(s, "keyed", False) for s in ins) \
+                    or any(_store_has_sv(s) for s in ins):
+                return self._host_map(stage, ins)
+            other, me = ins
+            ko, vo = self._all_rows(other)
+            km, vm = self._all_rows(me)
+            if self.world > 1:
+                from ..parallel.shuffle import gather_columns
+                km, vm = gather_columns(km, vm, device=self.device)
+            n_o, m = ko.numel(), km.numel()
+            out_k = torch.repeat_interleave(ko, m) if m else \
+                torch.zeros(0, dtype=torch.int64, device=self.device)
+            if n_o and m:
+                a = vm.repeat(n_o)                   # me value (v2)
+                b = torch.repeat_interleave(vo, m)   # other value (v1)
+                if a.dtype != b.dtype:
+                    dt = torch.promote_types(a.dtype, b.dtype)
+                    a, b = a.to(dt), b.to(dt)
+                out_v = {"add": lambda: a + b,
+                         "mul": lambda: a * b,
+                         "min": lambda: torch.minimum(a, b),
+                         "max": lambda: torch.maximum(a, b)}[opn]()
+            else:
+                out_v = torch.zeros(0, dtype=torch.int64,
+                                    device=self.device)
+            st = self._partition(out_k, out_v,
+                                 fkeys=getattr(other, "fkeys", False))
+            st.str_table = getattr(other, "str_table", None)
+            return st
         if kind == "topk_local":
             # per-partition top-k candidates by value (K11); all
             # candidates meet in partition 0 for the global pass

@@ -246,6 +246,28 @@ def exchange_columns_varlen(keys, sv, pids, world, group=None):
             torch.cat(op).to(pids.device))
 
 
+def gather_columns(keys, vals, device=None, group=None):
+    """Union of every rank's (keys, vals) columns, identical on all
+    ranks — the broadcast-small-side collective for device cross joins
+    (K9; host-record analog: engine._host_gather)."""
+    world = dist.get_world_size(group)
+    if world == 1:
+        return keys, vals
+    gathered = [None] * world
+    dist.all_gather_object(gathered, (keys.cpu(), vals.cpu()),
+                           group=group)
+    ks = [g[0] for g in gathered]
+    vs = [g[1] for g in gathered]
+    dt = vs[0].dtype
+    for v in vs[1:]:
+        dt = torch.promote_types(dt, v.dtype)
+    k = torch.cat(ks)
+    v = torch.cat([x.to(dt) for x in vs])
+    if device is not None:
+        k, v = k.to(device), v.to(device)
+    return k, v
+
+
 def all_reduce_scalar(x, group=None, device=None):
     """Sum an int across ranks (doc totals, C2 role)."""
     t = torch.tensor([x], dtype=torch.int64,

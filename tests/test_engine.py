@@ -1355,3 +1355,70 @@ def test_str_values_sink(tmp_path):
         with open(os.path.join(path, f)) as fh:
             lines.extend(ln.strip() for ln in fh if ln.strip())
     assert sorted(lines) == ["aa", "bb"]
+
+
+# ------------------------------------------------------ device cross (K9)
+
+def test_cross_right_recognized_binop_device_matches_host():
+    """cross_right with a recognized commutative binop runs the K9
+    broadcast-apply on device; host engine is the oracle."""
+    import operator
+    vals = np.arange(1, 8, dtype=np.int64)
+    dev = sorted(
+        Dampr.columns(vals)
+        .cross_right(Dampr.columns(np.array([10], dtype=np.int64)),
+                     operator.mul)
+        .run(runner=GpuRunner).read())
+    host = sorted(
+        Dampr.memory([10])
+        .cross_left(Dampr.memory(list(range(1, 8))), operator.mul)
+        .run().read())
+    assert dev == host == [10 * v for v in range(1, 8)]
+
+
+def test_cross_left_recognized_binop_device_matches_host():
+    import operator
+    a = np.array([1, 2, 3], dtype=np.int64)
+    b = np.array([100, 200], dtype=np.int64)
+    dev = sorted(
+        Dampr.columns(a).cross_left(Dampr.columns(b), operator.add)
+        .run(runner=GpuRunner).read())
+    host = sorted(
+        Dampr.memory([1, 2, 3]).cross_left(Dampr.memory([100, 200]),
+                                           operator.add).run().read())
+    assert dev == host
+
+
+def test_cross_min_max_device():
+    a = np.array([5, 1], dtype=np.int64)
+    b = np.array([3], dtype=np.int64)
+    dev = sorted(Dampr.columns(a)
+                 .cross_left(Dampr.columns(b), min)
+                 .run(runner=GpuRunner).read())
+    assert dev == [1, 3]
+    dev2 = sorted(Dampr.columns(a)
+                  .cross_left(Dampr.columns(b), max)
+                  .run(runner=GpuRunner).read())
+    assert dev2 == [3, 5]
+
+
+def test_cross_float_promotion_device():
+    a = np.array([1.5, 2.5])
+    b = np.array([2], dtype=np.int64)
+    dev = sorted(Dampr.columns(a)
+                 .cross_left(Dampr.columns(b), funcs.mul)
+                 .run(runner=GpuRunner).read())
+    assert dev == [3.0, 5.0]
+
+
+def test_cross_opaque_still_host_fallback():
+    a = np.array([1, 2], dtype=np.int64)
+    b = np.array([10], dtype=np.int64)
+    got = sorted(Dampr.columns(a)
+                 .cross_left(Dampr.columns(b), lambda x, y: x - y)
+                 .run(runner=GpuRunner).read())
+    # host semantics: cross(v_me, v_other) per cross_left's _cross
+    host = sorted(Dampr.memory([1, 2])
+                  .cross_left(Dampr.memory([10]), lambda x, y: x - y)
+                  .run().read())
+    assert got == host
