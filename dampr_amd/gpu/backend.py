@@ -39,6 +39,14 @@ class _OpsBase(object):
     def hash_join(self, keys_l, keys_r, how="inner"):
         raise NotImplementedError
 
+    def merge_sorted_runs(self, ks, fkeys=False):
+        """K-way merge of key-sorted runs (K4).  ``ks`` is a list of
+        runs each already in host-ascending order (signed for i64 keys,
+        encoded-unsigned for fkeys).  Returns (merged_keys, perm) where
+        perm indexes into the runs' concatenation.  Stable: ties keep
+        run order, then within-run order."""
+        raise NotImplementedError
+
     # shared helpers ------------------------------------------------------
 
     def partition_of(self, keys, n_partitions):
@@ -92,6 +100,36 @@ class HipOps(_OpsBase):
         from .relational import hash_join
         return hash_join(keys_l, keys_r, how)
 
+    def merge_sorted_runs(self, ks, fkeys=False):
+        total = sum(k.numel() for k in ks)
+        if total >= (1 << 31):      # u32 payload limit (same as sort)
+            raise OverflowError("merge payload exceeds u32")
+        dev = ks[0].device
+        items = []
+        base = 0
+        for k in ks:
+            p = torch.arange(base, base + k.numel(), dtype=torch.int32,
+                             device=dev)
+            items.append((k.contiguous(), p))
+            base += k.numel()
+        bias = 0 if fkeys else 1    # signed compare for raw i64 keys
+        # pairwise merge tree over ADJACENT runs (keeps global stability)
+        while len(items) > 1:
+            nxt = []
+            for a in range(0, len(items) - 1, 2):
+                ka, pa = items[a]
+                kb, pb = items[a + 1]
+                n = ka.numel() + kb.numel()
+                out_k = torch.empty(n, dtype=ka.dtype, device=dev)
+                out_p = torch.empty(n, dtype=torch.int32, device=dev)
+                self.ext.mp_merge(ka, pa, kb, pb, bias, out_k, out_p)
+                nxt.append((out_k, out_p))
+            if len(items) % 2:
+                nxt.append(items[-1])
+            items = nxt
+        k, p = items[0]
+        return k, p.to(torch.int64)
+
 
 class TorchOps(_OpsBase):
     """Pure-torch oracle (CPU tests; never used on a CUDA device)."""
@@ -121,6 +159,12 @@ class TorchOps(_OpsBase):
         out[:n_seg] = out[:n_seg].scatter_reduce(
             0, seg, vals, reduce=red, include_self=True)
         return uniq, out[:n_seg]
+
+    def merge_sorted_runs(self, ks, fkeys=False):
+        keys = torch.cat(ks)
+        ordk = (keys ^ _I64_MIN) if fkeys else keys
+        perm = torch.argsort(ordk, stable=True)
+        return keys[perm], perm
 
     def hash_join(self, keys_l, keys_r, how="inner"):
         assert how in ("inner", "left", "outer")

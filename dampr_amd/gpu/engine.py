@@ -1032,15 +1032,21 @@ class GpuRunner(RunnerBase):
         fkeys = getattr(store, "fkeys", False)
         tbl = getattr(store, "str_table", None)
         ks, vs = [], []
+        all_sorted = True
         for p in sorted(store):
             for run in store[p]:
                 self.pool.touch(run, self.device)
                 ks.append(run.keys)
                 vs.append(run.vals)
+                all_sorted = all_sorted and run.sorted
                 self.pool.release(run)
         if not ks:
             z = torch.zeros(0, dtype=torch.int64)
             return ColumnDataset(z, z.clone(), keyed, fkeys, tbl)
+        if all_sorted and len(ks) > 1:
+            sk, perm = self.ops.merge_sorted_runs(ks, fkeys=fkeys)
+            return ColumnDataset(sk, _cat_vals(vs)[perm], keyed, fkeys,
+                                 tbl)
         keys = torch.cat(ks)
         vals = _cat_vals(vs)
         sk, sp = self._sort(keys, fkeys=fkeys)
@@ -1184,10 +1190,16 @@ class GpuRunner(RunnerBase):
             return None, None
         if len(ks) == 1 and (all_sorted or not need_sorted):
             return ks[0], vs[0]
+        if not need_sorted:
+            return torch.cat(ks), _cat_vals(vs)
+        if all_sorted and len(ks) > 1:
+            # K4: merge-path k-way merge of already-sorted runs — one
+            # log2(R)-deep pass tree instead of a full radix re-sort
+            # (reference analog: heapq.merge, dataset.py:567-588)
+            mk, perm = self.ops.merge_sorted_runs(ks, fkeys=fkeys)
+            return mk, _cat_vals(vs)[perm]
         keys = torch.cat(ks)
         vals = _cat_vals(vs)
-        if not need_sorted:
-            return keys, vals
         sk, sp = self._sort(keys, fkeys=fkeys)
         return sk, vals[sp.to(torch.int64)]
 

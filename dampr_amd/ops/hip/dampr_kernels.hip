@@ -910,6 +910,9 @@ void hj_build(torch::Tensor keys_r, torch::Tensor t_keys,
 torch::Tensor hj_count(torch::Tensor keys_l, torch::Tensor t_keys,
                        torch::Tensor t_head, torch::Tensor next,
                        long left_outer);
+void mp_merge(torch::Tensor ka, torch::Tensor pa, torch::Tensor kb,
+              torch::Tensor pb, long bias_signed, torch::Tensor out_k,
+              torch::Tensor out_p);
 void varlen_gather(torch::Tensor blob, torch::Tensor src_off,
                    torch::Tensor lens, torch::Tensor new_offs,
                    torch::Tensor out);
@@ -931,6 +934,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("hj_build", &hj_build, "hash-join build (chained)");
     m.def("varlen_gather", &varlen_gather,
           "row gather for byte-arena value columns");
+    m.def("mp_merge", &mp_merge,
+          "stable merge-path 2-way merge of sorted (key, payload) runs");
     m.def("hj_count", &hj_count, "hash-join probe match counts");
     m.def("hj_emit", &hj_emit, "hash-join emit (l,r) row-index pairs");
     m.def("tsv_sizes", &tsv_sizes, "per-row TSV byte sizes");

@@ -511,3 +511,86 @@ void varlen_gather(torch::Tensor blob, torch::Tensor src_off,
         lens.data_ptr<long>(), new_offs.data_ptr<long>(), n,
         (u8*)out.data_ptr());
 }
+
+// ------------------------------------------------------ merge-path (K4)
+// Stable 2-way merge of key-sorted runs (Merge Path, Green et al.): each
+// block owns one output tile, finds its (i, j) split with a diagonal
+// binary search, and its threads merge per-thread sub-segments found
+// the same way.  K-way merge of spilled sorted runs = pairwise tree of
+// these (log2 R passes) — replaces the cat+re-sort of sorted runs
+// (reference's heapq.merge role, dataset.py:567-588).  ``bias`` is
+// XORed into keys for comparisons only (1<<63 = signed order for i64
+// keys; 0 = raw unsigned for f64-encoded keys), so no separate
+// re-encode passes are needed.  Ties take run A first (stability).
+
+#define MP_TILE 4096
+#define MP_IPT 16
+
+__device__ __forceinline__ long mp_diag(const u64* __restrict__ a,
+                                        long na,
+                                        const u64* __restrict__ b,
+                                        long nb, long diag, u64 bias) {
+    long lo = diag > nb ? diag - nb : 0;
+    long hi = diag < na ? diag : na;
+    while (lo < hi) {
+        long mid = (lo + hi) >> 1;
+        if ((a[mid] ^ bias) <= (b[diag - 1 - mid] ^ bias))
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    return lo;
+}
+
+__global__ void __launch_bounds__(256)
+mp_merge_kernel(const u64* __restrict__ ka, const u32* __restrict__ pa,
+                long na, const u64* __restrict__ kb,
+                const u32* __restrict__ pb, long nb, u64 bias,
+                u64* __restrict__ out_k, u32* __restrict__ out_p) {
+    const long total = na + nb;
+    const long tile0 = (long)blockIdx.x * MP_TILE;
+    if (tile0 >= total) return;
+    const long tile1 = min(tile0 + (long)MP_TILE, total);
+    // block split
+    const long bi0 = mp_diag(ka, na, kb, nb, tile0, bias);
+    const long bi1 = mp_diag(ka, na, kb, nb, tile1, bias);
+    const long bj0 = tile0 - bi0;
+    const long bj1 = tile1 - bi1;
+    // per-thread split within the block's segments
+    const long d0 = min(tile0 + (long)threadIdx.x * MP_IPT, tile1);
+    const long d1 = min(d0 + (long)MP_IPT, tile1);
+    if (d0 >= d1) return;
+    long i = bi0 + mp_diag(ka + bi0, bi1 - bi0, kb + bj0, bj1 - bj0,
+                           d0 - tile0, bias);
+    long j = d0 - i;
+    for (long o = d0; o < d1; ++o) {
+        bool take_a;
+        if (i >= bi1) take_a = false;
+        else if (j >= bj1) take_a = true;
+        else take_a = (ka[i] ^ bias) <= (kb[j] ^ bias);
+        if (take_a) {
+            out_k[o] = ka[i];
+            out_p[o] = pa[i];
+            ++i;
+        } else {
+            out_k[o] = kb[j];
+            out_p[o] = pb[j];
+            ++j;
+        }
+    }
+}
+
+void mp_merge(torch::Tensor ka, torch::Tensor pa, torch::Tensor kb,
+              torch::Tensor pb, long bias_signed, torch::Tensor out_k,
+              torch::Tensor out_p) {
+    long na = ka.numel(), nb = kb.numel();
+    long total = na + nb;
+    if (total == 0) return;
+    u64 bias = bias_signed ? 0x8000000000000000ULL : 0ULL;
+    long nblocks = (total + MP_TILE - 1) / MP_TILE;
+    hipLaunchKernelGGL(mp_merge_kernel, dim3((u32)nblocks), dim3(256), 0,
+        cur_stream(), (const u64*)ka.data_ptr(),
+        (const u32*)pa.data_ptr(), na, (const u64*)kb.data_ptr(),
+        (const u32*)pb.data_ptr(), nb, bias, (u64*)out_k.data_ptr(),
+        (u32*)out_p.data_ptr());
+}

@@ -205,3 +205,47 @@ def test_radix_sort_input_not_clobbered(rng):
     torch.cuda.synchronize()
     np.testing.assert_array_equal(keys.cpu().numpy(), keys_np)
     assert torch.equal(keys[sp.to(torch.int64)], sk)
+
+
+def test_merge_sorted_runs_oracle(rng):
+    """Merge-path k-way merge vs numpy: signed i64 order, ties stable
+    (run order, then within-run order)."""
+    from dampr_amd.gpu.backend import HipOps
+    ops = HipOps()
+    runs_np = []
+    for r in range(7):                       # odd count: tree leftover
+        n = int(rng.integers(1, 40_000))
+        a = np.sort(rng.integers(-1000, 1000, size=n, dtype=np.int64))
+        runs_np.append(a)
+    ks = [torch.from_numpy(a).to(DEV) for a in runs_np]
+    mk, perm = ops.merge_sorted_runs(ks, fkeys=False)
+    cat = np.concatenate(runs_np)
+    order = np.argsort(cat, kind="stable")
+    np.testing.assert_array_equal(mk.cpu().numpy(), cat[order])
+    np.testing.assert_array_equal(perm.cpu().numpy(), order)
+
+
+def test_merge_sorted_runs_unsigned_fkeys(rng):
+    from dampr_amd.gpu.backend import HipOps
+    ops = HipOps()
+    runs_np = []
+    for r in range(3):
+        a = rng.integers(0, 1 << 63, size=30_000, dtype=np.int64) \
+            .view(np.uint64)
+        a = np.sort(a).view(np.int64)        # unsigned ascending
+        runs_np.append(a)
+    ks = [torch.from_numpy(a).to(DEV) for a in runs_np]
+    mk, perm = ops.merge_sorted_runs(ks, fkeys=True)
+    cat = np.concatenate(runs_np)
+    order = np.argsort(cat.view(np.uint64), kind="stable")
+    np.testing.assert_array_equal(mk.cpu().numpy(), cat[order])
+
+
+def test_merge_sorted_runs_empty_and_single():
+    from dampr_amd.gpu.backend import HipOps
+    ops = HipOps()
+    e = torch.zeros(0, dtype=torch.int64, device=DEV)
+    a = torch.tensor([1, 2, 3], dtype=torch.int64, device=DEV)
+    mk, perm = ops.merge_sorted_runs([e, a, e], fkeys=False)
+    assert mk.cpu().tolist() == [1, 2, 3]
+    assert perm.cpu().tolist() == [0, 1, 2]
