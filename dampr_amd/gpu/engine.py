@@ -341,7 +341,8 @@ class DeviceRun(object):
     /tmp spill files, dataset.py:119-188, re-expressed for 288 GB HBM +
     host DRAM + NVMe)."""
 
-    __slots__ = ("keys", "vals", "sorted", "_host", "_disk", "_meta")
+    __slots__ = ("keys", "vals", "sorted", "_host", "_disk", "_meta",
+                 "_evt")
 
     def __init__(self, keys, vals, sorted=False):
         self.keys = keys
@@ -350,6 +351,7 @@ class DeviceRun(object):
         self._host = None
         self._disk = None
         self._meta = None
+        self._evt = None           # in-flight async D2H spill marker
 
     @property
     def n(self):
@@ -383,6 +385,65 @@ class DeviceRun(object):
     def on_disk(self):
         return self._disk is not None
 
+    def _wait_spill(self):
+        """Block until an in-flight async spill's D2H copies land."""
+        if self._evt is not None:
+            self._evt.synchronize()
+            self._evt = None
+
+    def drop(self):
+        """Release storage on every tier (caller owns accounting and
+        disk unlink).  Waits for in-flight spill DMA first — freeing a
+        pinned buffer under an active copy corrupts host memory."""
+        self._wait_spill()
+        self.keys = None
+        self.vals = None
+        self._host = None
+        self._meta = None
+
+    def spill_async(self, stream):
+        """HBM -> pinned host on a dedicated D2H stream, overlapped
+        with compute on the main stream.  Device tensors are released
+        immediately (record_stream defers allocator reuse until the
+        copies complete); host-side readers must _wait_spill()."""
+        if stream is None or self.keys is None \
+                or self.keys.device.type != "cuda":
+            return self.spill()
+        if self._host is not None:
+            return
+        from .strvals import StrVals
+        main = torch.cuda.current_stream(self.keys.device)
+        with torch.cuda.stream(stream):
+            # spill reads must see the producing kernels' writes
+            stream.wait_stream(main)
+            hk = torch.empty_like(self.keys, device="cpu",
+                                  pin_memory=True)
+            hk.copy_(self.keys, non_blocking=True)
+            self.keys.record_stream(stream)
+            if _is_sv(self.vals):
+                hb = torch.empty_like(self.vals.blob, device="cpu",
+                                      pin_memory=True)
+                ho = torch.empty_like(self.vals.offs, device="cpu",
+                                      pin_memory=True)
+                hb.copy_(self.vals.blob, non_blocking=True)
+                ho.copy_(self.vals.offs, non_blocking=True)
+                self.vals.record_stream(stream)
+                hv = StrVals(hb, ho)
+                self._meta = (self.keys.numel(),
+                              ("str", self.vals.blob.numel()))
+            else:
+                hv = torch.empty_like(self.vals, device="cpu",
+                                      pin_memory=True)
+                hv.copy_(self.vals, non_blocking=True)
+                self.vals.record_stream(stream)
+                self._meta = (self.keys.numel(), self.vals.dtype)
+            evt = torch.cuda.Event()
+            evt.record(stream)
+        self._evt = evt
+        self._host = (hk, hv)
+        self.keys = None
+        self.vals = None
+
     def spill(self):
         """HBM -> (pinned) host memory."""
         if self._host is not None or self.keys is None:
@@ -414,6 +475,7 @@ class DeviceRun(object):
         """Host -> NVMe file (raw little-endian columns, no pickle)."""
         if self._host is None:
             return
+        self._wait_spill()
         hk, hv = self._host
         with open(path, "wb") as fh:
             fh.write(hk.numpy().tobytes())
@@ -456,6 +518,7 @@ class DeviceRun(object):
 
     def load(self, device):
         if self.keys is None:
+            self._wait_spill()
             self._load_host()
             hk, hv = self._host
             self.keys = hk.to(device, non_blocking=True)
@@ -486,6 +549,9 @@ class HbmPool(object):
         self.host_used = 0
         self._host_lru = {}
         self.spill_dir = spill_dir or settings.spill_dir
+        # dedicated D2H stream: evictions overlap main-stream compute
+        # (set by the engine on CUDA devices; None = synchronous spill)
+        self.spill_stream = None
         self._run_tag = "dampr_amd_{}".format(uuid.uuid4().hex[:10])
         self._file_ctr = 0
         self._disk_paths = []
@@ -579,7 +645,7 @@ class HbmPool(object):
                 return
             del self._lru[victim]
             self.used -= victim.nbytes
-            victim.spill()
+            victim.spill_async(self.spill_stream)
             self.spilled_host += victim.nbytes
             self.host_used += victim.nbytes
             self._host_lru[victim] = None
@@ -673,6 +739,10 @@ class GpuRunner(RunnerBase):
         self.exchanged_rows = 0
         self._side_stream = (torch.cuda.Stream(device=self.device)
                              if self.device.type == "cuda" else None)
+        # separate D2H stream: spills and prefetches use the DMA
+        # engines in both directions concurrently with compute
+        self.pool.spill_stream = (torch.cuda.Stream(device=self.device)
+                                  if self.device.type == "cuda" else None)
         if n_partitions:
             self.n_partitions = n_partitions
         elif self.world == 1 and self._inputs_fit(cap):
@@ -775,10 +845,7 @@ class GpuRunner(RunnerBase):
                     except OSError:
                         pass
                     run._disk = None
-                run.keys = None
-                run.vals = None
-                run._host = None
-                run._meta = None
+                run.drop()
             if kept:
                 store[part] = kept
             else:
