@@ -618,6 +618,10 @@ class HbmPool(object):
         # rebalances on the main stream.
 
     def touch(self, run, device):
+        """Page the run in (if spilled) and refresh it to MRU.  ``used``
+        counts every tracked resident run — round 1's model decremented
+        on release() and double-decremented on repeated touch/release
+        cycles, driving ``used`` negative and disabling eviction."""
         if not run.resident:
             if run in self._host_lru:
                 del self._host_lru[run]
@@ -627,12 +631,26 @@ class HbmPool(object):
             self.used += run.nbytes
             self._lru[run] = None
             self.balance(exclude=run)
+        elif run in self._lru:
+            del self._lru[run]          # MRU refresh
+            self._lru[run] = None
         return run
 
     def release(self, run):
-        if run.resident:
-            self.used -= run.nbytes
-        self._lru.pop(run, None)
+        """The caller is done with the run for this stage.  Accounting
+        is unchanged — the run stays tracked and evictable (an evicted
+        run's tensors survive through the caller's own references until
+        it drops them)."""
+
+    def forget(self, run):
+        """Remove a run from all tier tracking (it is being freed)."""
+        if run in self._lru:
+            del self._lru[run]
+            if run.resident:
+                self.used -= run.nbytes
+        if run in self._host_lru:
+            del self._host_lru[run]
+            self.host_used -= run.nbytes
 
     def balance(self, exclude=None):
         while self.used > self.capacity and self._lru:
@@ -835,10 +853,7 @@ class GpuRunner(RunnerBase):
                 if id(run) in live_ids:
                     kept.append(run)
                     continue
-                self.pool.release(run)
-                if run in self.pool._host_lru:
-                    del self.pool._host_lru[run]
-                    self.pool.host_used -= run.nbytes
+                self.pool.forget(run)
                 if run.on_disk:
                     try:
                         self.pool._os.unlink(run._disk)

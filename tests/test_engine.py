@@ -1422,3 +1422,30 @@ def test_cross_opaque_still_host_fallback():
                   .cross_left(Dampr.memory([10]), lambda x, y: x - y)
                   .run().read())
     assert got == host
+
+
+def test_pool_accounting_stays_consistent():
+    """Regression: repeated touch/release cycles drove ``used`` negative
+    in round 1 (double decrement), silently disabling eviction."""
+    from dampr_amd.gpu.engine import DeviceRun, HbmPool
+    pool = HbmPool(4096)
+    runs = []
+    for i in range(4):
+        k = torch.arange(128, dtype=torch.int64)    # 2 KB/run
+        r = DeviceRun(k, k.clone(), sorted=True)
+        runs.append(r)
+        pool.admit(r)
+    for _cycle in range(5):
+        for r in runs:
+            pool.touch(r, torch.device("cpu"))
+            _ = r.keys
+            pool.release(r)
+    assert pool.used >= 0
+    resident_bytes = sum(r.nbytes for r in runs if r.resident)
+    assert pool.used == resident_bytes
+    # pool stays bounded by capacity (+ at most one run of slack)
+    assert pool.used <= 4096 + runs[0].nbytes
+    for r in runs:
+        pool.forget(r)
+        r.drop()
+    assert pool.used == 0 and pool.host_used == 0
