@@ -805,6 +805,18 @@ class GpuRunner(RunnerBase):
             log.info("[device] Stage %s/%s: %r", stage_id + 1,
                      len(self.graph.stages), stage)
             ins = [data[i] for i in stage.inputs]
+            # consume-as-you-go: inputs whose LAST consumer is this
+            # stage may free each partition right after it is reduced /
+            # joined (out-of-core jobs would otherwise re-evict dead
+            # partition data until stage-end cleanup)
+            self._consume_flags = [
+                cleanup and consumers.get(src, 0) ==
+                stage.inputs.count(src) and src not in outputs
+                for src in stage.inputs]
+            self._stage_live = {
+                id(r) for k, v in data.items()
+                if isinstance(v, PartStore) and k not in set(stage.inputs)
+                for runs in v.values() for r in runs}
             with trace_stage(repr(stage), self.device):
                 if isinstance(stage, GMap):
                     out = self.run_map(stage, ins)
@@ -841,6 +853,31 @@ class GpuRunner(RunnerBase):
             log.info("[device] run stats: %s", st)
         self.pool.cleanup()
         return rets
+
+    def _consume_partition(self, ins, p):
+        """Free partition ``p`` of fully-consumed input stores (their
+        last consumer is the running stage) once its reduce/join is
+        done — bounds the tier churn of >pool jobs."""
+        flags = getattr(self, "_consume_flags", None)
+        if not flags or len(flags) != len(ins):
+            return
+        live = getattr(self, "_stage_live", frozenset())
+        seen = set()
+        for st, f in zip(ins, flags):
+            if not f or not isinstance(st, PartStore):
+                continue
+            for run in st.pop(p, []):
+                if id(run) in live or id(run) in seen:
+                    continue
+                seen.add(id(run))
+                self.pool.forget(run)
+                if run.on_disk:
+                    try:
+                        self.pool._os.unlink(run._disk)
+                    except OSError:
+                        pass
+                    run._disk = None
+                run.drop()
 
     def _free_store(self, store, live_ids=frozenset()):
         """Release a fully-consumed store's memory across all tiers
@@ -1741,6 +1778,7 @@ class GpuRunner(RunnerBase):
                 self._prefetch_partition(ins, parts, i + 1)
                 self._wait_prefetch()
                 uk, agg = self._reduce_partition(ins, p, kind)
+                self._consume_partition(ins, p)
                 if uk is None:
                     continue
                 run = DeviceRun(uk, agg, sorted=True)
@@ -1757,6 +1795,7 @@ class GpuRunner(RunnerBase):
                     ins, p, "sum",
                     vt=lambda v: torch.ones_like(v,
                                                  dtype=torch.float64))
+                self._consume_partition(ins, p)
                 run = DeviceRun(uk, sm / c, sorted=True)
                 out.setdefault(p, []).append(run)
                 self.pool.admit(run)
@@ -1766,6 +1805,7 @@ class GpuRunner(RunnerBase):
             # within equal keys, so the segment head is the first seen
             for p in self._parts(ins):
                 uk, fv = self._reduce_partition(ins, p, "first")
+                self._consume_partition(ins, p)
                 if uk is None:
                     continue
                 run = DeviceRun(uk, fv, sorted=True)
@@ -1938,6 +1978,7 @@ class GpuRunner(RunnerBase):
                                 sorted=True)
                 out.setdefault(p, []).append(run)
                 self.pool.admit(run)
+            self._consume_partition([left, right], p)
         return out
 
     @staticmethod
