@@ -854,11 +854,22 @@ class GpuRunner(RunnerBase):
         self.pool.cleanup()
         return rets
 
-    def _consume_partition(self, ins, p):
+    def _free_run(self, run):
+        self.pool.forget(run)
+        if run.on_disk:
+            try:
+                self.pool._os.unlink(run._disk)
+            except OSError:
+                pass
+            run._disk = None
+        run.drop()
+
+    def _consume_partition(self, ins, p, flags=None):
         """Free partition ``p`` of fully-consumed input stores (their
         last consumer is the running stage) once its reduce/join is
         done — bounds the tier churn of >pool jobs."""
-        flags = getattr(self, "_consume_flags", None)
+        if flags is None:
+            flags = getattr(self, "_consume_flags", None)
         if not flags or len(flags) != len(ins):
             return
         live = getattr(self, "_stage_live", frozenset())
@@ -870,14 +881,7 @@ class GpuRunner(RunnerBase):
                 if id(run) in live or id(run) in seen:
                     continue
                 seen.add(id(run))
-                self.pool.forget(run)
-                if run.on_disk:
-                    try:
-                        self.pool._os.unlink(run._disk)
-                    except OSError:
-                        pass
-                    run._disk = None
-                run.drop()
+                self._free_run(run)
 
     def _free_store(self, store, live_ids=frozenset()):
         """Release a fully-consumed store's memory across all tiers
@@ -1379,7 +1383,15 @@ class GpuRunner(RunnerBase):
                 return out
 
             def batches():
-                for store in ins:
+                # stream-consume: a batch's source runs are freed as
+                # soon as the kv transform's output exists (this stage
+                # is their last consumer), so the routed copy does not
+                # coexist with the whole unrouted input on >pool jobs
+                flags = getattr(self, "_consume_flags", None)
+                if not flags or len(flags) != len(ins):
+                    flags = [False] * len(ins)
+                live = getattr(self, "_stage_live", frozenset())
+                for store, f in zip(ins, flags):
                     if isinstance(store, PartStore) \
                             and not store.partitioned:
                         # record-wise op: stream run by run
@@ -1388,6 +1400,10 @@ class GpuRunner(RunnerBase):
                             k, v = run.keys, run.vals
                             self.pool.release(run)
                             yield k, v
+                            if f and id(run) not in live:
+                                self._free_run(run)
+                        if f:
+                            store.pop(0, None)
                     else:
                         for p in self._parts([store]):
                             keys, vals = self._merged_partition(
@@ -1395,6 +1411,8 @@ class GpuRunner(RunnerBase):
                             if keys is None:
                                 continue
                             yield keys, vals
+                            self._consume_partition([store], p,
+                                                    flags=[f])
 
             if self.world > 1:
                 # collectives must line up across ranks: apply the column
