@@ -948,9 +948,14 @@ class GpuRunner(RunnerBase):
                 st = self._partition(keys, vals, fkeys=inp.fkeys)
                 st.str_table = inp.str_table
                 return st
-            # lazy ingest: batched unpartitioned runs — partitioning by
-            # the input keys is wasted work (and a wasted spill round
-            # trip) when the first stage re-keys anyway
+            # lazy ingest: batched unpartitioned runs as VIEWS of the
+            # resident input — partitioning by the input keys is wasted
+            # work when the first stage re-keys, and cloning would send
+            # a full extra copy of the input through the spill tiers
+            # (measured 2x tier traffic on a 120 GB job).  View runs
+            # are NOT pool-admitted: the parent tensor is resident for
+            # the stage regardless (ColumnSource holds it), so they are
+            # never evicted and cost the pool nothing.
             store = PartStore(partitioned=False,
                               str_table=inp.str_table,
                               fkeys=inp.fkeys,
@@ -960,12 +965,8 @@ class GpuRunner(RunnerBase):
             step = max(1, settings.gpu_batch_records)
             for lo in range(0, n, step):
                 hi = min(lo + step, n)
-                # clone, not view: a spilled run must actually release
-                # its HBM (views pin the whole parent tensor)
-                run = DeviceRun(keys[lo:hi].clone(),
-                                vals[lo:hi].clone(), sorted=False)
+                run = DeviceRun(keys[lo:hi], vals[lo:hi], sorted=False)
                 store[0].append(run)
-                self.pool.admit(run)
             return store
         # host dataset / chunker: stream records; numeric records become
         # columns, object records stay host-side (HostStore)
