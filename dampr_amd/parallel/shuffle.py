@@ -46,6 +46,8 @@ def exchange_pairs(keys, vals, group=None):
 
     # gloo fallback (CPU tests): gather everything, keep our partition.
     rank = dist.get_rank(group)
+    dev = keys.device
+    keys, vals = keys.cpu(), vals.cpu()
     sizes = [torch.zeros(1, dtype=torch.int64) for _ in range(world)]
     dist.all_gather(sizes, torch.tensor([keys.numel()]), group=group)
     maxn = max(int(s.item()) for s in sizes)
@@ -65,7 +67,7 @@ def exchange_pairs(keys, vals, group=None):
         mine = owner_of(k, world) == rank
         outs_k.append(k[mine])
         outs_v.append(v[mine])
-    return torch.cat(outs_k), torch.cat(outs_v)
+    return torch.cat(outs_k).to(dev), torch.cat(outs_v).to(dev)
 
 
 def reorder_blob(blob, lens, order):
@@ -122,9 +124,12 @@ def exchange_keyed_payload(keys, vals, blob, lens, group=None):
         dist.all_to_all_single(rb, s_blob, out_bp, in_bp, group=group)
         return rk, rv, rb, rl
 
-    # gloo emulation
+    # gloo emulation — results must come back on the INPUT device: a
+    # CUDA caller (HipOps under gloo, e.g. multi-process CPU-backend
+    # tests on a GPU box) would otherwise feed CPU pointers into HIP
+    # kernels (observed GPU memory faults)
     rank = dist.get_rank(group)
-    obj = [None]
+    dev = keys.device
     gathered = [None] * world
     dist.all_gather_object(
         gathered,
@@ -141,8 +146,8 @@ def exchange_keyed_payload(keys, vals, blob, lens, group=None):
                          for o, n in zip(offs[own], l[own])]) \
             if int(own.sum()) else torch.empty(0, dtype=torch.int64)
         outs[2].append(b[sel])
-    return (torch.cat(outs[0]), torch.cat(outs[1]),
-            torch.cat(outs[2]), torch.cat(outs[3]))
+    return (torch.cat(outs[0]).to(dev), torch.cat(outs[1]).to(dev),
+            torch.cat(outs[2]).to(dev), torch.cat(outs[3]).to(dev))
 
 
 def exchange_columns(keys, vals, pids, world, group=None):
