@@ -1449,3 +1449,58 @@ def test_pool_accounting_stays_consistent():
         pool.forget(r)
         r.drop()
     assert pool.used == 0 and pool.host_used == 0
+
+
+def _engine_rank_sv_cross(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import operator
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        # string VALUES through join at world>1 (varlen exchange)
+        lk = np.array(["a", "b", "a", "c"] * 6)
+        lv = np.array(["v" + str(i) for i in range(24)])
+        rk = np.array(["a", "c"])
+        rv = np.array(["R_a", "R_c"])
+        out = Dampr.columns(lv, keys=lk) \
+            .join(Dampr.columns(rv, keys=rk)) \
+            .reduce(funcs.pair_right, many=True).run(runner=GpuRunner)
+        pairs = sorted(out.read())
+        gathered = [None] * world
+        dist.all_gather_object(gathered, pairs)
+        merged = sorted(p for lst in gathered for p in lst)
+        want = sorted([("a", "R_a")] * 12 + [("c", "R_c")] * 6)
+        assert merged == want, merged[:4]
+        # device cross join at world>1 (gather_columns collective)
+        got = Dampr.columns(np.array([1, 2, 3, 4], dtype=np.int64)) \
+            .cross_right(Dampr.columns(np.array([10], dtype=np.int64)),
+                         operator.mul).run(runner=GpuRunner).read()
+        gathered2 = [None] * world
+        dist.all_gather_object(gathered2, sorted(got))
+        merged2 = sorted(v for lst in gathered2 for v in lst)
+        assert merged2 == [10, 20, 30, 40], merged2
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [2])
+def test_engine_gloo_strvals_and_cross(world):
+    """world>1: var-len value exchange + broadcast cross join stay
+    rank-consistent (collective sequences must align)."""
+    port = 29000 + (os.getpid() + 300 + world) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_sv_cross,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
