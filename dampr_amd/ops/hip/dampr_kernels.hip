@@ -281,7 +281,7 @@ __device__ __forceinline__ int lds_set_insert(u64* set, u64 h) {
     return -1;
 }
 
-__global__ void __launch_bounds__(DOC_WAVES * WAVE)
+__global__ void __launch_bounds__(DOC_WAVES * WAVE, 5)
 tfidf_docs_kernel(const u8* __restrict__ text, long n,
                   const u32* __restrict__ nl_pos, long n_nl, long n_docs,
                   u64* __restrict__ cnt_keys, u64* __restrict__ cnt_vals,
