@@ -35,6 +35,14 @@ __device__ __forceinline__ bool mark_at(int mode, const u8* text,
 // ---------------------------------------------------------------- scan pass 1
 // Each block owns a contiguous [base, base + iters*TILE) byte range and
 // counts its marks.
+// SWAR newline detection: high bit of each byte equal to '\n' (5 VALU
+// ops per 4 bytes instead of ~3 per byte) — lines are ~100 bytes, so
+// matches are rare and position extraction is off the hot path.
+__device__ __forceinline__ u32 nlmask32(u32 w) {
+    u32 x = w ^ 0x0A0A0A0Au;
+    return (x - 0x01010101u) & ~x & 0x80808080u;
+}
+
 __global__ void count_marks_kernel(const u8* __restrict__ text, long n,
                                    int mode, int iters,
                                    u32* __restrict__ counts) {
@@ -46,20 +54,18 @@ __global__ void count_marks_kernel(const u8* __restrict__ text, long n,
         if (off >= n) break;
         if (off + VBYTES <= n) {
             uint4 v = *reinterpret_cast<const uint4*>(text + off);
-            const u8* b = reinterpret_cast<const u8*>(&v);
-            #pragma unroll
-            for (int j = 0; j < VBYTES; ++j) {
-                u8 c = b[j];
-                bool m;
-                if (mode == MODE_NEWLINE) {
-                    m = (c == '\n');
-                } else {
+            if (mode == MODE_NEWLINE) {
+                local += __popc(nlmask32(v.x)) + __popc(nlmask32(v.y))
+                       + __popc(nlmask32(v.z)) + __popc(nlmask32(v.w));
+            } else {
+                const u8* b = reinterpret_cast<const u8*>(&v);
+                #pragma unroll
+                for (int j = 0; j < VBYTES; ++j) {
                     bool prev_word = (j > 0)
                         ? is_word(b[j - 1])
                         : (off > 0 ? is_word(text[off - 1]) : false);
-                    m = is_word(c) && !prev_word;
+                    local += is_word(b[j]) && !prev_word;
                 }
-                local += m;
             }
         } else {
             for (long i = off; i < n; ++i)
@@ -95,19 +101,27 @@ __device__ __forceinline__ u32 gather_marks(const u8* __restrict__ text,
     if (off >= n) return 0;
     if (off + VBYTES <= n) {
         uint4 v = *reinterpret_cast<const uint4*>(text + off);
+        if (mode == MODE_NEWLINE) {
+            const u32 words[4] = {v.x, v.y, v.z, v.w};
+            #pragma unroll
+            for (int wi = 0; wi < 4; ++wi) {
+                u32 m = nlmask32(words[wi]);
+                while (m) {
+                    int bit = __ffs(m) - 1;
+                    rel[cnt++] = (u8)(wi * 4 + (bit >> 3));
+                    m &= m - 1;
+                }
+            }
+            return cnt;
+        }
         const u8* b = reinterpret_cast<const u8*>(&v);
         #pragma unroll
         for (int j = 0; j < VBYTES; ++j) {
             u8 c = b[j];
-            bool m;
-            if (mode == MODE_NEWLINE) {
-                m = (c == '\n');
-            } else {
-                bool prev_word = (j > 0)
-                    ? is_word(b[j - 1])
-                    : (off > 0 ? is_word(text[off - 1]) : false);
-                m = is_word(c) && !prev_word;
-            }
+            bool prev_word = (j > 0)
+                ? is_word(b[j - 1])
+                : (off > 0 ? is_word(text[off - 1]) : false);
+            bool m = is_word(c) && !prev_word;
             if (m) rel[cnt++] = (u8)j;
         }
     } else {
