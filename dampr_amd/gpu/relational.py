@@ -82,21 +82,35 @@ def segment_ids(sorted_keys):
 
 def group_reduce_sorted(sorted_keys, vals, op=OP_SUM):
     """Segmented reduce over a key-sorted column; returns (unique_keys,
-    aggregates)."""
+    aggregates).  Fused form (K5+K7): per-tile segment-start counts +
+    a tiny tile-base scan, then ONE pass that re-derives boundaries
+    from neighbor keys, emits each segment head's key and folds values
+    — no materialized per-element segment ids (the flags -> cumsum ->
+    nonzero -> gather chain cost 4+ full-column passes and a host
+    sync)."""
     ext = native.require()
-    seg, uniq = segment_ids(sorted_keys)
-    n_seg = uniq.numel()
+    n = sorted_keys.numel()
+    if n == 0:
+        return sorted_keys, vals[:0]
+    sorted_keys = sorted_keys.contiguous()
+    vals = vals.contiguous()
+    counts = ext.seg_count(sorted_keys)
+    scan = torch.cumsum(counts, 0, dtype=torch.int64)
+    tile_base = scan - counts
+    n_seg = int(scan[-1].item())
     if vals.dtype == torch.float64:
         init = {OP_SUM: 0.0, OP_MIN: float("inf"),
                 OP_MAX: float("-inf")}[op]
-        out = torch.full((max(n_seg, 1),), init, dtype=torch.float64,
+        out = torch.full((n_seg,), init, dtype=torch.float64,
                          device=vals.device)
     else:
         init = {OP_SUM: 0, OP_MIN: _I64_MAX, OP_MAX: _I64_MIN}[op]
-        out = torch.full((max(n_seg, 1),), init, dtype=torch.int64,
+        out = torch.full((n_seg,), init, dtype=torch.int64,
                          device=vals.device)
-    ext.seg_reduce(seg, vals, out, op)
-    return uniq, out[:n_seg]
+    uniq = torch.empty(n_seg, dtype=torch.int64,
+                       device=sorted_keys.device)
+    ext.seg_reduce_fused(sorted_keys, vals, tile_base, op, uniq, out)
+    return uniq, out
 
 
 def group_sum(keys, vals):

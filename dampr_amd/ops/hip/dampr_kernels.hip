@@ -924,6 +924,10 @@ void hj_build(torch::Tensor keys_r, torch::Tensor t_keys,
 torch::Tensor hj_count(torch::Tensor keys_l, torch::Tensor t_keys,
                        torch::Tensor t_head, torch::Tensor next,
                        long left_outer);
+torch::Tensor seg_count(torch::Tensor keys);
+void seg_reduce_fused(torch::Tensor keys, torch::Tensor vals,
+                      torch::Tensor tile_base, long op,
+                      torch::Tensor out_keys, torch::Tensor out_vals);
 void mp_merge(torch::Tensor ka, torch::Tensor pa, torch::Tensor kb,
               torch::Tensor pb, long bias_signed, torch::Tensor out_k,
               torch::Tensor out_p);
@@ -948,6 +952,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("hj_build", &hj_build, "hash-join build (chained)");
     m.def("varlen_gather", &varlen_gather,
           "row gather for byte-arena value columns");
+    m.def("seg_count", &seg_count,
+          "segment-start counts per tile (fused group-by, pass 1)");
+    m.def("seg_reduce_fused", &seg_reduce_fused,
+          "fused boundary-derive + segmented reduce over sorted keys");
     m.def("mp_merge", &mp_merge,
           "stable merge-path 2-way merge of sorted (key, payload) runs");
     m.def("hj_count", &hj_count, "hash-join probe match counts");

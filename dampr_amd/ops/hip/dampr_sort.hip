@@ -594,3 +594,174 @@ void mp_merge(torch::Tensor ka, torch::Tensor pa, torch::Tensor kb,
         (const u32*)pb.data_ptr(), nb, bias, (u64*)out_k.data_ptr(),
         (u32*)out_p.data_ptr());
 }
+
+// ----------------------------------- fused segmented reduce (K5 + K7)
+// Group-by over a key-sorted column WITHOUT materializing per-element
+// segment ids: kernel 1 counts segment starts per tile; a (tiny)
+// device scan of tile counts gives each tile its base segment id;
+// kernel 2 re-derives boundary flags from neighbor keys, scans them
+// within the tile, writes each segment head's key to out_keys and
+// folds values with the wave-segmented scan + one atomic per
+// (wave, segment) tail.  Replaces the flags -> cumsum -> nonzero ->
+// gather chain (4+ full-column passes and a host sync) with two
+// column reads.
+
+#define SEG_TPB 16
+#define SEG_SPAN (RS_BLOCK * SEG_TPB)
+
+__global__ void seg_count_kernel(const u64* __restrict__ keys, long n,
+                                 u32* __restrict__ counts) {
+    const long start = (long)blockIdx.x * SEG_SPAN;
+    const int count = (int)min((long)SEG_SPAN, n - start);
+    u32 local = 0;
+    for (int j = threadIdx.x; j < count; j += blockDim.x) {
+        const long i = start + j;
+        local += (i == 0) || (keys[i] != keys[i - 1]);
+    }
+    #pragma unroll
+    for (int d = WAVE / 2; d > 0; d >>= 1)
+        local += __shfl_down(local, d, WAVE);
+    __shared__ u32 wsum[RS_BLOCK / WAVE];
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    if (lane == 0) wsum[wid] = local;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        u32 tot = 0;
+        for (int w = 0; w < RS_BLOCK / WAVE; ++w) tot += wsum[w];
+        counts[blockIdx.x] = tot;
+    }
+}
+
+__device__ __forceinline__ void seg_atomic_fold(long* addr, long v,
+                                                int op) {
+    if (op == OP_SUM) {
+        atomicAdd((u64*)addr, (u64)v);
+    } else if (op == OP_MIN) {
+        atomicMin((long long*)addr, (long long)v);
+    } else {
+        atomicMax((long long*)addr, (long long)v);
+    }
+}
+
+__device__ __forceinline__ void seg_atomic_fold(double* addr, double v,
+                                                int op) {
+    if (op == OP_SUM) {
+        atomicAdd(addr, v);
+        return;
+    }
+    u64* a = (u64*)addr;
+    u64 old = *a;
+    while (true) {
+        double cur = __longlong_as_double((long long)old);
+        double nv = (op == OP_MIN) ? min(cur, v) : max(cur, v);
+        if (nv == cur) break;
+        u64 assumed = old;
+        old = atomicCAS(a, assumed, (u64)__double_as_longlong(nv));
+        if (old == assumed) break;
+    }
+}
+
+template <typename V>
+__global__ void __launch_bounds__(RS_BLOCK)
+seg_reduce_fused_kernel(const u64* __restrict__ keys,
+                        const V* __restrict__ vals, long n,
+                        const long* __restrict__ tile_base, int op,
+                        u64* __restrict__ out_keys,
+                        V* __restrict__ out_vals) {
+    const long start = (long)blockIdx.x * SEG_SPAN;
+    const int count = (int)min((long)SEG_SPAN, n - start);
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    __shared__ long wseg[RS_BLOCK / WAVE];  // per-wave start counts
+    __shared__ long round_tot;
+    long base = tile_base[blockIdx.x];      // starts before this round
+    for (int t = 0; t < SEG_TPB; ++t) {
+        const int j = t * RS_BLOCK + threadIdx.x;
+        const bool valid = j < count;
+        const long i = start + j;
+        u64 k = 0;
+        V v = V(0);
+        long f = 0;
+        if (valid) {
+            k = keys[i];
+            v = vals[i];
+            f = (i == 0) || (k != keys[i - 1]);
+        }
+        // wave-inclusive scan of start flags
+        long fs = f;
+        #pragma unroll
+        for (int d = 1; d < WAVE; d <<= 1) {
+            long x = __shfl_up(fs, d, WAVE);
+            if (lane >= d) fs += x;
+        }
+        const long wave_tot = __shfl(fs, WAVE - 1, WAVE);
+        if (lane == 0) wseg[wid] = wave_tot;
+        __syncthreads();
+        long wbase = base;
+        for (int w = 0; w < wid; ++w) wbase += wseg[w];
+        // seg id: starts at-or-before this element, minus one (fs == 0
+        // for a run continuing from the left -> wbase - 1)
+        const long seg = wbase + fs - 1;
+        if (valid && f)
+            out_keys[seg] = k;
+        // wave-segmented value fold (runs are contiguous)
+        V acc = v;
+        long s = valid ? seg : -1;
+        #pragma unroll
+        for (int d = 1; d < WAVE; d <<= 1) {
+            V v2 = __shfl_up(acc, d, WAVE);
+            long s2 = __shfl_up(s, d, WAVE);
+            if (lane >= d && s2 == s) {
+                if (op == OP_SUM) acc += v2;
+                else if (op == OP_MIN) acc = min(acc, v2);
+                else acc = max(acc, v2);
+            }
+        }
+        long s_next = __shfl_down(s, 1, WAVE);
+        if (valid && (lane == WAVE - 1 || s_next != s))
+            seg_atomic_fold(&out_vals[s], acc, op);
+        if (threadIdx.x == 0) {
+            long tot = 0;
+            for (int w = 0; w < RS_BLOCK / WAVE; ++w) tot += wseg[w];
+            round_tot = tot;
+        }
+        __syncthreads();
+        base += round_tot;
+        __syncthreads();
+    }
+}
+
+torch::Tensor seg_count(torch::Tensor keys) {
+    long n = keys.numel();
+    long nblocks = (n + SEG_SPAN - 1) / SEG_SPAN;
+    auto counts = torch::empty({std::max(nblocks, 1L)},
+        torch::TensorOptions().dtype(torch::kInt)
+            .device(keys.device()));
+    if (n)
+        hipLaunchKernelGGL(seg_count_kernel, dim3((u32)nblocks),
+            dim3(RS_BLOCK), 0, cur_stream(),
+            (const u64*)keys.data_ptr(), n, (u32*)counts.data_ptr());
+    return counts;
+}
+
+void seg_reduce_fused(torch::Tensor keys, torch::Tensor vals,
+                      torch::Tensor tile_base, long op,
+                      torch::Tensor out_keys, torch::Tensor out_vals) {
+    long n = keys.numel();
+    if (n == 0) return;
+    long nblocks = (n + SEG_SPAN - 1) / SEG_SPAN;
+    if (vals.dtype() == torch::kFloat64) {
+        hipLaunchKernelGGL(seg_reduce_fused_kernel<double>,
+            dim3((u32)nblocks), dim3(RS_BLOCK), 0, cur_stream(),
+            (const u64*)keys.data_ptr(), vals.data_ptr<double>(), n,
+            tile_base.data_ptr<long>(), (int)op,
+            (u64*)out_keys.data_ptr(), out_vals.data_ptr<double>());
+    } else {
+        hipLaunchKernelGGL(seg_reduce_fused_kernel<long>,
+            dim3((u32)nblocks), dim3(RS_BLOCK), 0, cur_stream(),
+            (const u64*)keys.data_ptr(), vals.data_ptr<long>(), n,
+            tile_base.data_ptr<long>(), (int)op,
+            (u64*)out_keys.data_ptr(), out_vals.data_ptr<long>());
+    }
+}
