@@ -36,8 +36,16 @@ class _OpsBase(object):
     def seg_reduce_sorted(self, sorted_keys, vals, op="sum"):
         raise NotImplementedError
 
-    def hash_join(self, keys_l, keys_r, how="inner"):
+    def hash_join(self, keys_l, keys_r, how="inner", table=None):
         raise NotImplementedError
+
+    def hash_join_build(self, keys_r):
+        """Opaque reusable build-side state for batched probing."""
+        return None
+
+    def probe_order(self, keys):
+        """Optional probe-locality permutation (None = keep order)."""
+        return None
 
     def merge_sorted_runs(self, ks, fkeys=False):
         """K-way merge of key-sorted runs (K4).  ``ks`` is a list of
@@ -96,9 +104,24 @@ class HipOps(_OpsBase):
         from .relational import group_reduce_sorted
         return group_reduce_sorted(sorted_keys, vals, _OP_IDS[op])
 
-    def hash_join(self, keys_l, keys_r, how="inner"):
+    def hash_join(self, keys_l, keys_r, how="inner", table=None):
         from .relational import hash_join
-        return hash_join(keys_l, keys_r, how)
+        return hash_join(keys_l, keys_r, how, table=table)
+
+    def hash_join_build(self, keys_r):
+        from .relational import hj_build_table
+        return hj_build_table(keys_r)
+
+    def probe_order(self, keys):
+        """Permutation clustering probe keys by their low bits (hash
+        slot = key & mask): probes then walk L2-resident table regions
+        instead of random HBM lines.  Two low-byte passes give 64K-key
+        clusters (~1 MB table window each)."""
+        if keys.numel() < (1 << 20):
+            return None
+        from .relational import radix_sort_pairs
+        _k, perm = radix_sort_pairs(keys, low_passes=2)
+        return perm.to(torch.int64)
 
     def merge_sorted_runs(self, ks, fkeys=False):
         total = sum(k.numel() for k in ks)
@@ -166,11 +189,14 @@ class TorchOps(_OpsBase):
         perm = torch.argsort(ordk, stable=True)
         return keys[perm], perm
 
-    def hash_join(self, keys_l, keys_r, how="inner"):
+    def hash_join_build(self, keys_r):
+        return self.sort_pairs(keys_r)
+
+    def hash_join(self, keys_l, keys_r, how="inner", table=None):
         assert how in ("inner", "left", "outer")
         dev = keys_l.device
         # sort right side; binary-search each left key's run
-        sr, pr = self.sort_pairs(keys_r)
+        sr, pr = table if table is not None else self.sort_pairs(keys_r)
         fl = keys_l ^ _I64_MIN
         fr = sr ^ _I64_MIN
         lo = torch.searchsorted(fr, fl, side="left")

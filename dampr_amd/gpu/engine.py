@@ -1940,11 +1940,15 @@ class GpuRunner(RunnerBase):
         cap = int(os.environ.get("DAMPR_JOIN_PROBE_ROWS",
                                  settings.gpu_join_probe_rows))
         for p in self._parts([left, right]):
-            # sorted probe keys touch table slots near-sequentially
-            # (slot = key & mask), turning random HBM probes into
-            # streaming ones — measured 3.5x on 200M-row joins
-            lk, lv = self._merged_partition([left], p)
-            rk, rv = self._merged_partition([right], p)
+            # probe keys are CLUSTERED by their low bits (hash slot =
+            # key & mask) rather than fully sorted: probes walk
+            # L2-resident table regions (measured 3.5x over random
+            # probing round 1) and the partial sort costs 2 radix
+            # passes instead of up to 8
+            lk, lv = self._merged_partition([left], p,
+                                            need_sorted=False)
+            rk, rv = self._merged_partition([right], p,
+                                            need_sorted=False)
             if lk is None and rk is None:
                 continue
             if lk is None:
@@ -1962,6 +1966,12 @@ class GpuRunner(RunnerBase):
                     and rk.numel() > lk.numel())
             if swap:
                 lk, lv, rk, rv = rk, rv, lk, lv
+            perm = self.ops.probe_order(lk)
+            if perm is not None:
+                lk = lk[perm]
+                lv = lv[perm]
+            # build the table ONCE; the skew guard probes in batches
+            table = self.ops.hash_join_build(rk)
             # inner/left matches of a probe row are independent of every
             # other probe row, so probing distributes over contiguous
             # batches; full-outer tracks unmatched RIGHT rows across the
@@ -1970,7 +1980,8 @@ class GpuRunner(RunnerBase):
             step = n if (how == "outer" or n <= cap) else cap
             for a in range(0, max(n, 1), max(step, 1)):
                 b = min(n, a + max(step, 1))
-                li, ri = self.ops.hash_join(lk[a:b], rk, how)
+                li, ri = self.ops.hash_join(lk[a:b], rk, how,
+                                            table=table)
                 li = torch.where(li >= 0, li + a, li)
                 if swap:
                     li, ri = ri, li
