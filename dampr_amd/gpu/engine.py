@@ -1966,23 +1966,27 @@ class GpuRunner(RunnerBase):
                     and rk.numel() > lk.numel())
             if swap:
                 lk, lv, rk, rv = rk, rv, lk, lv
+            # probe through a low-bit-clustered VIEW of the keys; only
+            # the (small) emitted index set maps back through the
+            # permutation — the value column is never re-gathered
             perm = self.ops.probe_order(lk)
-            if perm is not None:
-                lk = lk[perm]
-                lv = lv[perm]
+            probe_k = lk[perm] if perm is not None else lk
             # build the table ONCE; the skew guard probes in batches
             table = self.ops.hash_join_build(rk)
             # inner/left matches of a probe row are independent of every
             # other probe row, so probing distributes over contiguous
             # batches; full-outer tracks unmatched RIGHT rows across the
             # whole probe and must run in one piece.
-            n = lk.numel()
+            n = probe_k.numel()
             step = n if (how == "outer" or n <= cap) else cap
             for a in range(0, max(n, 1), max(step, 1)):
                 b = min(n, a + max(step, 1))
-                li, ri = self.ops.hash_join(lk[a:b], rk, how,
+                li, ri = self.ops.hash_join(probe_k[a:b], rk, how,
                                             table=table)
                 li = torch.where(li >= 0, li + a, li)
+                if perm is not None:
+                    li = torch.where(
+                        li >= 0, perm[torch.clamp(li, min=0)], li)
                 if swap:
                     li, ri = ri, li
                     klk, krk, klv, krv = rk, lk, rv, lv
@@ -2003,9 +2007,11 @@ class GpuRunner(RunnerBase):
                                                  valid_l, valid_r)
                 if keys.numel() == 0 and n > 0:
                     continue
-                sk, sp = self._sort(keys, fkeys=out.fkeys)
-                run = DeviceRun(sk, merged[sp.to(torch.int64)],
-                                sorted=True)
+                # emitted unsorted: consumers (collect/reduce) sort the
+                # whole output once instead of per-batch sort + merge
+                run = DeviceRun(keys.contiguous(), merged.contiguous()
+                                if not _is_sv(merged) else merged,
+                                sorted=False)
                 out.setdefault(p, []).append(run)
                 self.pool.admit(run)
             self._consume_partition([left, right], p)
