@@ -35,7 +35,7 @@ def radix_sort_pairs(keys, payload=None, low_passes=None):
         payload = torch.arange(n, dtype=torch.int32, device=keys.device)
     if n <= 1:
         return keys.clone(), payload.clone()
-    RS_SPAN = 4096
+    RS_SPAN = ext.rs_span()
     nblocks = (n + RS_SPAN - 1) // RS_SPAN
     # The caller's tensors are read-only: the first executed pass scatters
     # out of them into an owned buffer, and later passes ping-pong between

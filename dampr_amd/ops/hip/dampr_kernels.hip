@@ -924,6 +924,7 @@ void hj_build(torch::Tensor keys_r, torch::Tensor t_keys,
 torch::Tensor hj_count(torch::Tensor keys_l, torch::Tensor t_keys,
                        torch::Tensor t_head, torch::Tensor next,
                        long left_outer);
+long rs_span();
 torch::Tensor seg_count(torch::Tensor keys);
 void seg_reduce_fused(torch::Tensor keys, torch::Tensor vals,
                       torch::Tensor tile_base, long op,
@@ -952,6 +953,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("hj_build", &hj_build, "hash-join build (chained)");
     m.def("varlen_gather", &varlen_gather,
           "row gather for byte-arena value columns");
+    m.def("rs_span", &rs_span,
+          "elements per radix block (dampr_sort.hip RS_SPAN)");
     m.def("seg_count", &seg_count,
           "segment-start counts per tile (fused group-by, pass 1)");
     m.def("seg_reduce_fused", &seg_reduce_fused,

@@ -205,6 +205,8 @@ torch::Tensor rs_digit_fold(torch::Tensor keys) {
     return and_or;
 }
 
+long rs_span() { return RS_SPAN; }
+
 torch::Tensor rs_hist(torch::Tensor keys, long shift, long nblocks) {
     // int32 storage (counts <= RS_SPAN): torch.cumsum upcasts to i64 in
     // one fused pass (dtype=), so no separate conversion kernel/alloc
