@@ -335,6 +335,67 @@ def _encode_f64_sortable(x):
     return torch.where(b < 0, ~b, b ^ sign_bit)
 
 
+class _PinnedPool(object):
+    """Recycled pinned-host buffers for the spill tier: cudaHostAlloc of
+    a multi-hundred-MB buffer costs tens of ms, and a >pool job spills
+    hundreds of runs.  Power-of-two u8 buckets, views carved to size;
+    a buffer with an in-flight H2D reload is only reused after its
+    event completes."""
+
+    def __init__(self, cap_bytes=32 << 30):
+        self.free = {}                 # bucket bytes -> [u8 base]
+        self.free_bytes = 0
+        self.cap = cap_bytes
+        self._pending = []             # (base, bucket, event)
+
+    @staticmethod
+    def _bucket(nbytes):
+        return 1 << max(12, int(nbytes - 1).bit_length())
+
+    def _drain(self):
+        still = []
+        for base, b, evt in self._pending:
+            if evt is not None and not evt.query():
+                still.append((base, b, evt))
+                continue
+            if self.free_bytes + b <= self.cap:
+                self.free.setdefault(b, []).append(base)
+                self.free_bytes += b
+        self._pending = still
+
+    def get_like(self, t):
+        """Pinned host tensor with t's shape/dtype."""
+        nbytes = t.numel() * t.element_size()
+        if nbytes == 0:
+            return torch.empty_like(t, device="cpu", pin_memory=True)
+        self._drain()
+        b = self._bucket(nbytes)
+        lst = self.free.get(b)
+        if lst:
+            base = lst.pop()
+            self.free_bytes -= b
+        else:
+            base = torch.empty(b, dtype=torch.uint8, pin_memory=True)
+        return base[:nbytes].view(t.dtype).view(t.shape)
+
+    def put(self, t, event=None):
+        """Return a buffer; with ``event``, reuse waits for it (an
+        async H2D may still be reading the pinned memory)."""
+        if t is None or not isinstance(t, torch.Tensor) \
+                or not t.is_pinned():
+            return
+        base = t
+        while getattr(base, "_base", None) is not None:
+            base = base._base
+        nb = base.numel() * base.element_size()
+        if base.dtype != torch.uint8 or nb != self._bucket(nb):
+            return                     # not one of our bucket bases
+        self._pending.append((base, nb, event))
+
+
+_PIN = _PinnedPool()
+
+
 class DeviceRun(object):
     """One (keys, vals) run of a partition, spillable down the tier
     hierarchy HBM -> pinned host -> NVMe file (the reference's gzip-pickle
@@ -396,6 +457,14 @@ class DeviceRun(object):
         disk unlink).  Waits for in-flight spill DMA first — freeing a
         pinned buffer under an active copy corrupts host memory."""
         self._wait_spill()
+        if self._host is not None:
+            hk, hv = self._host
+            _PIN.put(hk)
+            if _is_sv(hv):
+                _PIN.put(hv.blob)
+                _PIN.put(hv.offs)
+            else:
+                _PIN.put(hv)
         self.keys = None
         self.vals = None
         self._host = None
@@ -416,15 +485,12 @@ class DeviceRun(object):
         with torch.cuda.stream(stream):
             # spill reads must see the producing kernels' writes
             stream.wait_stream(main)
-            hk = torch.empty_like(self.keys, device="cpu",
-                                  pin_memory=True)
+            hk = _PIN.get_like(self.keys)
             hk.copy_(self.keys, non_blocking=True)
             self.keys.record_stream(stream)
             if _is_sv(self.vals):
-                hb = torch.empty_like(self.vals.blob, device="cpu",
-                                      pin_memory=True)
-                ho = torch.empty_like(self.vals.offs, device="cpu",
-                                      pin_memory=True)
+                hb = _PIN.get_like(self.vals.blob)
+                ho = _PIN.get_like(self.vals.offs)
                 hb.copy_(self.vals.blob, non_blocking=True)
                 ho.copy_(self.vals.offs, non_blocking=True)
                 self.vals.record_stream(stream)
@@ -432,8 +498,7 @@ class DeviceRun(object):
                 self._meta = (self.keys.numel(),
                               ("str", self.vals.blob.numel()))
             else:
-                hv = torch.empty_like(self.vals, device="cpu",
-                                      pin_memory=True)
+                hv = _PIN.get_like(self.vals)
                 hv.copy_(self.vals, non_blocking=True)
                 self.vals.record_stream(stream)
                 self._meta = (self.keys.numel(), self.vals.dtype)
@@ -449,22 +514,24 @@ class DeviceRun(object):
         if self._host is not None or self.keys is None:
             return
         pin = self.keys.device.type == "cuda"
-        hk = torch.empty_like(self.keys, device="cpu", pin_memory=pin)
+
+        def _halloc(t):
+            return _PIN.get_like(t) if pin else \
+                torch.empty_like(t, device="cpu")
+
+        hk = _halloc(self.keys)
         hk.copy_(self.keys)
         if _is_sv(self.vals):
             from .strvals import StrVals
-            hb = torch.empty_like(self.vals.blob, device="cpu",
-                                  pin_memory=pin)
-            ho = torch.empty_like(self.vals.offs, device="cpu",
-                                  pin_memory=pin)
+            hb = _halloc(self.vals.blob)
+            ho = _halloc(self.vals.offs)
             hb.copy_(self.vals.blob)
             ho.copy_(self.vals.offs)
             hv = StrVals(hb, ho)
             self._meta = (self.keys.numel(),
                           ("str", self.vals.blob.numel()))
         else:
-            hv = torch.empty_like(self.vals, device="cpu",
-                                  pin_memory=pin)
+            hv = _halloc(self.vals)
             hv.copy_(self.vals)
             self._meta = (self.keys.numel(), self.vals.dtype)
         self._host = (hk, hv)
@@ -486,6 +553,12 @@ class DeviceRun(object):
                 fh.write(hv.numpy().tobytes())
         self._disk = path
         self._host = None
+        _PIN.put(hk)
+        if _is_sv(hv):
+            _PIN.put(hv.blob)
+            _PIN.put(hv.offs)
+        else:
+            _PIN.put(hv)
 
     def _load_host(self):
         if self._host is None and self._disk is not None:
@@ -524,6 +597,15 @@ class DeviceRun(object):
             self.keys = hk.to(device, non_blocking=True)
             self.vals = hv.to(device, non_blocking=True)
             self._host = None
+            if torch.device(device).type == "cuda":
+                evt = torch.cuda.Event()
+                evt.record(torch.cuda.current_stream(device))
+                _PIN.put(hk, evt)
+                if _is_sv(hv):
+                    _PIN.put(hv.blob, evt)
+                    _PIN.put(hv.offs, evt)
+                else:
+                    _PIN.put(hv, evt)
         return self
 
 
