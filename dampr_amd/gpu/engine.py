@@ -305,12 +305,16 @@ def _cat_vals(vs):
 
 
 def _run_has_sv(run):
-    """Does this run carry var-len values (any tier)?"""
-    if run.keys is not None:
-        return _is_sv(run.vals)
-    if run._host is not None:
-        return _is_sv(run._host[1])
-    return run._meta is not None and isinstance(run._meta[1], tuple)
+    """Does this run carry var-len values (any tier)?  Reads snapshot
+    attributes (IO threads may migrate the run between tiers)."""
+    vals = run.vals
+    if run.keys is not None and vals is not None:
+        return _is_sv(vals)
+    host = run._host
+    if host is not None:
+        return _is_sv(host[1])
+    meta = run._meta
+    return meta is not None and isinstance(meta[1], tuple)
 
 
 def _store_has_sv(store):
@@ -343,10 +347,12 @@ class _PinnedPool(object):
     event completes."""
 
     def __init__(self, cap_bytes=32 << 30):
+        import threading
         self.free = {}                 # bucket bytes -> [u8 base]
         self.free_bytes = 0
         self.cap = cap_bytes
         self._pending = []             # (base, bucket, event)
+        self._lock = threading.Lock()  # IO threads stage into the pool
 
     @staticmethod
     def _bucket(nbytes):
@@ -365,18 +371,26 @@ class _PinnedPool(object):
 
     def get_like(self, t):
         """Pinned host tensor with t's shape/dtype."""
-        nbytes = t.numel() * t.element_size()
+        return self.get(t.numel(), t.dtype)
+
+    def get(self, numel, dtype):
+        nbytes = numel * torch._utils._element_size(dtype)
+        pin = torch.cuda.is_available()
         if nbytes == 0:
-            return torch.empty_like(t, device="cpu", pin_memory=True)
-        self._drain()
-        b = self._bucket(nbytes)
-        lst = self.free.get(b)
-        if lst:
-            base = lst.pop()
-            self.free_bytes -= b
-        else:
-            base = torch.empty(b, dtype=torch.uint8, pin_memory=True)
-        return base[:nbytes].view(t.dtype).view(t.shape)
+            return torch.empty(0, dtype=dtype, pin_memory=pin)
+        with self._lock:
+            self._drain()
+            b = self._bucket(nbytes)
+            lst = self.free.get(b)
+            if lst:
+                base = lst.pop()
+                self.free_bytes -= b
+            else:
+                base = None
+        if base is None:
+            base = torch.empty(self._bucket(nbytes), dtype=torch.uint8,
+                               pin_memory=pin)
+        return base[:nbytes].view(dtype)
 
     def put(self, t, event=None):
         """Return a buffer; with ``event``, reuse waits for it (an
@@ -390,7 +404,8 @@ class _PinnedPool(object):
         nb = base.numel() * base.element_size()
         if base.dtype != torch.uint8 or nb != self._bucket(nb):
             return                     # not one of our bucket bases
-        self._pending.append((base, nb, event))
+        with self._lock:
+            self._pending.append((base, nb, event))
 
 
 _PIN = _PinnedPool()
@@ -403,7 +418,7 @@ class DeviceRun(object):
     host DRAM + NVMe)."""
 
     __slots__ = ("keys", "vals", "sorted", "_host", "_disk", "_meta",
-                 "_evt")
+                 "_evt", "_dfut", "_lfut")
 
     def __init__(self, keys, vals, sorted=False):
         self.keys = keys
@@ -413,6 +428,8 @@ class DeviceRun(object):
         self._disk = None
         self._meta = None
         self._evt = None           # in-flight async D2H spill marker
+        self._dfut = None          # in-flight threaded disk WRITE
+        self._lfut = None          # in-flight threaded disk READ-AHEAD
 
     @property
     def n(self):
@@ -452,11 +469,32 @@ class DeviceRun(object):
             self._evt.synchronize()
             self._evt = None
 
+    def _wait_io(self):
+        """Join any threaded disk write/read touching this run."""
+        if self._dfut is not None:
+            self._dfut.result()
+            self._dfut = None
+        if self._lfut is not None:
+            self._lfut.result()
+            self._lfut = None
+
+    def _stage_host(self):
+        """IO-thread body: page the run's bytes from NVMe into pinned
+        host buffers (read-ahead; the H2D happens at touch time).
+        Joins a pending disk WRITE first (executor FIFO guarantees the
+        write already started, so this cannot self-deadlock)."""
+        self._wait_spill()
+        if self._dfut is not None:
+            self._dfut.result()
+            self._dfut = None
+        self._load_host()
+
     def drop(self):
         """Release storage on every tier (caller owns accounting and
         disk unlink).  Waits for in-flight spill DMA first — freeing a
         pinned buffer under an active copy corrupts host memory."""
         self._wait_spill()
+        self._wait_io()
         if self._host is not None:
             hk, hv = self._host
             _PIN.put(hk)
@@ -545,12 +583,12 @@ class DeviceRun(object):
         self._wait_spill()
         hk, hv = self._host
         with open(path, "wb") as fh:
-            fh.write(hk.numpy().tobytes())
+            hk.numpy().tofile(fh)
             if _is_sv(hv):
-                fh.write(hv.offs.numpy().tobytes())
-                fh.write(hv.blob.numpy().tobytes())
+                hv.offs.numpy().tofile(fh)
+                hv.blob.numpy().tofile(fh)
             else:
-                fh.write(hv.numpy().tobytes())
+                hv.numpy().tofile(fh)
         self._disk = path
         self._host = None
         _PIN.put(hk)
@@ -562,25 +600,25 @@ class DeviceRun(object):
 
     def _load_host(self):
         if self._host is None and self._disk is not None:
-            import numpy as np
             n, vdt = self._meta
+
+            def _readinto(fh, numel, dtype):
+                t = _PIN.get(numel, dtype)
+                if numel:
+                    mv = memoryview(t.numpy()).cast("B")
+                    got = fh.readinto(mv)
+                    assert got == len(mv), "short spill-file read"
+                return t
+
             with open(self._disk, "rb") as fh:
-                raw_k = fh.read(n * 8)
-                raw_v = fh.read()
-            hk = torch.from_numpy(
-                np.frombuffer(raw_k, dtype=np.int64).copy())
-            if isinstance(vdt, tuple):
-                from .strvals import StrVals
-                off_b = (n + 1) * 8
-                ho = torch.from_numpy(
-                    np.frombuffer(raw_v[:off_b], dtype=np.int64).copy())
-                hb = torch.from_numpy(
-                    np.frombuffer(raw_v[off_b:], dtype=np.uint8).copy())
-                hv = StrVals(hb, ho)
-            else:
-                vnp = (np.float64 if vdt == torch.float64 else np.int64)
-                hv = torch.from_numpy(
-                    np.frombuffer(raw_v, dtype=vnp).copy())
+                hk = _readinto(fh, n, torch.int64)
+                if isinstance(vdt, tuple):
+                    from .strvals import StrVals
+                    ho = _readinto(fh, n + 1, torch.int64)
+                    hb = _readinto(fh, vdt[1], torch.uint8)
+                    hv = StrVals(hb, ho)
+                else:
+                    hv = _readinto(fh, n, vdt)
             os_mod = __import__("os")
             try:
                 os_mod.unlink(self._disk)
@@ -592,6 +630,7 @@ class DeviceRun(object):
     def load(self, device):
         if self.keys is None:
             self._wait_spill()
+            self._wait_io()
             self._load_host()
             hk, hv = self._host
             self.keys = hk.to(device, non_blocking=True)
@@ -641,6 +680,13 @@ class HbmPool(object):
         self.spilled_disk = 0
         self.reloaded = 0
         self._os = os
+        self._io = None            # lazy ThreadPoolExecutor (NVMe IO)
+
+    def _executor(self):
+        if self._io is None:
+            from concurrent.futures import ThreadPoolExecutor
+            self._io = ThreadPoolExecutor(max_workers=2)
+        return self._io
 
     def _next_path(self):
         self._os.makedirs(self.spill_dir, exist_ok=True)
@@ -654,6 +700,9 @@ class HbmPool(object):
     def cleanup(self):
         """Unlink any spill files still on disk (runs never paged back
         before the job finished)."""
+        if self._io is not None:
+            self._io.shutdown(wait=True)
+            self._io = None
         for p in self._disk_paths:
             try:
                 self._os.unlink(p)
@@ -679,9 +728,18 @@ class HbmPool(object):
         if stream is None:
             return
         main = torch.cuda.current_stream(device)
+        # disk-resident runs: stage file -> pinned on the IO thread NOW
+        # (the H2D happens at touch); host-resident runs take the H2D
+        # side-stream path below
+        for run in runs:
+            if run.resident or run._lfut is not None:
+                continue
+            if run.on_disk or run._dfut is not None:
+                run._lfut = self._executor().submit(run._stage_host)
         with torch.cuda.stream(stream):
             for run in runs:
-                if run.resident:
+                if run.resident or run._lfut is not None \
+                        or run._dfut is not None or run.on_disk:
                     continue
                 if run in self._host_lru:
                     del self._host_lru[run]
@@ -754,7 +812,13 @@ class HbmPool(object):
             del self._host_lru[v]
             self.host_used -= v.nbytes
             self.spilled_disk += v.nbytes
-            v.spill_to_disk(self._next_path())
+            path = self._next_path()
+            if self.spill_stream is None:
+                v.spill_to_disk(path)      # CPU/test path stays sync
+            else:
+                # NVMe writes ride an IO thread, overlapped with
+                # compute; readers join via _wait_io
+                v._dfut = self._executor().submit(v.spill_to_disk, path)
 
 
 # --------------------------------------------------------------------------
@@ -938,13 +1002,13 @@ class GpuRunner(RunnerBase):
 
     def _free_run(self, run):
         self.pool.forget(run)
+        run.drop()                  # joins in-flight spill/disk IO
         if run.on_disk:
             try:
                 self.pool._os.unlink(run._disk)
             except OSError:
                 pass
             run._disk = None
-        run.drop()
 
     def _consume_partition(self, ins, p, flags=None):
         """Free partition ``p`` of fully-consumed input stores (their
@@ -977,13 +1041,13 @@ class GpuRunner(RunnerBase):
                     kept.append(run)
                     continue
                 self.pool.forget(run)
+                run.drop()          # joins in-flight spill/disk IO
                 if run.on_disk:
                     try:
                         self.pool._os.unlink(run._disk)
                     except OSError:
                         pass
                     run._disk = None
-                run.drop()
             if kept:
                 store[part] = kept
             else:
