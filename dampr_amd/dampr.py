@@ -466,10 +466,19 @@ class PMap(PBase):
 
     def cross_set(self, other, cross, agg=None, **options):
         """Cross each value against the whole of ``other``, aggregated once
-        by ``agg`` (default list)."""
+        by ``agg`` (default list).  With a recognized commutative
+        ``cross`` (funcs.CROSS_BINOPS) AND a recognized scalar ``agg``
+        (sum/min/max), the columnar engine folds the broadcast side to
+        one scalar on device and fuses the apply (K9)."""
         def _cross(k1, v1, right):
             yield k1, cross(v1, right)
 
+        cname = funcs.cross_binop_name(cross)
+        aname = funcs.set_agg_name(agg) if agg is not None else None
+        if cname and aname:
+            options = dict(options)
+            options.setdefault("device_map",
+                               ("cross_set", cname, aname))
         if agg is None:
             agg = list
 

@@ -117,6 +117,21 @@ CROSS_BINOPS = {
 }
 
 
+# cross_set aggregate: the whole broadcast side folds to ONE scalar
+SET_AGGS = {
+    sum: "sum",
+    min: "min",
+    max: "max",
+}
+
+
+def set_agg_name(f):
+    try:
+        return SET_AGGS.get(f)
+    except TypeError:
+        return None
+
+
 def cross_binop_name(f):
     try:
         return CROSS_BINOPS.get(f)

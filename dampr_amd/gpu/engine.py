@@ -1703,6 +1703,37 @@ class GpuRunner(RunnerBase):
                                  fkeys=getattr(other, "fkeys", False))
             st.str_table = getattr(other, "str_table", None)
             return st
+        if kind == "cross_set":
+            # K9 broadcast-set: fold the broadcast side (ins[1]) to one
+            # scalar, apply cross(v_streamed, scalar) over ins[0]
+            opn, aggn = spec[1], spec[2]
+            if len(ins) != 2 \
+                    or any(not isinstance(s, PartStore) for s in ins) \
+                    or any(getattr(s, "keyed", False) for s in ins) \
+                    or any(_store_has_sv(s) for s in ins):
+                return self._host_map(stage, ins)
+            other, me = ins
+            ko, vo = self._all_rows(other)
+            km, vm = self._all_rows(me)
+            if self.world > 1:
+                from ..parallel.shuffle import gather_columns
+                km, vm = gather_columns(km, vm, device=self.device)
+            if vm.numel() == 0:
+                # host semantics: sum() of nothing is 0, min/max raise
+                return self._host_map(stage, ins)
+            scalar = {"sum": vm.sum, "min": vm.min, "max": vm.max}[aggn]()
+            a, b = vo, scalar
+            if a.dtype != b.dtype:
+                dt = torch.promote_types(a.dtype, b.dtype)
+                a, b = a.to(dt), b.to(dt)
+            out_v = {"add": lambda: a + b,
+                     "mul": lambda: a * b,
+                     "min": lambda: torch.minimum(a, b),
+                     "max": lambda: torch.maximum(a, b)}[opn]()
+            st = self._partition(ko, out_v,
+                                 fkeys=getattr(other, "fkeys", False))
+            st.str_table = getattr(other, "str_table", None)
+            return st
         if kind == "topk_local":
             # per-partition top-k candidates by value (K11); all
             # candidates meet in partition 0 for the global pass

@@ -1504,3 +1504,42 @@ def test_engine_gloo_strvals_and_cross(world):
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def test_cross_set_recognized_device_matches_host():
+    """cross_set with recognized cross + scalar agg folds the broadcast
+    side to one scalar on device."""
+    import operator
+    a = np.array([1, 2, 3], dtype=np.int64)       # streamed side
+    b = np.array([5, 7, 2], dtype=np.int64)       # folded side
+    dev = sorted(
+        Dampr.columns(a)
+        .cross_set(Dampr.columns(b), operator.add, agg=sum)
+        .run(runner=GpuRunner).read())
+    host = sorted(
+        Dampr.memory([1, 2, 3])
+        .cross_set(Dampr.memory([5, 7, 2]), operator.add, agg=sum)
+        .run().read())
+    assert dev == host
+    dev2 = sorted(
+        Dampr.columns(a)
+        .cross_set(Dampr.columns(b), operator.mul, agg=max)
+        .run(runner=GpuRunner).read())
+    host2 = sorted(
+        Dampr.memory([1, 2, 3])
+        .cross_set(Dampr.memory([5, 7, 2]), operator.mul, agg=max)
+        .run().read())
+    assert dev2 == host2
+
+
+def test_cross_set_default_agg_still_host():
+    got = sorted(
+        Dampr.columns(np.array([1, 2, 3], dtype=np.int64))
+        .cross_set(Dampr.columns(np.array([2, 3], dtype=np.int64)),
+                   lambda v, right: v in right, agg=set)
+        .run(runner=GpuRunner).read())
+    host = sorted(
+        Dampr.memory([1, 2, 3])
+        .cross_set(Dampr.memory([2, 3]),
+                   lambda v, right: v in right, agg=set).run().read())
+    assert got == host

@@ -145,3 +145,56 @@ def test_string_key_join_engines_agree(lk, rk):
             if k == k2:
                 want.append((k, int(lva[i] + rva[j])))
     assert got == sorted(want)
+
+
+@hyp_settings(max_examples=15, deadline=None, derandomize=True)
+@given(
+    lk=st.lists(st.integers(min_value=0, max_value=12), min_size=1,
+                max_size=40),
+    rk=st.lists(st.integers(min_value=0, max_value=12), min_size=1,
+                max_size=40),
+    side=st.sampled_from(["left", "right"]),
+)
+def test_string_value_join_engines_agree(lk, rk, side):
+    """Var-len VALUE columns through the device join vs host records."""
+    lka = np.array([str(k) for k in lk])
+    rka = np.array([str(k) for k in rk])
+    lva = np.array(["L{}_{}".format(k, i) for i, k in enumerate(lk)])
+    rva = np.array(["R{}_{}".format(k, i) for i, k in enumerate(rk)])
+    agg = funcs.pair_left if side == "left" else funcs.pair_right
+    dev = sorted(
+        Dampr.columns(lva, keys=lka)
+        .join(Dampr.columns(rva, keys=rka))
+        .reduce(agg, many=True).run().read())
+    host = sorted(
+        Dampr.memory(list(zip(lka.tolist(), lva.tolist())))
+        .group_by(lambda kv: kv[0], lambda kv: kv[1])
+        .join(Dampr.memory(list(zip(rka.tolist(), rva.tolist())))
+              .group_by(lambda kv: kv[0], lambda kv: kv[1]))
+        .reduce(agg, many=True)
+        .run(runner=MTRunner, n_maps=2, n_reducers=2).read())
+    assert dev == host, (dev[:4], host[:4])
+
+
+@hyp_settings(max_examples=15, deadline=None, derandomize=True)
+@given(
+    a=st.lists(st.integers(min_value=-50, max_value=50), min_size=1,
+               max_size=40),
+    b=st.lists(st.integers(min_value=-50, max_value=50), min_size=1,
+               max_size=6),
+    opname=st.sampled_from(["add", "mul", "min", "max"]),
+)
+def test_cross_engines_agree(a, b, opname):
+    """Device broadcast cross joins vs host MapCrossJoin, any op in
+    funcs.CROSS_BINOPS."""
+    import operator
+    op = {"add": operator.add, "mul": operator.mul,
+          "min": min, "max": max}[opname]
+    dev = sorted(
+        Dampr.columns(np.array(a, dtype=np.int64))
+        .cross_left(Dampr.columns(np.array(b, dtype=np.int64)), op)
+        .run().read())
+    host = sorted(
+        Dampr.memory(a).cross_left(Dampr.memory(b), op)
+        .run(runner=MTRunner, n_maps=2, n_reducers=2).read())
+    assert dev == host, (dev[:5], host[:5])

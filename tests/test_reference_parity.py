@@ -78,6 +78,13 @@ PIPELINES = {
     "cross_set": lambda D: D.memory([1, 2, 3])
         .cross_set(D.memory([2, 3]), lambda v, right: v in right,
                    agg=set),
+    # recognized (cross, agg) pair — the device engine folds the
+    # broadcast side to one scalar; the reference executes its Python
+    # MapAllJoin.  Parity pins the (streams-other, aggregates-self)
+    # direction quirk too (SURVEY §8 note).
+    "cross_set_sum": lambda D: D.memory([1, 2, 3])
+        .cross_set(D.memory([4, 5]), __import__("operator").add,
+                   agg=sum),
     # each key's 3 values are identical, so "first value seen" is
     # deterministic (arrival order within a key is scheduling-dependent
     # in both engines)
