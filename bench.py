@@ -41,12 +41,15 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    # modulo lets an N-rank run share fewer GPUs (RCCL smoke tests on a
+    # 1-GPU box); on a full node it is the identity mapping
+    dev_idx = local_rank % max(torch.cuda.device_count(), 1)
     dist_mode = world > 1
     if dist_mode:
         import torch.distributed as dist
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(dev_idx)
         dist.init_process_group("nccl")
-    device = torch.device("cuda", local_rank)
+    device = torch.device("cuda", dev_idx)
     torch.cuda.set_device(device)
 
     from dampr_amd.gpu.corpus import synth_corpus_device
