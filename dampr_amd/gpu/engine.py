@@ -1106,6 +1106,7 @@ class GpuRunner(RunnerBase):
                               str_table=inp.str_table,
                               fkeys=inp.fkeys,
                               svals=_is_sv(inp.vals))
+            store.vdtype = None if _is_sv(vals) else vals.dtype
             store[0] = []
             n = keys.numel()
             step = max(1, settings.gpu_batch_records)
@@ -1345,6 +1346,7 @@ class GpuRunner(RunnerBase):
         partition ids, stable sort by id, slice contiguous segments."""
         P = self.n_partitions
         store = PartStore(keyed=keyed, fkeys=fkeys, svals=_is_sv(vals))
+        store.vdtype = None if _is_sv(vals) else vals.dtype
         if keys.numel() == 0 and self.world == 1:
             return store
         if P == 1 and self.world == 1:
@@ -1361,14 +1363,33 @@ class GpuRunner(RunnerBase):
         pid = self.ops.partition_of(keys, P)
         order = torch.argsort(pid, stable=True)
         keys, vals, pid = keys[order], vals[order], pid[order]
-        counts = torch.bincount(pid, minlength=P)
         if self.world > 1:
             keys, vals, pid = self._exchange(keys, vals, pid)
             # received rows arrive grouped by sender, not by partition:
             # restore pid order for the contiguous slicing below
             order = torch.argsort(pid, stable=True)
             keys, vals, pid = keys[order], vals[order], pid[order]
-            counts = torch.bincount(pid, minlength=P)
+        self._slice_into(store, keys, vals, pid,
+                         already_sorted=already_sorted)
+        return store
+
+    def _count_batches(self, ins):
+        """Upper bound on the (k, v) batches ``batches()`` yields,
+        computed WITHOUT consuming anything — the local term of the
+        rank-agreed chunk count for the exchange pipeline."""
+        c = 0
+        for store in ins:
+            if isinstance(store, PartStore) and not store.partitioned:
+                c += len(store.get(0, []))
+            elif isinstance(store, PartStore):
+                c += len(self._parts([store]))
+        return c
+
+    def _slice_into(self, store, keys, vals, pid, already_sorted=False):
+        """Append contiguous partition slices of pid-ordered columns to
+        ``store`` (this rank's owned partitions only at world > 1)."""
+        P = self.n_partitions
+        counts = torch.bincount(pid, minlength=P)
         offs = torch.cumsum(counts, 0) - counts
         counts_l = counts.tolist()
         offs_l = offs.tolist()
@@ -1562,21 +1583,76 @@ class GpuRunner(RunnerBase):
                                                     flags=[f])
 
             if self.world > 1:
-                # collectives must line up across ranks: apply the column
-                # funcs locally, then ONE exchange-partition for the
-                # whole stage (ranks may own different partition counts)
-                acc_k, acc_v = [], []
-                for k, v in batches():
-                    acc_k.append(self._apply_colfunc(keyf, k, v))
-                    acc_v.append(self._apply_colfunc(valf, k, v))
-                nk = torch.cat(acc_k) if acc_k else torch.zeros(
-                    0, dtype=torch.int64, device=self.device)
-                nv = _cat_vals(acc_v) if acc_v else torch.zeros(
-                    0, dtype=torch.int64, device=self.device)
-                fkeys = nk.dtype == torch.float64
-                if fkeys:
-                    nk = _encode_f64_sortable(nk)
-                return self._partition(nk, nv, fkeys=fkeys)
+                # CHUNKED exchange pipeline: ranks agree on the chunk
+                # COUNT (one max-reduce; the collective sequence per
+                # stage stays identical on every rank), then per chunk:
+                # column funcs + partition routing on the compute
+                # stream, the data all-to-all on RCCL's comm stream.
+                # Routing chunk i+1 is issued while chunk i's exchange
+                # is in flight (only the small counts collective syncs
+                # the host); received chunks slice into runs at the
+                # end.  (SURVEY §2.3 overlap; gloo executes the same
+                # sequence synchronously.)
+                import torch.distributed as dist
+                from ..parallel.shuffle import (exchange_columns,
+                                                exchange_columns_varlen)
+                t = torch.tensor([self._count_batches(ins)])
+                dist.all_reduce(t, op=dist.ReduceOp.MAX)
+                C = int(t.item())
+                # rank-deterministic empty-chunk layout (an empty rank
+                # must match the others' collective dtypes)
+                vd = next((getattr(s, "vdtype", None) for s in ins
+                           if getattr(s, "vdtype", None) is not None),
+                          torch.int64)
+                in_sv = any(_store_has_sv(s) for s in ins)
+                it = batches()
+                fkeys = None
+                parts = []
+                for _i in range(C):
+                    try:
+                        k, v = next(it)
+                    except StopIteration:
+                        k = v = None
+                    if k is None:
+                        nk = torch.zeros(0, dtype=torch.int64,
+                                         device=self.device)
+                        if valf == "identity" and in_sv:
+                            from .strvals import StrVals
+                            nv = StrVals.empty(self.device)
+                        else:
+                            nv = torch.zeros(
+                                0,
+                                dtype=(vd if valf == "identity"
+                                       else torch.int64),
+                                device=self.device)
+                    else:
+                        nk = self._apply_colfunc(keyf, k, v)
+                        nv = self._apply_colfunc(valf, k, v)
+                    if fkeys is None:
+                        fkeys = nk.dtype == torch.float64
+                    if fkeys:
+                        nk = _encode_f64_sortable(nk)
+                    pid = self.ops.partition_of(nk, self.n_partitions)
+                    order = torch.argsort(pid, stable=True)
+                    nk, nv, pid = nk[order], nv[order], pid[order]
+                    self.exchanged_rows += nk.numel()
+                    if _is_sv(nv):
+                        parts.append(exchange_columns_varlen(
+                            nk, nv, pid, self.world))
+                    else:
+                        parts.append(exchange_columns(
+                            nk, nv, pid, self.world))
+                out = PartStore(fkeys=bool(fkeys))
+                rk = torch.cat([e[0] for e in parts]) if parts else \
+                    torch.zeros(0, dtype=torch.int64, device=self.device)
+                rv = _cat_vals([e[1] for e in parts]) if parts else \
+                    torch.zeros(0, dtype=torch.int64, device=self.device)
+                rp = torch.cat([e[2] for e in parts]) if parts else \
+                    torch.zeros(0, dtype=torch.int64, device=self.device)
+                order = torch.argsort(rp, stable=True)
+                self._slice_into(out, rk[order], rv[order], rp[order])
+                out.svals = _is_sv(rv)
+                return out
             for k, v in batches():
                 out = fold(out, kv_batch(k, v))
             return out if out is not None else PartStore()
