@@ -3,6 +3,7 @@ tests/test_engine.py executed on cuda:0 with the HipOps backend (gfx950
 kernels), checked against exact Python oracles.  A HipOps/TorchOps
 cross-check guards backend divergence."""
 import collections
+import os
 
 import numpy as np
 import pytest
@@ -346,3 +347,23 @@ def test_str_values_spill_device(tmp_path):
     tiny = run(hbm_bytes=65536, host_bytes=131072,
                spill_dir=str(tmp_path))
     assert tiny == full and len(full) > 0
+
+
+def test_disk_tier_threaded_device(tmp_path):
+    """Force all three tiers on hardware: tiny HBM + host pools push
+    runs through the threaded NVMe writes and read-ahead; results must
+    match the resident run exactly."""
+    rng = np.random.default_rng(21)
+    vals = rng.integers(0, 500, size=2_000_000).astype(np.int64)
+
+    def run(**kw):
+        return sorted(_run_dev(Dampr.columns(vals).count(),
+                               **kw).read())
+
+    full = run()
+    tiny = run(hbm_bytes=1 << 21, host_bytes=1 << 21,
+               spill_dir=str(tmp_path), n_partitions=32)
+    assert tiny == full and len(full) > 0
+    # spill files are cleaned up after the job
+    left = [f for f in os.listdir(str(tmp_path)) if f.endswith(".run")]
+    assert left == [], left
