@@ -78,28 +78,51 @@ def bench_join(rows_l, rows_r, card, device, iters, warmup):
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--rows", type=int, default=200_000_000)
+    ap.add_argument("--rows", type=int, default=200_000_000,
+                    help="rows PER RANK (weak scaling at world > 1)")
     ap.add_argument("--card", type=int, default=1_000_000)
     ap.add_argument("--iters", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--trace", action="store_true",
                     help="print per-stage timings of the last run")
     args = ap.parse_args()
+    # multi-rank (BASELINE configs 3/4 name 8x MI355X): launch via
+    # torch.distributed.run; rows are PER RANK and the engine exchanges
+    # partitions over RCCL (one process per GPU)
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        import torch.distributed as dist
+        lr = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(lr % max(torch.cuda.device_count(), 1))
+        dist.init_process_group("nccl")
     device = "cuda:0" if torch.cuda.is_available() else "cpu"
+    if world > 1:
+        device = "cuda:{}".format(
+            int(os.environ.get("LOCAL_RANK", "0"))
+            % max(torch.cuda.device_count(), 1))
     if device == "cpu":
         args.rows = min(args.rows, 1_000_000)
+    if world > 1:
+        # engine contract: every rank passes the IDENTICAL logical
+        # input and keeps its 1/world slice — scale the total so the
+        # per-rank share stays at --rows (weak scaling)
+        args.rows *= world
 
     from dampr_amd.utils.trace import get_trace
     r = bench_group_sum_values(args.rows, args.card, device, args.iters,
                                args.warmup)
     r["device"] = device
-    print(json.dumps(r))
+    r["world"] = world
+    if world <= 1 or int(os.environ.get("RANK", "0")) == 0:
+        print(json.dumps(r))
     if args.trace:
         print(get_trace().report())
     r = bench_join(args.rows, args.rows // 8, max(args.rows, 1), device,
                    args.iters, args.warmup)
     r["device"] = device
-    print(json.dumps(r))
+    r["world"] = world
+    if world <= 1 or int(os.environ.get("RANK", "0")) == 0:
+        print(json.dumps(r))
     if args.trace:
         print(get_trace().report())
 
