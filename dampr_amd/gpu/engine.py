@@ -1652,6 +1652,7 @@ class GpuRunner(RunnerBase):
                 order = torch.argsort(rp, stable=True)
                 self._slice_into(out, rk[order], rv[order], rp[order])
                 out.svals = _is_sv(rv)
+                out.vdtype = None if _is_sv(rv) else rv.dtype
                 return out
             for k, v in batches():
                 out = fold(out, kv_batch(k, v))
@@ -1997,6 +1998,9 @@ class GpuRunner(RunnerBase):
                             for s in stores
                             if getattr(s, "str_table", None) is not None),
                            None))
+        out.vdtype = next((getattr(s, "vdtype", None) for s in stores
+                           if getattr(s, "vdtype", None) is not None),
+                          None)
         for s in stores:
             for p, runs in s.items():
                 out.setdefault(p, []).extend(runs)
