@@ -1605,8 +1605,15 @@ class GpuRunner(RunnerBase):
                            if getattr(s, "vdtype", None) is not None),
                           torch.int64)
                 in_sv = any(_store_has_sv(s) for s in ins)
+                # fkeys derives from world-agreed metadata, NOT from an
+                # observed batch: a rank with zero local batches must
+                # still treat RECEIVED keys as f64-encoded when the
+                # other ranks encode
+                fkeys = (keyf == "identity" and vd == torch.float64) \
+                    or (keyf == "key"
+                        and any(getattr(s, "fkeys", False)
+                                for s in ins))
                 it = batches()
-                fkeys = None
                 parts = []
                 for _i in range(C):
                     try:
@@ -1628,8 +1635,8 @@ class GpuRunner(RunnerBase):
                     else:
                         nk = self._apply_colfunc(keyf, k, v)
                         nv = self._apply_colfunc(valf, k, v)
-                    if fkeys is None:
-                        fkeys = nk.dtype == torch.float64
+                        assert (nk.dtype == torch.float64) == fkeys, \
+                            "kv key dtype disagrees with store metadata"
                     if fkeys:
                         nk = _encode_f64_sortable(nk)
                     pid = self.ops.partition_of(nk, self.n_partitions)
@@ -1900,6 +1907,7 @@ class GpuRunner(RunnerBase):
             ns = PartStore(keyed=getattr(s, "keyed", False),
                            partitioned=False, str_table=merged,
                            svals=getattr(s, "svals", False))
+            ns.vdtype = getattr(s, "vdtype", None)
             ns[0] = []
             for p in sorted(s):
                 for run in s[p]:
@@ -1934,6 +1942,7 @@ class GpuRunner(RunnerBase):
                            partitioned=False,
                            str_table=getattr(s, "str_table", None),
                            svals=getattr(s, "svals", False))
+            ns.vdtype = getattr(s, "vdtype", None)
             ns[0] = []
             for p in sorted(s):
                 for run in s[p]:
