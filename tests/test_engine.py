@@ -1543,3 +1543,42 @@ def test_cross_set_default_agg_still_host():
         .cross_set(Dampr.memory([2, 3]),
                    lambda v, right: v in right, agg=set).run().read())
     assert got == host
+
+
+def _engine_rank_float_tiny(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        # 2 float rows < 3 ranks: an empty rank must still treat the
+        # RECEIVED keys as f64-encoded (regression: fkeys inferred from
+        # the empty first chunk flagged them as ints)
+        got = Dampr.columns(np.array([1.5, 1.5])).count().run() \
+            .read()
+        gathered = [None] * world
+        dist.all_gather_object(gathered, got)
+        merged = sorted(p for lst in gathered for p in lst)
+        assert merged == [(1.5, 2)], merged
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [3])
+def test_engine_gloo_float_keys_empty_rank(world):
+    port = 29000 + (os.getpid() + 410 + world) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_float_tiny,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
