@@ -367,3 +367,44 @@ def test_disk_tier_threaded_device(tmp_path):
     # spill files are cleaned up after the job
     left = [f for f in os.listdir(str(tmp_path)) if f.endswith(".run")]
     assert left == [], left
+
+
+def test_large_join_probe_perm_cross_backend():
+    """2M+ probe rows trigger the low-bit probe-clustering permutation;
+    emitted indices must map back exactly (TorchOps never permutes, so
+    cross-backend equality pins the mapping)."""
+    rng = np.random.default_rng(31)
+    lk = rng.integers(0, 50_000, size=2_100_000).astype(np.int64)
+    lv = rng.integers(0, 1 << 40, size=lk.size).astype(np.int64)
+    rk = rng.integers(0, 50_000, size=40_000).astype(np.int64)
+    rv = rng.integers(0, 1 << 40, size=rk.size).astype(np.int64)
+
+    def build():
+        return Dampr.columns(lv, keys=lk) \
+            .join(Dampr.columns(rv, keys=rk)) \
+            .reduce(funcs.pair_sum, many=True)
+
+    gpu = sorted(_run_dev(build()).read())
+    cpu = sorted(build().run(device="cpu").read())
+    assert gpu == cpu and len(gpu) > 0
+
+
+def test_join_swap_plus_perm_cross_backend(monkeypatch):
+    """Build-side swap (skewed right side) combined with the probe
+    permutation: cap forced low so the big right side becomes the
+    probed side at a size that activates clustering."""
+    monkeypatch.setenv("DAMPR_JOIN_PROBE_ROWS", "100000")
+    rng = np.random.default_rng(33)
+    lk = rng.integers(0, 20_000, size=150_000).astype(np.int64)
+    lv = rng.integers(0, 1 << 30, size=lk.size).astype(np.int64)
+    rk = rng.integers(0, 20_000, size=1_200_000).astype(np.int64)
+    rv = rng.integers(0, 1 << 30, size=rk.size).astype(np.int64)
+
+    def build():
+        return Dampr.columns(lv, keys=lk) \
+            .join(Dampr.columns(rv, keys=rk)) \
+            .reduce(funcs.pair_sum, many=True)
+
+    gpu = sorted(_run_dev(build()).read())
+    cpu = sorted(build().run(device="cpu").read())
+    assert gpu == cpu and len(gpu) > 0
