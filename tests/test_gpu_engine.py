@@ -395,9 +395,9 @@ def test_join_swap_plus_perm_cross_backend(monkeypatch):
     probed side at a size that activates clustering."""
     monkeypatch.setenv("DAMPR_JOIN_PROBE_ROWS", "100000")
     rng = np.random.default_rng(33)
-    lk = rng.integers(0, 20_000, size=150_000).astype(np.int64)
+    lk = rng.integers(0, 100_000, size=120_000).astype(np.int64)
     lv = rng.integers(0, 1 << 30, size=lk.size).astype(np.int64)
-    rk = rng.integers(0, 20_000, size=1_200_000).astype(np.int64)
+    rk = rng.integers(0, 100_000, size=1_200_000).astype(np.int64)
     rv = rng.integers(0, 1 << 30, size=rk.size).astype(np.int64)
 
     def build():
