@@ -170,7 +170,98 @@ def main_float(trials=200, seed=77):
     return fails
 
 
+def main_r2(trials=200, seed=909):
+    """Round-2 surfaces: string VALUES (join pair_left/right, first,
+    order), cross joins, cross_set with recognized ops."""
+    import operator
+    rng = random.Random(seed)
+    fails = 0
+    ops_cross = [operator.add, operator.mul, min, max]
+    for trial in range(trials):
+        kind = rng.choice(["svjoin", "svfirst", "cross", "cross_set"])
+        try:
+            if kind == "svjoin":
+                nl = rng.randint(1, 120)
+                nr = rng.randint(1, 120)
+                card = rng.randint(1, 25)
+                lk = np.array([str(rng.randint(0, card))
+                               for _ in range(nl)])
+                rk = np.array([str(rng.randint(0, card))
+                               for _ in range(nr)])
+                lv = np.array(["L%d" % i for i in range(nl)])
+                rv = np.array(["R%d" % i for i in range(nr)])
+                agg = rng.choice([funcs.pair_left, funcs.pair_right])
+                dev = sorted(Dampr.columns(lv, keys=lk)
+                             .join(Dampr.columns(rv, keys=rk))
+                             .reduce(agg, many=True).run().read())
+                host = sorted(
+                    Dampr.memory(list(zip(lk.tolist(), lv.tolist())))
+                    .group_by(lambda kv: kv[0], lambda kv: kv[1])
+                    .join(Dampr.memory(list(zip(rk.tolist(),
+                                                rv.tolist())))
+                          .group_by(lambda kv: kv[0],
+                                    lambda kv: kv[1]))
+                    .reduce(agg, many=True)
+                    .run(runner=MTRunner, n_maps=2,
+                         n_reducers=2).read())
+                ok = dev == host
+            elif kind == "svfirst":
+                n = rng.randint(1, 150)
+                vals = np.array([str(rng.randint(0, 20))
+                                 for _ in range(n)])
+                dev = sorted(Dampr.columns(vals).a_group_by().first()
+                             .run().read())
+                host = sorted(Dampr.memory(vals.tolist()).a_group_by()
+                              .first().run(runner=MTRunner, n_maps=2,
+                                           n_reducers=2).read())
+                ok = dev == host
+            elif kind == "cross":
+                a = [rng.randint(-50, 50) for _ in range(
+                    rng.randint(1, 60))]
+                b = [rng.randint(-50, 50) for _ in range(
+                    rng.randint(1, 8))]
+                op = rng.choice(ops_cross)
+                dev = sorted(Dampr.columns(
+                    np.array(a, dtype=np.int64))
+                    .cross_left(Dampr.columns(
+                        np.array(b, dtype=np.int64)), op)
+                    .run().read())
+                host = sorted(Dampr.memory(a).cross_left(
+                    Dampr.memory(b), op)
+                    .run(runner=MTRunner, n_maps=2,
+                         n_reducers=2).read())
+                ok = dev == host
+            else:
+                a = [rng.randint(-50, 50) for _ in range(
+                    rng.randint(1, 60))]
+                b = [rng.randint(-50, 50) for _ in range(
+                    rng.randint(1, 20))]
+                op = rng.choice(ops_cross)
+                agg = rng.choice([sum, min, max])
+                dev = sorted(Dampr.columns(
+                    np.array(a, dtype=np.int64))
+                    .cross_set(Dampr.columns(
+                        np.array(b, dtype=np.int64)), op, agg=agg)
+                    .run().read())
+                host = sorted(Dampr.memory(a).cross_set(
+                    Dampr.memory(b), op, agg=agg)
+                    .run(runner=MTRunner, n_maps=2,
+                         n_reducers=2).read())
+                ok = dev == host
+        except Exception as e:       # noqa: BLE001
+            ok = False
+            print("EXC", kind, trial, repr(e)[:200])
+        if not ok:
+            fails += 1
+            print("FAIL", kind, "trial", trial)
+        if (trial + 1) % 50 == 0:
+            print("... r2", trial + 1, "trials, fails:", fails)
+    print("done r2:", trials, "trials, fails:", fails)
+    return fails
+
+
 if __name__ == "__main__":
     t = int(sys.argv[1]) if len(sys.argv) > 1 else 500
     sys.exit(1 if (main(t) + main_str(max(t // 2, 100))
-                   + main_float(max(t // 2, 100))) else 0)
+                   + main_float(max(t // 2, 100))
+                   + (main_r2(max(t // 2, 100)) or 0)) else 0)
