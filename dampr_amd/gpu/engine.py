@@ -1429,8 +1429,18 @@ class GpuRunner(RunnerBase):
                 self.pool.release(run)
             k = torch.cat(ks) if ks else torch.zeros(
                 0, dtype=torch.int64, device=self.device)
-            v = _cat_vals(vs) if vs else torch.zeros(
-                0, dtype=torch.int64, device=self.device)
+            if vs:
+                v = _cat_vals(vs)
+            elif getattr(store, "svals", False):
+                # empty rank: the exchange layout must still match the
+                # other ranks' (var-len blob wire shape, value dtype)
+                from .strvals import StrVals
+                v = StrVals.empty(self.device)
+            else:
+                v = torch.zeros(
+                    0,
+                    dtype=getattr(store, "vdtype", None) or torch.int64,
+                    device=self.device)
             out = self._partition(k, v, keyed=store.keyed,
                                   fkeys=store.fkeys)
             out.str_table = getattr(store, "str_table", None)

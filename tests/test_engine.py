@@ -1582,3 +1582,51 @@ def test_engine_gloo_float_keys_empty_rank(world):
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def _engine_rank_w8(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        # string-VALUE join whose right side is smaller than the world:
+        # some ranks hold empty var-len shards and must still exchange
+        # the var-len wire shape (round-2 regression: empty ranks built
+        # numeric empties and desynced the collective arity)
+        lk = np.array(["k%d" % (i % 13) for i in range(200)])
+        lv = np.array(["v%d" % i for i in range(200)])
+        rk = np.array(["k%d" % i for i in range(0, 13, 2)])
+        rv = np.array(["R%d" % i for i in range(0, 13, 2)])
+        out = Dampr.columns(lv, keys=lk) \
+            .join(Dampr.columns(rv, keys=rk)) \
+            .reduce(funcs.pair_right, many=True).run().read()
+        g = [None] * world
+        dist.all_gather_object(g, sorted(out))
+        merged = sorted(p for lst in g for p in lst)
+        want_n = sum(1 for k in lk if int(k[1:]) % 2 == 0)
+        assert len(merged) == want_n, (len(merged), want_n)
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [8])
+def test_engine_gloo_world8_strvals_join(world):
+    """The driver's scale shape (8 ranks) on gloo: var-len exchange
+    with empty shards stays rank-consistent."""
+    port = 29000 + (os.getpid() + 510 + world) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_w8,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
