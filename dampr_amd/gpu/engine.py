@@ -1868,9 +1868,22 @@ class GpuRunner(RunnerBase):
                     topk_batch(vals)
             store = PartStore(fkeys=fkeys)
             if not cand_k:
-                return store
-            ck = torch.cat(cand_k)
-            cv = torch.cat(cand_v)
+                if self.world == 1:
+                    return store
+                # candidate-less rank (input shard smaller than the
+                # world) must still join the exchange with the agreed
+                # layout, or the collective sequence desyncs
+                vd = next((getattr(s, "vdtype", None) for s in ins
+                           if getattr(s, "vdtype", None) is not None),
+                          torch.int64)
+                fkeys = vd == torch.float64
+                store.fkeys = fkeys
+                ck = torch.zeros(0, dtype=torch.int64,
+                                 device=self.device)
+                cv = torch.zeros(0, dtype=vd, device=self.device)
+            else:
+                ck = torch.cat(cand_k)
+                cv = torch.cat(cand_v)
             if self.world > 1:
                 zeros = torch.zeros_like(ck)
                 ck, cv, _ = self._exchange(ck, cv, zeros)

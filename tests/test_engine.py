@@ -1630,3 +1630,47 @@ def test_engine_gloo_world8_strvals_join(world):
         p.join(timeout=60)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def _engine_rank_topk_tiny(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        # 2 rows over 3 ranks: the candidate-less rank must still join
+        # the top-k exchange (regression: early return desynced it)
+        got = Dampr.columns(np.array([5, 3], dtype=np.int64)).topk(1) \
+            .run().read()
+        g = [None] * world
+        dist.all_gather_object(g, got)
+        merged = sorted(v for lst in g for v in lst)
+        assert merged == [5], merged
+        # same with float values (empty rank must agree on f64 layout)
+        got2 = Dampr.columns(np.array([1.5, 9.5])).topk(1).run().read()
+        g2 = [None] * world
+        dist.all_gather_object(g2, got2)
+        merged2 = sorted(v for lst in g2 for v in lst)
+        assert merged2 == [9.5], merged2
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [3])
+def test_engine_gloo_topk_empty_rank(world):
+    port = 29000 + (os.getpid() + 620 + world) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_topk_tiny,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
