@@ -346,7 +346,7 @@ class _PinnedPool(object):
     a buffer with an in-flight H2D reload is only reused after its
     event completes."""
 
-    def __init__(self, cap_bytes=32 << 30):
+    def __init__(self, cap_bytes=192 << 30):
         import threading
         self.free = {}                 # bucket bytes -> [u8 base]
         self.free_bytes = 0
