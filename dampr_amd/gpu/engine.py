@@ -705,6 +705,8 @@ class HbmPool(object):
         self.spilled_disk = 0
         self.reloaded = 0
         self.clean_bytes = 0       # resident runs with a host copy
+        from collections import deque
+        self._wb_queue = deque()   # writeback candidates, admit order
         self._os = os
         self._io = None            # lazy ThreadPoolExecutor (NVMe IO)
 
@@ -745,6 +747,7 @@ class HbmPool(object):
     def admit(self, run):
         self.used += run.nbytes
         self._lru[run] = None
+        self._wb_queue.append(run)
         self.balance()
 
     def prefetch(self, runs, device, stream):
@@ -796,6 +799,7 @@ class HbmPool(object):
             run.load(device)
             self.used += run.nbytes
             self._lru[run] = None
+            self._wb_queue.append(run)
             self.balance(exclude=run)
         elif run in self._lru:
             del self._lru[run]          # MRU refresh
@@ -830,13 +834,19 @@ class HbmPool(object):
         if self.spill_stream is not None \
                 and self.used > self.capacity // 2 \
                 and self.clean_bytes * 2 < self.capacity:
+            # FIFO candidate queue (admit order ~ LRU): O(1) amortized
+            # — scanning the LRU dict would walk its clean prefix on
+            # every admit
             wrote = 0
-            for r in self._lru:
-                if wrote >= 4:
+            while wrote < 4 and self._wb_queue \
+                    and self.clean_bytes * 2 < self.capacity:
+                r = self._wb_queue.popleft()
+                if r is exclude:
+                    self._wb_queue.append(r)
                     break
-                if r is exclude or not r.resident \
-                        or r._host is not None:
-                    continue
+                if not r.resident or r._host is not None \
+                        or r not in self._lru:
+                    continue            # freed/evicted/already clean
                 r.writeback_async(self.spill_stream)
                 self.clean_bytes += r.nbytes
                 wrote += 1
