@@ -274,7 +274,9 @@ class SinkStore(object):
     """A sink stage's output: lazy view over the durable text part
     files it wrote (host parity — _SinkWorker returns its
     TextLineDatasets, so a sink's output is readable downstream without
-    re-materializing the sunk bytes in memory)."""
+    re-materializing the sunk bytes; reference: the sink's
+    TextLineDataset handles flow back as normal data,
+    dampr/dataset.py:280-282, runner.py:195-197)."""
 
     keyed = False
     fkeys = False
@@ -1072,7 +1074,10 @@ class GpuRunner(RunnerBase):
     def _consume_partition(self, ins, p, flags=None):
         """Free partition ``p`` of fully-consumed input stores (their
         last consumer is the running stage) once its reduce/join is
-        done — bounds the tier churn of >pool jobs."""
+        done — bounds the tier churn of >pool jobs.  (The reference
+        deletes intermediates only after ALL outputs are known,
+        runner.py:207-228; per-partition granularity is what lets a
+        120 GB job fit a 24 GB pool without re-evicting dead data.)"""
         if flags is None:
             flags = getattr(self, "_consume_flags", None)
         if not flags or len(flags) != len(ins):

@@ -10,7 +10,10 @@ value streams of dampr/dataset.py:119-159, re-laid-out for HBM).
 All ops are torch-tensor ops (work on CPU for tests and CUDA for the
 engine); the row gather — the hot reorder under sort/join/partition —
 uses the wave-per-row HIP kernel when the extension is loaded on a CUDA
-device, with a pure-torch fallback as the CPU oracle.
+device, with a pure-torch fallback as the CPU oracle.  (Reference
+analog: values are opaque pickled streams the engine never interprets,
+dampr/dataset.py:119-159; here they are opaque BYTES the kernels never
+interpret — only keys compare on device.)
 """
 import numpy as np
 import torch
