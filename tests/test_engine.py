@@ -1674,3 +1674,68 @@ def test_engine_gloo_topk_empty_rank(world):
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", "rank {} failed:\n{}".format(rank, status)
+
+
+def _engine_rank_sink_all(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import operator
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from dampr_amd.dampr import Dampr as DD
+        rng = np.random.default_rng(5)
+        lk = np.array([str(x) for x in rng.integers(0, 40, size=3000)])
+        lv = np.array(["val%d" % i for i in range(3000)])
+        rk = np.array([str(x) for x in range(0, 40, 3)])
+        rv = np.array(["R%s" % k for k in rk])
+        joined = Dampr.columns(lv, keys=lk) \
+            .join(Dampr.columns(rv, keys=rk)) \
+            .reduce(funcs.pair_right, many=True)
+        counts = Dampr.columns(
+            np.array([str(x) for x in
+                      rng.integers(0, 25, size=2000)])).count()
+        cvals = rng.integers(0, 50, size=500)
+        crossed = Dampr.columns(cvals) \
+            .cross_set(Dampr.columns(np.array([7], dtype=np.int64)),
+                       operator.add, agg=sum)
+        e1, e2, e3 = DD.run(joined, counts, crossed,
+                            hbm_bytes=1 << 16, host_bytes=1 << 16,
+                            spill_dir="/tmp/ks_%d_%d"
+                            % (os.getpid(), rank))
+        r = (sorted(e1.read()), sorted(e2.read()), sorted(e3.read()))
+        g = [None] * world
+        dist.all_gather_object(g, r)
+        m1 = sorted(p for lst in g for p in lst[0])
+        m2 = sorted(p for lst in g for p in lst[1])
+        m3 = sorted(v for lst in g for v in lst[2])
+        rset = set(rk.tolist())
+        want1 = sorted((k, "R" + k) for k in lk if k in rset)
+        assert m1 == want1, (len(m1), len(want1))
+        assert sum(c for _k, c in m2) == 2000
+        assert m3 == [7 + int(cvals.sum())], m3
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:       # noqa: BLE001
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.parametrize("world", [3])
+def test_engine_gloo_kitchen_sink(world):
+    """One multi-output Dampr.run at world 3 under 64 KB pools: shared
+    subgraphs, string-value joins, string-key counts and a device
+    cross_set all exchanging in one collective sequence."""
+    port = 29000 + (os.getpid() + 820 + world) % 900
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_engine_rank_sink_all,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", "rank {} failed:\n{}".format(rank, status)
